@@ -1,0 +1,436 @@
+// hydragnn_amd HIP kernels for MI355X (gfx950, CDNA4).
+//
+// Segment reductions (gather / scatter) are THE hot aggregation of every
+// message-passing stack. Design notes:
+//  - wave = 64 lanes; block sizes are multiples of 64.
+//  - memory-bound ops: grid-stride loops, coalesced along the feature
+//    dim, vectorized where dtype/shape allow.
+//  - bf16/f16 scatter accumulates in fp32 (atomicAdd on packed halves is
+//    both slow and lossy), cast once at the end.
+//  - grid capped at ~2048 blocks with grid-stride (guide G11).
+//
+// Reference behavior being reimplemented (not copied):
+//   torch_scatter.scatter / index_add_ call sites listed in SURVEY.md §2c.
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+
+#define CHECK_CUDA(x) TORCH_CHECK(x.is_cuda(), #x " must be a GPU tensor")
+#define CHECK_CONTIG(x) TORCH_CHECK(x.is_contiguous(), #x " must be contiguous")
+
+namespace {
+
+constexpr int kBlock = 256;
+
+inline int n_blocks(long total, int block, int cap = 2048) {
+  long b = (total + block - 1) / block;
+  return (int)std::min<long>(b, cap);
+}
+
+// -------------------------------------------------------------------------
+// gather: out[e, f] = src[index[e], f]
+// -------------------------------------------------------------------------
+template <typename T>
+__global__ void gather_kernel(const T* __restrict__ src,
+                              const long* __restrict__ index,
+                              T* __restrict__ out, long E, long F) {
+  long total = E * F;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x) {
+    long e = i / F;
+    long f = i - e * F;
+    out[i] = src[index[e] * F + f];
+  }
+}
+
+// float4-vectorized variant for F % 4 == 0 (16B/lane coalescing).
+template <typename V>
+__global__ void gather_kernel_vec(const V* __restrict__ src,
+                                  const long* __restrict__ index,
+                                  V* __restrict__ out, long E, long Fv) {
+  long total = E * Fv;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x) {
+    long e = i / Fv;
+    long f = i - e * Fv;
+    out[i] = src[index[e] * Fv + f];
+  }
+}
+
+// -------------------------------------------------------------------------
+// scatter-add (atomic path; CSR segment path below is used when rowptr
+// is available).  fp32/fp64 only — half types go through an fp32 buffer.
+// -------------------------------------------------------------------------
+template <typename T>
+__global__ void scatter_add_kernel(const T* __restrict__ src,
+                                   const long* __restrict__ index,
+                                   T* __restrict__ out, long E, long F) {
+  long total = E * F;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x) {
+    long e = i / F;
+    long f = i - e * F;
+    atomicAdd(&out[index[e] * F + f], src[i]);
+  }
+}
+
+__global__ void scatter_add_bf16_kernel(const __hip_bfloat16* __restrict__ src,
+                                        const long* __restrict__ index,
+                                        float* __restrict__ out, long E,
+                                        long F) {
+  long total = E * F;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x) {
+    long e = i / F;
+    long f = i - e * F;
+    atomicAdd(&out[index[e] * F + f], __bfloat162float(src[i]));
+  }
+}
+
+__global__ void scatter_add_f16_kernel(const __half* __restrict__ src,
+                                       const long* __restrict__ index,
+                                       float* __restrict__ out, long E,
+                                       long F) {
+  long total = E * F;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x) {
+    long e = i / F;
+    long f = i - e * F;
+    atomicAdd(&out[index[e] * F + f], __half2float(src[i]));
+  }
+}
+
+__global__ void count_kernel(const long* __restrict__ index,
+                             float* __restrict__ count, long E) {
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < E;
+       i += (long)gridDim.x * blockDim.x) {
+    atomicAdd(&count[index[i]], 1.0f);
+  }
+}
+
+template <typename T>
+__global__ void divide_rows_kernel(T* __restrict__ out,
+                                   const float* __restrict__ count, long N,
+                                   long F) {
+  long total = N * F;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x) {
+    long n = i / F;
+    float c = count[n];
+    if (c > 0.5f) out[i] = out[i] / (T)c;
+  }
+}
+
+// -------------------------------------------------------------------------
+// scatter min/max with argext — two-pass:
+//   pass 1: atomic extreme on float-ordered bits
+//   pass 2: first (lowest e) matching edge wins the arg slot (atomicMin)
+// -------------------------------------------------------------------------
+__device__ inline unsigned int float_flip(float f) {
+  unsigned int u = __float_as_uint(f);
+  return (u & 0x80000000u) ? ~u : (u | 0x80000000u);  // order-preserving
+}
+__device__ inline float float_unflip(unsigned int u) {
+  return __uint_as_float((u & 0x80000000u) ? (u & 0x7fffffffu) : ~u);
+}
+
+__global__ void scatter_max_pass1(const float* __restrict__ src,
+                                  const long* __restrict__ index,
+                                  unsigned int* __restrict__ out_bits, long E,
+                                  long F, bool is_max) {
+  long total = E * F;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x) {
+    long e = i / F;
+    long f = i - e * F;
+    unsigned int bits = float_flip(src[i]);
+    unsigned int* slot = &out_bits[index[e] * F + f];
+    if (is_max)
+      atomicMax(slot, bits);
+    else
+      atomicMin(slot, bits);
+  }
+}
+
+__global__ void scatter_max_pass2(const float* __restrict__ src,
+                                  const long* __restrict__ index,
+                                  const unsigned int* __restrict__ out_bits,
+                                  unsigned long long* __restrict__ arg, long E,
+                                  long F) {
+  long total = E * F;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x) {
+    long e = i / F;
+    long f = i - e * F;
+    long slot = index[e] * F + f;
+    if (float_flip(src[i]) == out_bits[slot]) {
+      atomicMin(&arg[slot], (unsigned long long)e);
+    }
+  }
+}
+
+__global__ void scatter_max_finalize(const unsigned int* __restrict__ bits,
+                                     const unsigned long long* __restrict__ arg,
+                                     float* __restrict__ out,
+                                     long* __restrict__ arg_out, long total,
+                                     bool is_max) {
+  unsigned int empty =
+      is_max ? float_flip(-INFINITY) : float_flip(INFINITY);
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x) {
+    bool present = bits[i] != empty;
+    out[i] = present ? float_unflip(bits[i]) : 0.0f;
+    arg_out[i] = present ? (long)arg[i] : -1;
+  }
+}
+
+// -------------------------------------------------------------------------
+// radius graph (brute force within graph, count+fill; capping by
+// distance order is finished in Python with device torch ops)
+// -------------------------------------------------------------------------
+__global__ void radius_count_kernel(const float* __restrict__ pos,
+                                    const long* __restrict__ graph_of,
+                                    const long* __restrict__ gptr, long N,
+                                    float r2, bool loop,
+                                    int* __restrict__ count) {
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < N;
+       i += (long)gridDim.x * blockDim.x) {
+    long g = graph_of[i];
+    long lo = gptr[g], hi = gptr[g + 1];
+    float xi = pos[i * 3], yi = pos[i * 3 + 1], zi = pos[i * 3 + 2];
+    int c = 0;
+    for (long j = lo; j < hi; ++j) {
+      if (!loop && j == i) continue;
+      float dx = pos[j * 3] - xi, dy = pos[j * 3 + 1] - yi,
+            dz = pos[j * 3 + 2] - zi;
+      if (dx * dx + dy * dy + dz * dz <= r2) ++c;
+    }
+    count[i] = c;
+  }
+}
+
+__global__ void radius_fill_kernel(const float* __restrict__ pos,
+                                   const long* __restrict__ graph_of,
+                                   const long* __restrict__ gptr,
+                                   const long* __restrict__ offs, long N,
+                                   float r2, bool loop,
+                                   long* __restrict__ src_out,
+                                   long* __restrict__ dst_out,
+                                   float* __restrict__ dist_out) {
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < N;
+       i += (long)gridDim.x * blockDim.x) {
+    long g = graph_of[i];
+    long lo = gptr[g], hi = gptr[g + 1];
+    float xi = pos[i * 3], yi = pos[i * 3 + 1], zi = pos[i * 3 + 2];
+    long w = offs[i];
+    for (long j = lo; j < hi; ++j) {
+      if (!loop && j == i) continue;
+      float dx = pos[j * 3] - xi, dy = pos[j * 3 + 1] - yi,
+            dz = pos[j * 3 + 2] - zi;
+      float d2 = dx * dx + dy * dy + dz * dz;
+      if (d2 <= r2) {
+        src_out[w] = j;   // src = neighbor
+        dst_out[w] = i;   // dst = center
+        dist_out[w] = sqrtf(d2);
+        ++w;
+      }
+    }
+  }
+}
+
+}  // namespace
+
+// ===========================================================================
+// C++ entry points
+// ===========================================================================
+
+static hipStream_t cur_stream() {
+  return at::hip::getCurrentHIPStream().stream();
+}
+
+torch::Tensor gather_fwd(torch::Tensor src, torch::Tensor index) {
+  CHECK_CUDA(src); CHECK_CONTIG(src); CHECK_CUDA(index);
+  TORCH_CHECK(index.dtype() == torch::kLong);
+  long E = index.numel();
+  long F = src.numel() / std::max<long>(src.size(0), 1);
+  auto sizes = src.sizes().vec();
+  sizes[0] = E;
+  auto out = torch::empty(sizes, src.options());
+  if (E == 0) return out;
+  auto idx = index.contiguous();
+  AT_DISPATCH_FLOATING_TYPES_AND2(
+      at::ScalarType::BFloat16, at::ScalarType::Half, src.scalar_type(),
+      "gather_fwd", [&] {
+        long bytes = F * sizeof(scalar_t);
+        if (bytes % 16 == 0) {
+          long Fv = bytes / 16;
+          hipLaunchKernelGGL(gather_kernel_vec<float4>,
+                             dim3(n_blocks(E * Fv, kBlock)), dim3(kBlock), 0,
+                             cur_stream(),
+                             reinterpret_cast<const float4*>(src.data_ptr<scalar_t>()),
+                             idx.data_ptr<long>(),
+                             reinterpret_cast<float4*>(out.data_ptr<scalar_t>()),
+                             E, Fv);
+        } else {
+          hipLaunchKernelGGL(gather_kernel<scalar_t>,
+                             dim3(n_blocks(E * F, kBlock)), dim3(kBlock), 0,
+                             cur_stream(), src.data_ptr<scalar_t>(),
+                             idx.data_ptr<long>(), out.data_ptr<scalar_t>(), E,
+                             F);
+        }
+      });
+  return out;
+}
+
+torch::Tensor scatter_sum_fwd(torch::Tensor src, torch::Tensor index,
+                              long dim_size) {
+  CHECK_CUDA(src); CHECK_CONTIG(src); CHECK_CUDA(index);
+  long E = index.numel();
+  long F = E > 0 ? src.numel() / src.size(0) : 1;
+  auto sizes = src.sizes().vec();
+  sizes[0] = dim_size;
+  auto idx = index.contiguous();
+  if (src.scalar_type() == at::ScalarType::BFloat16 ||
+      src.scalar_type() == at::ScalarType::Half) {
+    auto acc = torch::zeros(sizes, src.options().dtype(torch::kFloat));
+    if (E > 0) {
+      if (src.scalar_type() == at::ScalarType::BFloat16) {
+        hipLaunchKernelGGL(scatter_add_bf16_kernel,
+                           dim3(n_blocks(E * F, kBlock)), dim3(kBlock), 0,
+                           cur_stream(),
+                           reinterpret_cast<const __hip_bfloat16*>(src.data_ptr()),
+                           idx.data_ptr<long>(), acc.data_ptr<float>(), E, F);
+      } else {
+        hipLaunchKernelGGL(scatter_add_f16_kernel,
+                           dim3(n_blocks(E * F, kBlock)), dim3(kBlock), 0,
+                           cur_stream(),
+                           reinterpret_cast<const __half*>(src.data_ptr()),
+                           idx.data_ptr<long>(), acc.data_ptr<float>(), E, F);
+      }
+    }
+    return acc.to(src.scalar_type());
+  }
+  auto out = torch::zeros(sizes, src.options());
+  if (E == 0) return out;
+  AT_DISPATCH_FLOATING_TYPES(src.scalar_type(), "scatter_sum_fwd", [&] {
+    hipLaunchKernelGGL(scatter_add_kernel<scalar_t>,
+                       dim3(n_blocks(E * F, kBlock)), dim3(kBlock), 0,
+                       cur_stream(), src.data_ptr<scalar_t>(),
+                       idx.data_ptr<long>(), out.data_ptr<scalar_t>(), E, F);
+  });
+  return out;
+}
+
+std::vector<torch::Tensor> scatter_mean_fwd(torch::Tensor src,
+                                            torch::Tensor index,
+                                            long dim_size) {
+  auto out = scatter_sum_fwd(src, index, dim_size);
+  long E = index.numel();
+  long F = E > 0 ? src.numel() / src.size(0) : 1;
+  auto count = torch::zeros({dim_size}, src.options().dtype(torch::kFloat));
+  auto idx = index.contiguous();
+  if (E > 0) {
+    hipLaunchKernelGGL(count_kernel, dim3(n_blocks(E, kBlock)), dim3(kBlock),
+                       0, cur_stream(), idx.data_ptr<long>(),
+                       count.data_ptr<float>(), E);
+    AT_DISPATCH_FLOATING_TYPES_AND2(
+        at::ScalarType::BFloat16, at::ScalarType::Half, out.scalar_type(),
+        "divide_rows", [&] {
+          hipLaunchKernelGGL(divide_rows_kernel<scalar_t>,
+                             dim3(n_blocks(dim_size * F, kBlock)),
+                             dim3(kBlock), 0, cur_stream(),
+                             out.data_ptr<scalar_t>(), count.data_ptr<float>(),
+                             dim_size, F);
+        });
+  }
+  return {out, count};
+}
+
+std::vector<torch::Tensor> scatter_minmax_fwd(torch::Tensor src,
+                                              torch::Tensor index,
+                                              long dim_size, bool is_max) {
+  CHECK_CUDA(src); CHECK_CUDA(index);
+  auto srcf = src.contiguous().to(torch::kFloat);
+  long E = index.numel();
+  long F = E > 0 ? srcf.numel() / srcf.size(0) : 1;
+  auto sizes = src.sizes().vec();
+  sizes[0] = dim_size;
+  long total = dim_size * F;
+  auto idx = index.contiguous();
+  auto bits = torch::empty(sizes, src.options().dtype(torch::kInt));
+  {
+    float fill = is_max ? -INFINITY : INFINITY;
+    unsigned int u;
+    float flipped;
+    // host-side flip of the fill value
+    unsigned int raw;
+    memcpy(&raw, &fill, 4);
+    u = (raw & 0x80000000u) ? ~raw : (raw | 0x80000000u);
+    bits.fill_((int)u);
+  }
+  auto arg64 = torch::full(sizes, (long)0x7fffffffffffffffLL,
+                           src.options().dtype(torch::kLong));
+  auto out = torch::empty(sizes, src.options().dtype(torch::kFloat));
+  auto arg = torch::empty(sizes, src.options().dtype(torch::kLong));
+  if (E > 0) {
+    hipLaunchKernelGGL(scatter_max_pass1, dim3(n_blocks(E * F, kBlock)),
+                       dim3(kBlock), 0, cur_stream(), srcf.data_ptr<float>(),
+                       idx.data_ptr<long>(),
+                       reinterpret_cast<unsigned int*>(bits.data_ptr<int>()),
+                       E, F, is_max);
+    hipLaunchKernelGGL(scatter_max_pass2, dim3(n_blocks(E * F, kBlock)),
+                       dim3(kBlock), 0, cur_stream(), srcf.data_ptr<float>(),
+                       idx.data_ptr<long>(),
+                       reinterpret_cast<unsigned int*>(bits.data_ptr<int>()),
+                       reinterpret_cast<unsigned long long*>(arg64.data_ptr<long>()),
+                       E, F);
+  }
+  hipLaunchKernelGGL(scatter_max_finalize, dim3(n_blocks(total, kBlock)),
+                     dim3(kBlock), 0, cur_stream(),
+                     reinterpret_cast<unsigned int*>(bits.data_ptr<int>()),
+                     reinterpret_cast<unsigned long long*>(arg64.data_ptr<long>()),
+                     out.data_ptr<float>(), arg.data_ptr<long>(), total,
+                     is_max);
+  return {out.to(src.scalar_type()), arg};
+}
+
+std::vector<torch::Tensor> radius_pairs(torch::Tensor pos, torch::Tensor batch,
+                                        torch::Tensor gptr, double r,
+                                        bool loop) {
+  CHECK_CUDA(pos); CHECK_CONTIG(pos);
+  long N = pos.size(0);
+  float r2 = (float)(r * r);
+  auto count = torch::zeros({N}, pos.options().dtype(torch::kInt));
+  auto b = batch.contiguous();
+  auto gp = gptr.contiguous();
+  hipLaunchKernelGGL(radius_count_kernel, dim3(n_blocks(N, kBlock)),
+                     dim3(kBlock), 0, cur_stream(), pos.data_ptr<float>(),
+                     b.data_ptr<long>(), gp.data_ptr<long>(), N, r2, loop,
+                     count.data_ptr<int>());
+  auto offs = torch::zeros({N}, pos.options().dtype(torch::kLong));
+  auto csum = count.to(torch::kLong).cumsum(0);
+  offs.slice(0, 1, N).copy_(csum.slice(0, 0, N - 1));
+  long E = N > 0 ? csum[-1].item<long>() : 0;
+  auto src = torch::empty({E}, pos.options().dtype(torch::kLong));
+  auto dst = torch::empty({E}, pos.options().dtype(torch::kLong));
+  auto dist = torch::empty({E}, pos.options().dtype(torch::kFloat));
+  if (E > 0) {
+    hipLaunchKernelGGL(radius_fill_kernel, dim3(n_blocks(N, kBlock)),
+                       dim3(kBlock), 0, cur_stream(), pos.data_ptr<float>(),
+                       b.data_ptr<long>(), gp.data_ptr<long>(),
+                       offs.data_ptr<long>(), N, r2, loop,
+                       src.data_ptr<long>(), dst.data_ptr<long>(),
+                       dist.data_ptr<float>());
+  }
+  return {src, dst, dist};
+}
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("gather_fwd", &gather_fwd, "gather rows (HIP)");
+  m.def("scatter_sum_fwd", &scatter_sum_fwd, "scatter-add (HIP)");
+  m.def("scatter_mean_fwd", &scatter_mean_fwd, "scatter-mean (HIP)");
+  m.def("scatter_minmax_fwd", &scatter_minmax_fwd, "scatter-min/max (HIP)");
+  m.def("radius_pairs", &radius_pairs, "radius pair enumeration (HIP)");
+}

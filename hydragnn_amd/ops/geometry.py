@@ -1,0 +1,225 @@
+"""Geometric primitives: edge vectors/lengths and radius-graph builders.
+
+Mirrors the behavior of the reference's shared primitive
+get_edge_vectors_and_lengths (/root/reference/hydragnn/utils/model/
+operations.py:21) and the RadiusGraph / RadiusGraphPBC factories
+(/root/reference/hydragnn/preprocess/graph_samples_checks_and_updates.py:
+112-417, vesin-backed) — reimplemented from scratch: open-boundary
+neighbor search via a cell-list (numpy on CPU, HIP kernel on GPU for the
+per-layer dynamic rebuild SchNet needs), PBC via explicit shift-vector
+enumeration with mixed-PBC support and max-neighbor capping by distance.
+"""
+
+from __future__ import annotations
+
+import math
+from typing import Optional, Tuple
+
+import numpy as np
+import torch
+
+from ._extension import get_extension, use_eager
+
+__all__ = [
+    "get_edge_vectors_and_lengths",
+    "radius_graph",
+    "radius_graph_pbc",
+]
+
+
+def get_edge_vectors_and_lengths(
+    positions: torch.Tensor,
+    edge_index: torch.Tensor,
+    shifts: Optional[torch.Tensor] = None,
+    normalize: bool = False,
+    eps: float = 1e-9,
+) -> Tuple[torch.Tensor, torch.Tensor]:
+    """vectors[e] = pos[dst[e]] - pos[src[e]] + shifts[e]; lengths = |v|.
+
+    Differentiable (double-backward capable) — pure tensor ops, so the
+    force pass autograd.grad(E, pos, create_graph=True) flows through.
+    """
+    sender, receiver = edge_index[0], edge_index[1]
+    vectors = positions[receiver] - positions[sender]
+    if shifts is not None:
+        vectors = vectors + shifts.to(vectors.dtype)
+    lengths = torch.linalg.norm(vectors, dim=-1, keepdim=True)
+    if normalize:
+        vectors = vectors / (lengths + eps)
+    return vectors, lengths
+
+
+# ---------------------------------------------------------------------------
+# Open-boundary radius graph
+# ---------------------------------------------------------------------------
+def _radius_graph_torch(
+    pos: torch.Tensor,
+    r: float,
+    batch: Optional[torch.Tensor],
+    max_num_neighbors: int,
+    loop: bool,
+) -> torch.Tensor:
+    """Dense fallback: fine for per-sample preprocessing and small
+    per-layer rebuilds; HIP cell-list kernel handles the hot path."""
+    n = pos.shape[0]
+    if n == 0:
+        return torch.zeros(2, 0, dtype=torch.long, device=pos.device)
+    d = torch.cdist(pos, pos)
+    mask = d <= r
+    if not loop:
+        mask.fill_diagonal_(False)
+    if batch is not None:
+        mask &= batch.view(-1, 1) == batch.view(1, -1)
+    if max_num_neighbors < n:
+        # keep the max_num_neighbors closest sources per destination
+        d_masked = torch.where(mask, d, torch.full_like(d, float("inf")))
+        k = min(max_num_neighbors, n)
+        _, idx = torch.topk(d_masked, k, dim=1, largest=False)
+        keep = torch.zeros_like(mask)
+        keep.scatter_(1, idx, True)
+        mask &= keep
+    dst, src = mask.nonzero(as_tuple=True)
+    return torch.stack([src, dst], dim=0)
+
+
+def radius_graph(
+    pos: torch.Tensor,
+    r: float,
+    batch: Optional[torch.Tensor] = None,
+    max_num_neighbors: int = 32,
+    loop: bool = False,
+) -> torch.Tensor:
+    """edge_index [2, E] with src row 0, dst row 1; each dst keeps at most
+    max_num_neighbors closest sources within radius r (same contract as
+    torch_cluster.radius used by PyG RadiusGraph)."""
+    if pos.is_cuda and not use_eager():
+        ext = get_extension(required=True)
+        n = pos.shape[0]
+        if batch is None:
+            batch_t = torch.zeros(n, dtype=torch.long, device=pos.device)
+            gptr = torch.tensor([0, n], dtype=torch.long, device=pos.device)
+        else:
+            batch_t = batch.long()
+            counts = torch.bincount(batch_t)
+            gptr = torch.zeros(
+                counts.numel() + 1, dtype=torch.long, device=pos.device)
+            gptr[1:] = counts.cumsum(0)
+        src, dst, dist = ext.radius_pairs(
+            pos.contiguous().float(), batch_t, gptr, float(r), bool(loop))
+        if max_num_neighbors < n and dst.numel() > 0:
+            # cap: keep the max_num_neighbors closest srcs per dst
+            order = torch.argsort(dst * (dist.max() + 1.0) + dist)
+            src, dst, dist = src[order], dst[order], dist[order]
+            counts = torch.bincount(dst, minlength=n)
+            seg_start = torch.zeros(n, dtype=torch.long, device=pos.device)
+            seg_start[1:] = counts.cumsum(0)[:-1]
+            pos_in_seg = (
+                torch.arange(dst.numel(), device=pos.device)
+                - seg_start[dst])
+            keep = pos_in_seg < max_num_neighbors
+            src, dst = src[keep], dst[keep]
+        return torch.stack([src, dst], dim=0)
+    return _radius_graph_torch(pos, r, batch, max_num_neighbors, loop)
+
+
+# ---------------------------------------------------------------------------
+# Periodic (PBC) radius graph — preprocessing-time, CPU/numpy
+# ---------------------------------------------------------------------------
+def radius_graph_pbc(
+    pos: torch.Tensor,
+    r: float,
+    cell: torch.Tensor,
+    pbc=(True, True, True),
+    max_num_neighbors: int = 1000000,
+    loop: bool = False,
+):
+    """Periodic neighbor list with integer shift vectors.
+
+    Returns (edge_index [2,E], edge_shifts [E,3]) where
+    edge_shifts = S @ cell and vectors use pos[dst]-pos[src]+shift.
+    Replaces the reference's vesin path (graph_samples_checks_and_
+    updates.py:172) with an explicit image enumeration: number of images
+    per lattice direction chosen from the cell's perpendicular widths so
+    a cutoff larger than the box is still correct; mixed PBC simply
+    zeroes the non-periodic directions.
+    """
+    device = pos.device
+    dtype = pos.dtype
+    p = pos.detach().cpu().double().numpy()
+    c = cell.detach().cpu().double().numpy().reshape(3, 3)
+    pbc = np.asarray(pbc, dtype=bool).reshape(3)
+    n = p.shape[0]
+    if n == 0:
+        return (torch.zeros(2, 0, dtype=torch.long, device=device),
+                torch.zeros(0, 3, dtype=dtype, device=device))
+
+    # Perpendicular width of the cell along each lattice direction:
+    # h_i = V / |a_j x a_k| ; images needed: ceil(r / h_i).
+    vol = abs(np.linalg.det(c))
+    n_img = np.zeros(3, dtype=int)
+    for i in range(3):
+        if not pbc[i]:
+            continue
+        j, k = (i + 1) % 3, (i + 2) % 3
+        cross = np.cross(c[j], c[k])
+        area = np.linalg.norm(cross)
+        h = vol / area if area > 0 else np.inf
+        n_img[i] = int(math.ceil(r / h)) if h > 0 and np.isfinite(h) else 0
+
+    shifts_int = []
+    ranges = [range(-n_img[i], n_img[i] + 1) for i in range(3)]
+    for sx in ranges[0]:
+        for sy in ranges[1]:
+            for sz in ranges[2]:
+                shifts_int.append((sx, sy, sz))
+    shifts_int = np.array(shifts_int, dtype=np.float64)  # [S,3]
+    shift_cart = shifts_int @ c  # [S,3]
+
+    src_list, dst_list, sh_list = [], [], []
+    r2 = r * r
+    for s_idx in range(shift_cart.shape[0]):
+        sh = shift_cart[s_idx]
+        is_zero = np.all(shifts_int[s_idx] == 0)
+        # d[i,j] = |p[j] + sh - p[i]|  (edge i->j means vector p[j]-p[i]+(-sh)?)
+        # Convention: edge (src=i, dst=j) with shift S means
+        # vec = p[j] - p[i] + S@cell. Enumerate all i,j pairs.
+        diff = p[None, :, :] + sh[None, None, :] - p[:, None, :]  # [i,j,3]
+        d2 = np.einsum("ijk,ijk->ij", diff, diff)
+        m = d2 <= r2
+        if is_zero and not loop:
+            np.fill_diagonal(m, False)
+        ii, jj = np.nonzero(m)
+        if ii.size:
+            src_list.append(ii)
+            dst_list.append(jj)
+            sh_list.append(np.repeat(sh[None, :], ii.size, axis=0))
+
+    if not src_list:
+        return (torch.zeros(2, 0, dtype=torch.long, device=device),
+                torch.zeros(0, 3, dtype=dtype, device=device))
+
+    src = np.concatenate(src_list)
+    dst = np.concatenate(dst_list)
+    sh = np.concatenate(sh_list)
+
+    # Edge convention: (src -> dst): vec = p[dst] - p[src] + shift.
+    # Above we computed p[j] + sh - p[i] for pair (i, j) => src=i, dst=j,
+    # shift=sh. Good.
+
+    if max_num_neighbors < n * 27:
+        # cap neighbors per dst by distance order (lexsort (dst, length))
+        vec = p[dst] - p[src] + sh
+        length = np.sqrt(np.einsum("ij,ij->i", vec, vec))
+        order = np.lexsort((length, dst))
+        src, dst, sh = src[order], dst[order], sh[order]
+        counts = np.bincount(dst, minlength=n)
+        keep = np.ones(len(dst), dtype=bool)
+        if np.any(counts > max_num_neighbors):
+            pos_in_seg = np.arange(len(dst)) - np.concatenate(
+                ([0], np.cumsum(counts)[:-1]))[dst]
+            keep = pos_in_seg < max_num_neighbors
+        src, dst, sh = src[keep], dst[keep], sh[keep]
+
+    edge_index = torch.from_numpy(np.stack([src, dst])).long().to(device)
+    edge_shifts = torch.from_numpy(sh).to(dtype).to(device)
+    return edge_index, edge_shifts
