@@ -1,0 +1,120 @@
+"""Dataset splitting and dataloader creation.
+
+Reference: hydragnn/preprocess/load_data.py:234-516 —
+create_dataloaders with DistributedSampler (or distributed cost-aware
+node-budget batch sampler), split_dataset with stratified option.
+The loader collates our Data into Batch (data.py) and sorts edges by
+destination so the HIP segment kernels get CSR rows for free.
+"""
+
+from __future__ import annotations
+
+import os
+import random
+from typing import List, Optional, Sequence, Tuple
+
+import torch
+import torch.distributed as dist
+from torch.utils.data import DataLoader as TorchDataLoader
+from torch.utils.data.distributed import DistributedSampler
+
+from ..data import Batch, Data
+from .batch_sampler import CostAwareBatchSampler, DistributedCostAwareBatchSampler
+
+
+def _collate(data_list: Sequence[Data]) -> Batch:
+    return Batch.from_data_list(list(data_list))
+
+
+def split_dataset(dataset, perc_train: float,
+                  stratify_splitting: bool = False, seed: int = 0):
+    """Random (or composition-stratified) train/val/test split:
+    train = perc_train, val = test = (1-perc_train)/2."""
+    n = len(dataset)
+    idx = list(range(n))
+    rng = random.Random(seed)
+    if stratify_splitting:
+        from .compositional_splitting import compositional_stratified_split
+        return compositional_stratified_split(dataset, perc_train, seed)
+    rng.shuffle(idx)
+    n_train = int(n * perc_train)
+    n_val = int(n * (1.0 - perc_train) / 2.0)
+    train_idx = idx[:n_train]
+    val_idx = idx[n_train:n_train + n_val]
+    test_idx = idx[n_train + n_val:]
+    take = lambda ids: [dataset[i] for i in ids]
+    return take(train_idx), take(val_idx), take(test_idx)
+
+
+def create_dataloaders(trainset, valset, testset, batch_size: int,
+                       sampler_shuffle: bool = True, group=None,
+                       oversampling: bool = False,
+                       num_samples: Optional[int] = None,
+                       config=None):
+    """Build train/val/test loaders.  With torch.distributed
+    initialized, uses DistributedSampler; node-budget batching
+    (Training.Batching.mode = node_budget) uses the cost-aware
+    samplers."""
+    num_workers = int(os.getenv("HYDRAGNN_NUM_WORKERS", "0"))
+    batching = None
+    if config is not None:
+        batching = config["NeuralNetwork"]["Training"].get("Batching")
+
+    use_dist = dist.is_initialized() and dist.get_world_size() > 1
+
+    def make(ds, shuffle):
+        if ds is None or len(ds) == 0:
+            return TorchDataLoader([], batch_size=batch_size,
+                                   collate_fn=_collate)
+        if batching is not None and batching.get("mode") == "node_budget":
+            max_nodes = batching["max_nodes"]
+            if use_dist:
+                bs = DistributedCostAwareBatchSampler(
+                    ds, max_nodes=max_nodes, shuffle=shuffle,
+                    seed=batching.get("seed", 0),
+                    oversized_policy=batching.get("oversized_policy",
+                                                  "error"),
+                    drop_last=batching.get("drop_last", False))
+            else:
+                bs = CostAwareBatchSampler(
+                    ds, max_nodes=max_nodes, shuffle=shuffle,
+                    seed=batching.get("seed", 0),
+                    oversized_policy=batching.get("oversized_policy",
+                                                  "error"),
+                    drop_last=batching.get("drop_last", False))
+            return TorchDataLoader(ds, batch_sampler=bs,
+                                   collate_fn=_collate,
+                                   num_workers=num_workers)
+        if use_dist:
+            sampler = DistributedSampler(ds, shuffle=shuffle and
+                                         sampler_shuffle)
+            return TorchDataLoader(ds, batch_size=batch_size,
+                                   sampler=sampler, collate_fn=_collate,
+                                   num_workers=num_workers,
+                                   pin_memory=torch.cuda.is_available())
+        return TorchDataLoader(ds, batch_size=batch_size,
+                               shuffle=shuffle and sampler_shuffle,
+                               collate_fn=_collate,
+                               num_workers=num_workers,
+                               pin_memory=torch.cuda.is_available())
+
+    return make(trainset, True), make(valset, False), make(testset, False)
+
+
+def dataset_loading_and_splitting(config):
+    """Config-driven dataset load + split + loader creation for the
+    'unit_test'-style serialized flows (reference load_data.py:214)."""
+    ds_config = config["Dataset"]
+    fmt = ds_config.get("format", "pickle")
+    if fmt == "pickle":
+        from ..utils.datasets.pickledataset import SimplePickleDataset
+        paths = ds_config["path"]
+        trainset = SimplePickleDataset(paths["train"])
+        valset = SimplePickleDataset(paths["validate"])
+        testset = SimplePickleDataset(paths["test"])
+    else:
+        raise ValueError(f"Unsupported dataset format {fmt}")
+    return create_dataloaders(
+        trainset, valset, testset,
+        batch_size=config["NeuralNetwork"]["Training"]["batch_size"],
+        config=config)

@@ -1,0 +1,373 @@
+"""Training / validation / test loop.
+
+Behavioral parity with /root/reference/hydragnn/train/
+train_validate_test.py:182-1124: per-epoch train/val/test with
+multi-task loss aggregation, head-index bookkeeping over concatenated
+data.y, precision plumbing (bf16 autocast / fp32 / fp64), checkpoint
+trigger with warmup, early stopping, eval-only mode
+(HYDRAGNN_EVALONLY), HYDRAGNN_MAX_NUM_BATCH clamp, distributed metric
+reductions and padded eval gathers over RCCL.
+"""
+
+from __future__ import annotations
+
+import os
+from contextlib import nullcontext
+
+import torch
+import torch.distributed as dist
+
+from ..models.create import resolve_precision
+from ..utils.distributed import check_remaining_time, get_device
+from ..utils.model.model import Checkpoint, EarlyStopping, save_model
+from ..utils.print.print_utils import iterate_tqdm, log, print_distributed
+from ..utils.profiling_and_tracing import tracer as tr
+from ..utils.profiling_and_tracing.profile import Profiler
+from ..utils.profiling_and_tracing.time_utils import Timer
+
+
+def _unwrap(model):
+    return model.module if hasattr(model, "module") else model
+
+
+def move_batch_to_device(data, param_dtype):
+    device = get_device()
+    for key in list(data.keys()):
+        v = data[key]
+        if torch.is_tensor(v) and torch.is_floating_point(v):
+            data[key] = v.to(dtype=param_dtype)
+    return data.to(device, non_blocking=True)
+
+
+def get_autocast_and_scaler(precision):
+    precision, _, autocast_dtype = resolve_precision(precision)
+    if precision == "bf16":
+        device = get_device()
+        use_bf16 = device.type == "cuda" or bool(
+            getattr(torch.backends.cpu, "has_bf16", False))
+        if use_bf16:
+            return torch.autocast(device_type=device.type,
+                                  dtype=autocast_dtype), None
+    return nullcontext(), None
+
+
+def get_nbatch(loader):
+    nbatch = len(loader)
+    if os.getenv("HYDRAGNN_MAX_NUM_BATCH") is not None:
+        nbatch = min(nbatch, int(os.environ["HYDRAGNN_MAX_NUM_BATCH"]))
+    if dist.is_initialized() and dist.get_world_size() > 1:
+        t = torch.tensor([nbatch], dtype=torch.long)
+        dist.all_reduce(t, op=dist.ReduceOp.MIN)
+        nbatch = int(t.item())
+    return nbatch
+
+
+# ---------------------------------------------------------------------------
+# head indices over concatenated data.y
+# ---------------------------------------------------------------------------
+def get_head_indices(model, data):
+    m = _unwrap(model)
+    if all(t == "graph" for t in m.head_type):
+        return _head_indices_graph(m, data)
+    return _head_indices_node_or_mixed(m, data)
+
+
+def _head_indices_graph(m, data):
+    nsize = data.y.shape[0]
+    if m.num_heads == 1:
+        return [torch.arange(nsize, device=data.y.device)]
+    head_dims = m.head_dims
+    head_dimsum = sum(head_dims)
+    batch_size = int(data.batch.max()) + 1
+    head_index = []
+    dev = data.y.device
+    for ihead in range(m.num_heads):
+        per = torch.arange(head_dims[ihead], device=dev).repeat(batch_size)
+        shift = sum(head_dims[:ihead]) + torch.repeat_interleave(
+            torch.arange(batch_size, device=dev) * head_dimsum,
+            head_dims[ihead])
+        head_index.append(per + shift)
+    return head_index
+
+
+def _head_indices_node_or_mixed(m, data):
+    if m.num_heads == 1:
+        return [torch.arange(data.y.shape[0], device=data.y.device)]
+    y_loc = data.y_loc
+    batch_size = int(data.batch.max()) + 1
+    total_size = y_loc[:, -1]
+    sample_start = (torch.cumsum(total_size, 0) - total_size).view(-1, 1)
+    start_index = sample_start + y_loc[:, :-1]
+    end_index = sample_start + y_loc[:, 1:]
+    index_range = torch.arange(int(end_index[-1, -1]), device=y_loc.device)
+    head_index = []
+    for ihead in range(m.num_heads):
+        segs = [index_range[int(start_index[s, ihead]):
+                            int(end_index[s, ihead])]
+                for s in range(batch_size)]
+        head_index.append(torch.cat(segs, 0))
+    return head_index
+
+
+# ---------------------------------------------------------------------------
+# metric reductions
+# ---------------------------------------------------------------------------
+@torch.no_grad()
+def reduce_values_ranks(local_tensor):
+    if dist.is_initialized() and dist.get_world_size() > 1:
+        dist.all_reduce(local_tensor, op=dist.ReduceOp.SUM)
+        local_tensor = local_tensor / dist.get_world_size()
+    return local_tensor
+
+
+@torch.no_grad()
+def gather_tensor_ranks(head_values):
+    if not dist.is_initialized() or dist.get_world_size() == 1:
+        return head_values
+    head_values = head_values.to(get_device())
+    size_local = torch.tensor([head_values.shape[0]], dtype=torch.int64,
+                              device=head_values.device)
+    size_all = [torch.ones_like(size_local)
+                for _ in range(dist.get_world_size())]
+    dist.all_gather(size_all, size_local)
+    size_all = torch.cat(size_all, 0)
+    max_size = int(size_all.max())
+    padded = torch.zeros(max_size, *head_values.shape[1:],
+                         dtype=head_values.dtype, device=head_values.device)
+    padded[: head_values.shape[0]] = head_values
+    gathered = [torch.zeros_like(padded)
+                for _ in range(dist.get_world_size())]
+    dist.all_gather(gathered, padded)
+    return torch.cat([g[: int(s)] for g, s in zip(gathered, size_all)], 0)
+
+
+# ---------------------------------------------------------------------------
+# inner loops
+# ---------------------------------------------------------------------------
+def _compute_loss(model, data, use_interatomic: bool, create_graph: bool):
+    m = _unwrap(model)
+    if use_interatomic:
+        data.pos.requires_grad_(True)
+        pred = model(data)
+        loss, tasks_loss = m.energy_force_loss(
+            pred, data, create_graph=create_graph)
+    else:
+        head_index = get_head_indices(model, data)
+        pred = model(data)
+        loss, tasks_loss = m.loss(pred, data.y, head_index)
+    return loss, tasks_loss, pred
+
+
+def train(loader, model, opt, verbosity, precision="fp32", profiler=None):
+    if profiler is None:
+        profiler = Profiler()
+    m = _unwrap(model)
+    num_tasks = len(m.loss_weights) if not _use_interatomic(m) else 3
+    device = get_device()
+    total_error = torch.zeros(1, device=device)
+    tasks_error = torch.zeros(num_tasks, device=device)
+    num_samples_local = 0
+    model.train()
+    _, param_dtype, _ = resolve_precision(precision)
+    autocast, scaler = get_autocast_and_scaler(precision)
+    nbatch = get_nbatch(loader)
+    use_ia = _use_interatomic(m)
+
+    for ibatch, data in enumerate(iterate_tqdm(loader, verbosity)):
+        if ibatch >= nbatch:
+            break
+        tr.start("h2d")
+        data = move_batch_to_device(data, param_dtype)
+        tr.stop("h2d")
+        opt.zero_grad(set_to_none=True)
+        tr.start("forward")
+        with autocast:
+            loss, tasks_loss, _ = _compute_loss(model, data, use_ia,
+                                                create_graph=True)
+        tr.stop("forward")
+        tr.start("backward")
+        loss.backward()
+        tr.stop("backward")
+        tr.start("opt_step")
+        opt.step()
+        tr.stop("opt_step")
+        profiler.step()
+        n = data.num_graphs
+        total_error += loss.detach() * n
+        for it, tl in enumerate(tasks_loss):
+            tasks_error[it] += tl.detach() * n
+        num_samples_local += n
+
+    denom = max(num_samples_local, 1)
+    train_error = reduce_values_ranks(total_error / denom)
+    train_tasks_error = reduce_values_ranks(tasks_error / denom)
+    return train_error, train_tasks_error
+
+
+@torch.no_grad()
+def _eval_pass(loader, model, verbosity, precision, return_samples=False):
+    m = _unwrap(model)
+    use_ia = _use_interatomic(m)
+    num_tasks = len(m.loss_weights) if not use_ia else 3
+    device = get_device()
+    total_error = torch.zeros(1, device=device)
+    tasks_error = torch.zeros(num_tasks, device=device)
+    num_samples_local = 0
+    model.eval()
+    _, param_dtype, _ = resolve_precision(precision)
+    autocast, _ = get_autocast_and_scaler(precision)
+    nbatch = get_nbatch(loader)
+    true_values = [[] for _ in range(m.num_heads)]
+    predicted_values = [[] for _ in range(m.num_heads)]
+
+    for ibatch, data in enumerate(loader):
+        if ibatch >= nbatch:
+            break
+        data = move_batch_to_device(data, param_dtype)
+        if use_ia:
+            with torch.enable_grad():
+                data.pos.requires_grad_(True)
+                with autocast:
+                    pred = model(data)
+                    loss, tasks_loss = m.energy_force_loss(
+                        pred, data, create_graph=False)
+        else:
+            head_index = get_head_indices(model, data)
+            with autocast:
+                pred = model(data)
+                loss, tasks_loss = m.loss(pred, data.y, head_index)
+            if return_samples:
+                ytrue = data.y
+                for ihead in range(m.num_heads):
+                    p = pred[0][ihead] if m.var_output else pred[ihead]
+                    true_values[ihead].append(
+                        ytrue[head_index[ihead]].reshape(-1, 1).detach())
+                    predicted_values[ihead].append(
+                        p.reshape(-1, 1).detach())
+        n = data.num_graphs
+        total_error += loss.detach() * n
+        for it, tl in enumerate(tasks_loss):
+            tasks_error[it] += tl.detach() * n
+        num_samples_local += n
+
+    denom = max(num_samples_local, 1)
+    err = reduce_values_ranks(total_error / denom)
+    tasks_err = reduce_values_ranks(tasks_error / denom)
+    if return_samples:
+        tv = [gather_tensor_ranks(torch.cat(v, 0)) if v else torch.zeros(0, 1)
+              for v in true_values]
+        pv = [gather_tensor_ranks(torch.cat(v, 0)) if v else torch.zeros(0, 1)
+              for v in predicted_values]
+        return err, tasks_err, tv, pv
+    return err, tasks_err
+
+
+def validate(loader, model, verbosity, precision="fp32"):
+    return _eval_pass(loader, model, verbosity, precision)
+
+
+def test(loader, model, verbosity, precision="fp32", return_samples=True):
+    out = _eval_pass(loader, model, verbosity, precision,
+                     return_samples=return_samples)
+    if return_samples:
+        return out
+    err, tasks_err = out
+    return err, tasks_err, [], []
+
+
+def _use_interatomic(m) -> bool:
+    return hasattr(m, "energy_force_loss")
+
+
+# ---------------------------------------------------------------------------
+# outer loop
+# ---------------------------------------------------------------------------
+def train_validate_test(
+    model,
+    optimizer,
+    train_loader,
+    val_loader,
+    test_loader,
+    writer,
+    scheduler,
+    config,
+    log_name,
+    verbosity,
+    create_plots=False,
+    compute_grad_energy=False,
+):
+    num_epoch = config["Training"]["num_epoch"]
+    epoch_start = config["Training"].get("epoch_start", 0)
+    precision = config["Training"].get("precision", "fp32")
+    EarlyStop = config["Training"].get("EarlyStopping", False)
+    early_stopping = EarlyStopping(
+        patience=config["Training"].get("patience", 10)) if EarlyStop else None
+    use_checkpoint = config["Training"].get("Checkpoint", False)
+    checkpoint = Checkpoint(
+        name=log_name,
+        warmup=config["Training"].get("checkpoint_warmup", 0),
+    ) if use_checkpoint else None
+
+    profiler = Profiler(config.get("Profile", {}))
+    timer = Timer("train_validate_test")
+    timer.start()
+
+    if os.getenv("HYDRAGNN_EVALONLY"):
+        err, tasks_err, tv, pv = test(test_loader, model, verbosity,
+                                      precision)
+        print_distributed(verbosity, f"eval-only test error: {err}")
+        return
+
+    total_epochs = 0
+    import time as _time
+    for epoch in range(epoch_start, num_epoch):
+        epoch_t0 = _time.time()
+        for loader in (train_loader, val_loader, test_loader):
+            sampler = getattr(loader, "sampler", None)
+            if sampler is not None and hasattr(sampler, "set_epoch"):
+                sampler.set_epoch(epoch)
+            bsampler = getattr(loader, "batch_sampler", None)
+            if bsampler is not None and hasattr(bsampler, "set_epoch"):
+                bsampler.set_epoch(epoch)
+
+        profiler.set_epoch(epoch)
+        tr.start("train")
+        train_error, train_tasks_error = train(
+            train_loader, model, optimizer, verbosity, precision, profiler)
+        tr.stop("train")
+        val_error, val_tasks_error = validate(val_loader, model, verbosity,
+                                              precision)
+        test_error, test_tasks_error = _eval_pass(
+            test_loader, model, verbosity, precision)
+
+        if scheduler is not None:
+            scheduler.step(val_error)
+        if writer is not None:
+            writer.add_scalar("train_error", train_error.item(), epoch)
+            writer.add_scalar("val_error", val_error.item(), epoch)
+            writer.add_scalar("test_error", test_error.item(), epoch)
+
+        print_distributed(
+            verbosity,
+            f"Epoch {epoch}: train {train_error.item():.6f}, "
+            f"val {val_error.item():.6f}, test {test_error.item():.6f}")
+        log(f"Epoch {epoch}: train {train_error.item():.6f}, "
+            f"val {val_error.item():.6f}, test {test_error.item():.6f}")
+
+        if checkpoint is not None and checkpoint(
+                epoch, val_error.item()):
+            save_model(model, optimizer, log_name)
+
+        if early_stopping is not None:
+            early_stopping(val_error.item())
+            if early_stopping.early_stop:
+                print_distributed(verbosity,
+                                  f"Early stopping at epoch {epoch}")
+                break
+        total_epochs += 1
+        if check_remaining_time(None, _time.time() - epoch_t0):
+            print_distributed(verbosity, "Stopping: SLURM time limit near")
+            break
+
+    timer.stop()
+    tr.save(f"logs/{log_name}")

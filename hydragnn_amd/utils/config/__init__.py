@@ -1,0 +1,8 @@
+from .config_utils import (
+    update_config,
+    save_config,
+    merge_config,
+    get_log_name_config,
+    update_multibranch_heads,
+    parse_deepspeed_config,
+)

@@ -1,0 +1,265 @@
+"""JSON-config normalization.
+
+Behavioral parity with the reference config system
+(/root/reference/hydragnn/utils/input_config_parsing/config_utils.py:
+26-184 update_config, :446 save_config, :474 merge_config) and
+update_multibranch_heads (/root/reference/hydragnn/utils/model/
+model.py:316): defaults for the ~30 architecture keys, output-dim
+inference from y_loc, pna_deg / avg_num_neighbors injection, edge_dim
+rules, head-config modernization, deep merge.
+"""
+
+from __future__ import annotations
+
+import copy
+import json
+import os
+from typing import Any, Dict, List, Optional
+
+import torch
+
+_ARCH_DEFAULT_NONE = [
+    "radius", "radial_type", "distance_transform", "num_gaussians",
+    "num_filters", "envelope_exponent", "num_after_skip", "num_before_skip",
+    "basis_emb_size", "int_emb_size", "out_emb_size", "num_radial",
+    "num_spherical", "correlation", "max_ell", "node_max_ell",
+]
+
+
+def update_multibranch_heads(heads: Dict[str, Any]) -> Dict[str, Any]:
+    """Modernize output_heads: dict-style {'graph': {...}} becomes the
+    multi-branch list form [{'type': 'branch-0', 'architecture': {...}}]."""
+    out = {}
+    for level, cfg in heads.items():
+        if isinstance(cfg, list):
+            branches = []
+            for b in cfg:
+                if "architecture" in b:
+                    branches.append(b)
+                else:
+                    bid = b.get("type", f"branch-{len(branches)}")
+                    branches.append({"type": bid, "architecture": b})
+            out[level] = branches
+        else:
+            out[level] = [{"type": "branch-0", "architecture": cfg}]
+    return out
+
+
+def check_if_graph_size_variable(*loaders) -> bool:
+    sizes = set()
+    for loader in loaders:
+        if loader is None:
+            continue
+        ds = loader.dataset
+        n = min(len(ds), 50)
+        for i in range(n):
+            sizes.add(ds[i].num_nodes)
+        if len(sizes) > 1:
+            return True
+    return len(sizes) > 1
+
+
+def _gather_deg(dataset) -> torch.Tensor:
+    from ...ops import degree
+    max_deg = 0
+    hists = []
+    for data in dataset:
+        d = degree(data.edge_index[1], data.num_nodes, torch.long)
+        hists.append(torch.bincount(d))
+        max_deg = max(max_deg, hists[-1].numel())
+    out = torch.zeros(max_deg, dtype=torch.long)
+    for h in hists:
+        out[: h.numel()] += h
+    if torch.distributed.is_initialized():
+        torch.distributed.all_reduce(out)
+    return out
+
+
+def _calculate_avg_deg(dataset) -> float:
+    num_edges = 0
+    num_nodes = 0
+    for data in dataset:
+        num_edges += data.num_edges
+        num_nodes += data.num_nodes
+    t = torch.tensor([num_edges, num_nodes], dtype=torch.float64)
+    if torch.distributed.is_initialized():
+        torch.distributed.all_reduce(t)
+    return float(t[0] / t[1].clamp(min=1))
+
+
+def update_config(config, train_loader, val_loader, test_loader):
+    """Normalize the raw JSON config with dataset-derived fields."""
+    gsv = os.getenv("HYDRAGNN_USE_VARIABLE_GRAPH_SIZE")
+    if gsv is None:
+        graph_size_variable = check_if_graph_size_variable(
+            train_loader, val_loader, test_loader)
+    else:
+        graph_size_variable = bool(int(gsv))
+
+    nn = config["NeuralNetwork"]
+    arch = nn["Architecture"]
+    training = nn["Training"]
+    var = nn["Variables_of_interest"]
+
+    arch.setdefault("global_attn_engine", None)
+    arch.setdefault("global_attn_type", None)
+    arch.setdefault("global_attn_heads", 0)
+    arch.setdefault("pe_dim", 0)
+    training.setdefault("global_attn_redraw_interval", 1000)
+    arch.setdefault("equivariant_attn_lmax", 1)
+    arch.setdefault("equivariant_attn_num_radial", 16)
+    arch.setdefault("equivariant_attn_feedforward_multiplier", 2)
+    arch.setdefault("equivariant_attn_allow_scalar_only", False)
+    arch.setdefault("equivariant_attn_require_tensor_coupling", True)
+    arch.setdefault("equivariant_attn_chunk_size", 512)
+    arch.setdefault("equivariant_attn_coupling_mode", "parallel")
+
+    batching = training.get("Batching")
+    if batching is not None:
+        mode = batching.get("mode", "fixed")
+        if mode not in ("fixed", "node_budget"):
+            raise ValueError(f"unsupported batching mode: {mode}")
+        if mode == "node_budget" and "max_nodes" not in batching:
+            raise ValueError("node_budget batching requires max_nodes")
+
+    arch["output_heads"] = update_multibranch_heads(arch["output_heads"])
+
+    # --- output dims from data / y_loc ---
+    data0 = train_loader.dataset[0]
+    output_type = var["type"]
+    if arch.get("enable_interatomic_potential", False):
+        dims_list = var["output_dim"]
+    elif data0.get("y_loc") is not None:
+        dims_list = []
+        for ihead in range(len(output_type)):
+            span = int(data0.y_loc[0, ihead + 1]) - int(data0.y_loc[0, ihead])
+            if output_type[ihead] == "graph":
+                dims_list.append(span)
+            elif output_type[ihead] == "node":
+                if (graph_size_variable and
+                        arch["output_heads"]["node"][0]["architecture"]["type"]
+                        == "mlp_per_node"):
+                    raise ValueError(
+                        "mlp_per_node not allowed for variable graph size")
+                dims_list.append(span // data0.num_nodes)
+            else:
+                raise ValueError(f"Unknown output type {output_type[ihead]}")
+    else:
+        for t in output_type:
+            if t != "graph":
+                raise ValueError("y_loc needed for non-graph outputs")
+        dims_list = var["output_dim"]
+    arch["output_dim"] = dims_list
+    arch["output_type"] = output_type
+    arch["num_nodes"] = data0.num_nodes
+
+    var.setdefault("denormalize_output", False)
+
+    arch["input_dim"] = len(var["input_node_features"])
+
+    # --- PNA degree histogram / MACE avg_num_neighbors ---
+    if arch["mpnn_type"] in ("PNA", "PNAPlus", "PNAEq"):
+        if hasattr(train_loader.dataset, "pna_deg") and \
+                train_loader.dataset.pna_deg is not None:
+            deg = torch.tensor(train_loader.dataset.pna_deg)
+        else:
+            deg = _gather_deg(train_loader.dataset)
+        arch["pna_deg"] = deg.tolist()
+        arch["max_neighbours"] = len(deg) - 1
+    else:
+        arch["pna_deg"] = None
+
+    if arch["mpnn_type"] == "CGCNN" and not arch.get("global_attn_engine"):
+        arch["hidden_dim"] = arch["input_dim"]
+
+    if arch["mpnn_type"] == "MACE":
+        if hasattr(train_loader.dataset, "avg_num_neighbors") and \
+                train_loader.dataset.avg_num_neighbors is not None:
+            arch["avg_num_neighbors"] = float(
+                train_loader.dataset.avg_num_neighbors)
+        else:
+            arch["avg_num_neighbors"] = _calculate_avg_deg(
+                train_loader.dataset)
+    else:
+        arch["avg_num_neighbors"] = None
+
+    for key in _ARCH_DEFAULT_NONE:
+        arch.setdefault(key, None)
+    arch.setdefault("enable_interatomic_potential", False)
+
+    # --- edge dim rules ---
+    arch["edge_dim"] = None
+    edge_models = ["GAT", "PNA", "PNAPlus", "PAINN", "PNAEq", "CGCNN",
+                   "SchNet", "EGNN", "DimeNet", "MACE"]
+    if arch.get("edge_features"):
+        assert arch["mpnn_type"] in edge_models, (
+            "Edge features only with " + ",".join(edge_models))
+        arch["edge_dim"] = len(arch["edge_features"])
+        assert not arch.get("enable_interatomic_potential", False), (
+            "Edge features cannot be combined with interatomic potentials")
+    elif arch["mpnn_type"] == "CGCNN":
+        arch["edge_dim"] = 0
+
+    arch.setdefault("equivariance", None)
+    arch.setdefault("freeze_conv_layers", False)
+    arch.setdefault("initial_bias", None)
+    arch.setdefault("activation_function", "relu")
+    arch.setdefault("SyncBatchNorm", False)
+    training.setdefault("conv_checkpointing", False)
+    training.setdefault("loss_function_type", "mse")
+    training.setdefault("Optimizer", {"type": "AdamW", "learning_rate": 1e-3})
+    training.setdefault("precision", "fp32")
+    return config
+
+
+def save_config(config, log_name: str, path: str = "./logs/") -> None:
+    fname = os.path.join(path, log_name, "config.json")
+    os.makedirs(os.path.dirname(fname), exist_ok=True)
+    rank = 0
+    if torch.distributed.is_initialized():
+        rank = torch.distributed.get_rank()
+    if rank == 0:
+        with open(fname, "w") as f:
+            json.dump(config, f, indent=2)
+
+
+def merge_config(base: Dict, override: Dict) -> Dict:
+    """Deep merge (override wins)."""
+    out = copy.deepcopy(base)
+    for k, v in override.items():
+        if k in out and isinstance(out[k], dict) and isinstance(v, dict):
+            out[k] = merge_config(out[k], v)
+        else:
+            out[k] = copy.deepcopy(v)
+    return out
+
+
+def get_log_name_config(config) -> str:
+    arch = config["NeuralNetwork"]["Architecture"]
+    training = config["NeuralNetwork"]["Training"]
+    name = config.get("Dataset", {}).get("name", "dataset")
+    cut = name.rfind("_") if name.rfind("_") > 0 else None
+    return (
+        f"{arch['mpnn_type']}-r-{arch.get('radius')}"
+        f"-ncl-{arch['num_conv_layers']}-hd-{arch['hidden_dim']}"
+        f"-ne-{training['num_epoch']}"
+        f"-lr-{training['Optimizer']['learning_rate']}"
+        f"-bs-{training['batch_size']}"
+        f"-data-{name[:cut]}"
+    )
+
+
+def parse_deepspeed_config(config) -> Dict:
+    """The reference can emit a DeepSpeed config
+    (config_utils.py:455); we keep the entry point for API parity. On
+    MI355X the native path is DDP/FSDP over RCCL."""
+    training = config["NeuralNetwork"]["Training"]
+    return {
+        "train_batch_size": training.get("batch_size", 32),
+        "optimizer": {
+            "type": training.get("Optimizer", {}).get("type", "AdamW"),
+            "params": {
+                "lr": training.get("Optimizer", {}).get("learning_rate", 1e-3)
+            },
+        },
+    }

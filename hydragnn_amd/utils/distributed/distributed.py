@@ -1,0 +1,240 @@
+"""Process/comm bootstrap and distributed model wrappers.
+
+MI355X-native redesign of /root/reference/hydragnn/utils/distributed/
+distributed.py:113-644.  One process per GPU; torch.distributed with the
+"nccl" backend IS RCCL on ROCm, running bucketed all-reduce /
+reduce-scatter / all-gather over the 8-GPU xGMI clique.  Bootstrap comes
+from torchrun/env vars (RANK / WORLD_SIZE / LOCAL_RANK, with
+OMPI/SLURM fallbacks) — no mpi4py dependency.
+
+Env flags mirrored from the reference:
+  HYDRAGNN_BACKEND, HYDRAGNN_MASTER_ADDR / _PORT,
+  HYDRAGNN_USE_FSDP, HYDRAGNN_FSDP_VERSION, HYDRAGNN_FSDP_STRATEGY.
+"""
+
+from __future__ import annotations
+
+import datetime
+import os
+import socket
+from typing import Optional, Tuple
+
+import torch
+import torch.distributed as dist
+
+
+def init_comm_size_and_rank() -> Tuple[int, int]:
+    """World size / rank from the launcher environment."""
+    if os.getenv("WORLD_SIZE") is not None:
+        return int(os.environ["WORLD_SIZE"]), int(os.environ.get("RANK", 0))
+    if os.getenv("OMPI_COMM_WORLD_SIZE") is not None:
+        return (int(os.environ["OMPI_COMM_WORLD_SIZE"]),
+                int(os.environ["OMPI_COMM_WORLD_RANK"]))
+    if os.getenv("SLURM_NPROCS") is not None:
+        return (int(os.environ["SLURM_NPROCS"]),
+                int(os.environ["SLURM_PROCID"]))
+    return 1, 0
+
+
+def get_local_rank() -> int:
+    for var in ("LOCAL_RANK", "OMPI_COMM_WORLD_LOCAL_RANK",
+                "SLURM_LOCALID"):
+        if os.getenv(var) is not None:
+            return int(os.environ[var])
+    return 0
+
+
+def _select_backend() -> str:
+    backend = os.getenv("HYDRAGNN_BACKEND")
+    if backend:
+        return backend
+    if torch.cuda.is_available() and dist.is_nccl_available():
+        return "nccl"  # RCCL on ROCm
+    return "gloo"
+
+
+def setup_ddp(use_deepspeed: bool = False) -> Tuple[int, int]:
+    """Initialize the default process group. Returns (world_size, rank)."""
+    world_size, rank = init_comm_size_and_rank()
+    if dist.is_initialized():
+        return world_size, rank
+
+    master_addr = (os.getenv("HYDRAGNN_MASTER_ADDR")
+                   or os.getenv("MASTER_ADDR") or "127.0.0.1")
+    master_port = (os.getenv("HYDRAGNN_MASTER_PORT")
+                   or os.getenv("MASTER_PORT") or "8889")
+    os.environ["MASTER_ADDR"] = master_addr
+    os.environ["MASTER_PORT"] = str(master_port)
+    os.environ.setdefault("RANK", str(rank))
+    os.environ.setdefault("WORLD_SIZE", str(world_size))
+
+    backend = _select_backend()
+    retries = int(os.getenv("HYDRAGNN_MASTER_PORT_RETRIES", "10"))
+    port = int(master_port)
+    last_err = None
+    for _ in range(max(retries, 1)):
+        try:
+            os.environ["MASTER_PORT"] = str(port)
+            dist.init_process_group(
+                backend=backend, rank=rank, world_size=world_size,
+                timeout=datetime.timedelta(seconds=1800))
+            break
+        except (RuntimeError, OSError) as e:  # EADDRINUSE retry
+            last_err = e
+            if "address already in use" in str(e).lower() and world_size == 1:
+                port += 1
+                continue
+            raise
+    else:
+        raise RuntimeError(f"setup_ddp failed: {last_err}")
+
+    if torch.cuda.is_available():
+        torch.cuda.set_device(get_local_rank() % torch.cuda.device_count())
+    return world_size, rank
+
+
+def get_comm_size_and_rank() -> Tuple[int, int]:
+    if dist.is_initialized():
+        return dist.get_world_size(), dist.get_rank()
+    return 1, 0
+
+
+def get_device_name(use_gpu: bool = True) -> str:
+    if use_gpu and torch.cuda.is_available():
+        return f"cuda:{get_local_rank() % torch.cuda.device_count()}"
+    return "cpu"
+
+
+def get_device(use_gpu: bool = True, rank_per_model: int = 1,
+               verbosity_level: int = 0) -> torch.device:
+    return torch.device(get_device_name(use_gpu))
+
+
+def nsplit(lst, n: int):
+    """Split a sequence into n roughly-equal chunks (per-rank sharding,
+    reference abstractrawdataset.py:172)."""
+    k, m = divmod(len(lst), n)
+    return (lst[i * k + min(i, m):(i + 1) * k + min(i + 1, m)]
+            for i in range(n))
+
+
+def comm_reduce(value: torch.Tensor, op: str = "sum") -> torch.Tensor:
+    if not dist.is_initialized():
+        return value
+    ops = {"sum": dist.ReduceOp.SUM, "max": dist.ReduceOp.MAX,
+           "min": dist.ReduceOp.MIN}
+    dist.all_reduce(value, op=ops[op])
+    return value
+
+
+# ---------------------------------------------------------------------------
+# model wrappers: DDP / FSDP1 / FSDP2
+# ---------------------------------------------------------------------------
+class ModuleCompat(torch.nn.Module):
+    """.module shim so FSDP2-composable models look like DDP to the
+    training loop (reference distributed.py:37)."""
+
+    def __init__(self, model):
+        super().__init__()
+        self.module = model
+
+    def forward(self, *args, **kwargs):
+        return self.module(*args, **kwargs)
+
+    def __getattr__(self, name):
+        try:
+            return super().__getattr__(name)
+        except AttributeError:
+            return getattr(self.module, name)
+
+
+def get_distributed_model(model, verbosity: int = 0,
+                          find_unused_parameters: bool = False,
+                          process_group=None):
+    device = get_device()
+    world_size, _ = get_comm_size_and_rank()
+    if world_size <= 1 and not dist.is_initialized():
+        return model
+
+    use_fsdp = bool(int(os.getenv("HYDRAGNN_USE_FSDP", "0")))
+    if use_fsdp:
+        version = int(os.getenv("HYDRAGNN_FSDP_VERSION", "2"))
+        if version == 1:
+            from torch.distributed.fsdp import (
+                FullyShardedDataParallel as FSDP, ShardingStrategy)
+            strategy_name = os.getenv("HYDRAGNN_FSDP_STRATEGY", "FULL_SHARD")
+            strategy = getattr(ShardingStrategy, strategy_name)
+            return FSDP(model, sharding_strategy=strategy,
+                        device_id=device if device.type == "cuda" else None,
+                        process_group=process_group)
+        # FSDP v2 (composable fully_shard)
+        from torch.distributed.fsdp import fully_shard
+        reshard = os.getenv("HYDRAGNN_FSDP_RESHARD", "default")
+        kwargs = {}
+        if reshard in ("true", "1"):
+            kwargs["reshard_after_forward"] = True
+        elif reshard in ("false", "0"):
+            kwargs["reshard_after_forward"] = False
+        for submodule in model.graph_convs if hasattr(model, "graph_convs") \
+                else []:
+            fully_shard(submodule, **kwargs)
+        fully_shard(model, **kwargs)
+        return ModuleCompat(model)
+
+    ddp_kwargs = dict(find_unused_parameters=find_unused_parameters,
+                      process_group=process_group)
+    if device.type == "cuda":
+        ddp_kwargs["device_ids"] = [device]
+    return torch.nn.parallel.DistributedDataParallel(model, **ddp_kwargs)
+
+
+def distributed_model_wrapper(model, max_neighbours=None, verbosity: int = 0,
+                              find_unused_parameters: bool = False):
+    """Reference distributed.py:489: move to device and wrap."""
+    device = get_device()
+    model = model.to(device)
+    if dist.is_initialized() and dist.get_world_size() >= 1:
+        # EnhancedModelWrapper's second autograd pass can leave params
+        # unused in the first; mirror the reference's toggle.
+        from ...models.create import EnhancedModelWrapper
+        if isinstance(model, EnhancedModelWrapper):
+            find_unused_parameters = True
+        model = get_distributed_model(
+            model, verbosity, find_unused_parameters=find_unused_parameters)
+    return model
+
+
+def print_peak_memory(verbosity: int = 0, prefix: str = "") -> None:
+    if torch.cuda.is_available():
+        peak = torch.cuda.max_memory_allocated() / (1024 ** 3)
+        print(f"{prefix} peak GPU memory: {peak:.2f} GB")
+
+
+def check_remaining_time(start_time, epoch_time, broadcast: bool = True
+                         ) -> bool:
+    """SLURM remaining-time early stop (reference distributed.py:619).
+    Returns True if training should stop."""
+    import subprocess
+    import time
+    should_stop = torch.zeros(1, dtype=torch.uint8)
+    _, rank = get_comm_size_and_rank()
+    if rank == 0 and os.getenv("SLURM_JOB_ID"):
+        try:
+            out = subprocess.run(
+                ["squeue", "-h", "-j", os.environ["SLURM_JOB_ID"], "-o",
+                 "%L"], capture_output=True, text=True, timeout=10).stdout
+            parts = out.strip().split(":")
+            secs = 0
+            for p in parts:
+                if "-" in p:
+                    d, h = p.split("-")
+                    secs = secs * 60 + int(d) * 86400 + int(h) * 3600
+                else:
+                    secs = secs * 60 + int(p)
+            if secs < 1.5 * epoch_time:
+                should_stop[0] = 1
+        except Exception:
+            pass
+    if broadcast and dist.is_initialized():
+        dist.broadcast(should_stop, src=0)
+    return bool(should_stop.item())

@@ -1,0 +1,125 @@
+"""Checkpoint save/load, EarlyStopping, Checkpoint gating, tensorboard
+writer.
+
+On-disk format compatibility with the reference (SURVEY.md §5
+"Checkpoint / resume"): single .pk file torch.save dict with
+model_state_dict / optimizer_state_dict keys, per-epoch filenames +
+latest symlink under logs/<name>/ (reference: hydragnn/utils/model/
+model.py:106-313, 515-573)."""
+
+from __future__ import annotations
+
+import os
+from typing import Optional
+
+import torch
+import torch.distributed as dist
+
+
+def _rank() -> int:
+    return dist.get_rank() if dist.is_initialized() else 0
+
+
+def get_summary_writer(name: str, path: str = "./logs/"):
+    if _rank() != 0:
+        return None
+    try:
+        from torch.utils.tensorboard import SummaryWriter
+        return SummaryWriter(os.path.join(path, name))
+    except ImportError:
+        return None
+
+
+def _unwrap(model):
+    return model.module if hasattr(model, "module") else model
+
+
+def save_model(model, optimizer, name: str, epoch: Optional[int] = None,
+               path: str = "./logs/") -> None:
+    """torch.save {model_state_dict, optimizer_state_dict} to
+    logs/<name>/<name>[_epoch_E].pk + latest symlink."""
+    d = os.path.join(path, name)
+    os.makedirs(d, exist_ok=True)
+    if hasattr(optimizer, "consolidate_state_dict"):
+        optimizer.consolidate_state_dict()
+    m = _unwrap(model)
+    if _rank() == 0:
+        fname = (f"{name}_epoch_{epoch}.pk" if epoch is not None
+                 else f"{name}.pk")
+        fpath = os.path.join(d, fname)
+        torch.save({
+            "model_state_dict": m.state_dict(),
+            "optimizer_state_dict": optimizer.state_dict(),
+        }, fpath)
+        latest = os.path.join(d, f"{name}.pk")
+        if epoch is not None and fpath != latest:
+            if os.path.islink(latest) or os.path.exists(latest):
+                os.remove(latest)
+            os.symlink(fname, latest)
+    if dist.is_initialized():
+        dist.barrier()
+
+
+def load_existing_model(model, name: str, path: str = "./logs/",
+                        optimizer=None, map_location=None) -> None:
+    fpath = os.path.join(path, name, f"{name}.pk")
+    if map_location is None:
+        map_location = "cpu"
+    ckpt = torch.load(fpath, map_location=map_location, weights_only=False)
+    state = ckpt["model_state_dict"]
+    m = _unwrap(model)
+    # module-prefix fixup (checkpoints saved from DDP-wrapped models)
+    if any(k.startswith("module.") for k in state):
+        state = {k.removeprefix("module."): v for k, v in state.items()}
+    m.load_state_dict(state)
+    if optimizer is not None and "optimizer_state_dict" in ckpt:
+        optimizer.load_state_dict(ckpt["optimizer_state_dict"])
+
+
+def load_existing_model_config(model, config, path: str = "./logs/",
+                               optimizer=None) -> None:
+    if config.get("continue", 0):
+        name = config.get("startfrom")
+        if name:
+            load_existing_model(model, name, path, optimizer)
+
+
+class EarlyStopping:
+    """Stop when validation loss has not improved for `patience` epochs
+    (reference model.py:515-530)."""
+
+    def __init__(self, patience: int = 10, min_delta: float = 0.0):
+        self.patience = patience
+        self.min_delta = min_delta
+        self.counter = 0
+        self.best = None
+        self.early_stop = False
+
+    def __call__(self, val_loss: float) -> bool:
+        if self.best is None or val_loss < self.best - self.min_delta:
+            self.best = val_loss
+            self.counter = 0
+        else:
+            self.counter += 1
+            if self.counter >= self.patience:
+                self.early_stop = True
+        return self.early_stop
+
+
+class Checkpoint:
+    """Best-val-metric gated checkpointing with warmup
+    (reference model.py:533-573)."""
+
+    def __init__(self, name: str, warmup: int = 0, path: str = "./logs/"):
+        self.name = name
+        self.warmup = warmup
+        self.path = path
+        self.best = None
+
+    def __call__(self, epoch: int, val_loss: float) -> bool:
+        if epoch < self.warmup:
+            return False
+        if self.best is None or val_loss < self.best:
+            self.best = val_loss
+            return True
+        return False
