@@ -171,6 +171,65 @@ __global__ void scatter_max_pass2(const float* __restrict__ src,
   }
 }
 
+// ---- fp64 variants (order-preserving flip on 64-bit patterns) ----
+__device__ inline unsigned long long double_flip(double f) {
+  unsigned long long u = __double_as_longlong(f);
+  return (u & 0x8000000000000000ull) ? ~u : (u | 0x8000000000000000ull);
+}
+__device__ inline double double_unflip(unsigned long long u) {
+  return __longlong_as_double(
+      (u & 0x8000000000000000ull) ? (u & 0x7fffffffffffffffull) : ~u);
+}
+
+__global__ void scatter_max_pass1_f64(const double* __restrict__ src,
+                                      const long* __restrict__ index,
+                                      unsigned long long* __restrict__ out_bits,
+                                      long E, long F, bool is_max) {
+  long total = E * F;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x) {
+    long e = i / F;
+    long f = i - e * F;
+    unsigned long long bits = double_flip(src[i]);
+    unsigned long long* slot = &out_bits[index[e] * F + f];
+    if (is_max)
+      atomicMax(slot, bits);
+    else
+      atomicMin(slot, bits);
+  }
+}
+
+__global__ void scatter_max_pass2_f64(const double* __restrict__ src,
+                                      const long* __restrict__ index,
+                                      const unsigned long long* __restrict__ out_bits,
+                                      unsigned long long* __restrict__ arg,
+                                      long E, long F) {
+  long total = E * F;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x) {
+    long e = i / F;
+    long f = i - e * F;
+    long slot = index[e] * F + f;
+    if (double_flip(src[i]) == out_bits[slot]) {
+      atomicMin(&arg[slot], (unsigned long long)e);
+    }
+  }
+}
+
+__global__ void scatter_max_finalize_f64(
+    const unsigned long long* __restrict__ bits,
+    const unsigned long long* __restrict__ arg, double* __restrict__ out,
+    long* __restrict__ arg_out, long total, bool is_max) {
+  unsigned long long empty =
+      is_max ? double_flip(-INFINITY) : double_flip(INFINITY);
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x) {
+    bool present = bits[i] != empty;
+    out[i] = present ? double_unflip(bits[i]) : 0.0;
+    arg_out[i] = present ? (long)arg[i] : -1;
+  }
+}
+
 __global__ void scatter_max_finalize(const unsigned int* __restrict__ bits,
                                      const unsigned long long* __restrict__ arg,
                                      float* __restrict__ out,
@@ -348,10 +407,60 @@ std::vector<torch::Tensor> scatter_mean_fwd(torch::Tensor src,
   return {out, count};
 }
 
+std::vector<torch::Tensor> scatter_minmax_fwd_f64(torch::Tensor src,
+                                                  torch::Tensor index,
+                                                  long dim_size,
+                                                  bool is_max) {
+  auto srcd = src.contiguous();
+  long E = index.numel();
+  long F = E > 0 ? srcd.numel() / srcd.size(0) : 1;
+  auto sizes = src.sizes().vec();
+  sizes[0] = dim_size;
+  long total = dim_size * F;
+  auto idx = index.contiguous();
+  auto bits = torch::empty(sizes, src.options().dtype(torch::kLong));
+  {
+    double fill = is_max ? -INFINITY : INFINITY;
+    unsigned long long raw;
+    memcpy(&raw, &fill, 8);
+    unsigned long long u = (raw & 0x8000000000000000ull) ? ~raw
+                           : (raw | 0x8000000000000000ull);
+    bits.fill_((long)u);
+  }
+  auto arg64 = torch::full(sizes, (long)0x7fffffffffffffffLL,
+                           src.options().dtype(torch::kLong));
+  auto out = torch::empty(sizes, src.options());
+  auto arg = torch::empty(sizes, src.options().dtype(torch::kLong));
+  if (E > 0) {
+    hipLaunchKernelGGL(scatter_max_pass1_f64, dim3(n_blocks(E * F, kBlock)),
+                       dim3(kBlock), 0, cur_stream(), srcd.data_ptr<double>(),
+                       idx.data_ptr<long>(),
+                       reinterpret_cast<unsigned long long*>(bits.data_ptr<long>()),
+                       E, F, is_max);
+    hipLaunchKernelGGL(scatter_max_pass2_f64, dim3(n_blocks(E * F, kBlock)),
+                       dim3(kBlock), 0, cur_stream(), srcd.data_ptr<double>(),
+                       idx.data_ptr<long>(),
+                       reinterpret_cast<unsigned long long*>(bits.data_ptr<long>()),
+                       reinterpret_cast<unsigned long long*>(arg64.data_ptr<long>()),
+                       E, F);
+  }
+  hipLaunchKernelGGL(scatter_max_finalize_f64,
+                     dim3(n_blocks(total, kBlock)), dim3(kBlock), 0,
+                     cur_stream(),
+                     reinterpret_cast<unsigned long long*>(bits.data_ptr<long>()),
+                     reinterpret_cast<unsigned long long*>(arg64.data_ptr<long>()),
+                     out.data_ptr<double>(), arg.data_ptr<long>(), total,
+                     is_max);
+  return {out, arg};
+}
+
 std::vector<torch::Tensor> scatter_minmax_fwd(torch::Tensor src,
                                               torch::Tensor index,
                                               long dim_size, bool is_max) {
   CHECK_CUDA(src); CHECK_CUDA(index);
+  if (src.scalar_type() == at::ScalarType::Double) {
+    return scatter_minmax_fwd_f64(src, index, dim_size, is_max);
+  }
   auto srcf = src.contiguous().to(torch::kFloat);
   long E = index.numel();
   long F = E > 0 ? srcf.numel() / srcf.size(0) : 1;
