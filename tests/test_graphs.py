@@ -23,7 +23,8 @@ THRESHOLDS = {
     "MACE": 0.70,
 }
 
-SIMPLE_MODELS = ["GIN", "SAGE", "MFC", "GAT", "CGCNN", "PNA"]
+SIMPLE_MODELS = ["GIN", "SAGE", "MFC", "GAT", "CGCNN", "PNA", "PNAPlus",
+                 "SchNet", "EGNN", "DimeNet", "PAINN", "PNAEq"]
 
 
 @pytest.mark.parametrize("mpnn_type", SIMPLE_MODELS)
