@@ -1,0 +1,174 @@
+"""MACE stack: O(3)-equivariant higher-order message passing.
+
+Functional parity with /root/reference/hydragnn/models/MACEStack.py:
+74-605: per-graph position centering, one-hot(Z,118) node attrs,
+spherical-harmonic edge attrs, Bessel x polynomial-cutoff radial
+embedding, per-layer interaction (edge tensor product) + product basis
+(symmetric contraction) + per-layer multihead readouts summed over
+layers; the last layer produces scalars only.
+"""
+
+from __future__ import annotations
+
+from typing import List, Optional
+
+import torch
+from torch import nn
+
+from ...ops import (
+    get_edge_vectors_and_lengths,
+    scatter,
+    spherical_harmonics,
+)
+from ..base import Base
+from .blocks import (
+    EquivariantProductBasisBlock,
+    LinearReadoutBlock,
+    NonLinearReadoutBlock,
+    RadialEmbeddingBlock,
+    RealAgnosticResidualInteractionBlock,
+)
+from .o3 import IrrepsLinear, dim
+
+NUM_ELEMENTS = 118
+
+
+class MACEStack(Base):
+    def __init__(
+        self,
+        r_max: Optional[float] = None,
+        radial_type: Optional[str] = "bessel",
+        distance_transform: Optional[str] = None,
+        num_bessel: Optional[int] = 8,
+        max_ell: Optional[int] = 2,
+        node_max_ell: Optional[int] = 1,
+        avg_num_neighbors: Optional[float] = 10.0,
+        envelope_exponent: Optional[int] = 5,
+        correlation: Optional[int] = 2,
+        edge_dim: Optional[int] = None,
+        **kwargs,
+    ):
+        self.r_max = r_max or 5.0
+        self.radial_type = radial_type or "bessel"
+        self.distance_transform = distance_transform
+        self.num_bessel = num_bessel or 8
+        self.max_ell = max_ell if max_ell is not None else 2
+        self.node_max_ell = node_max_ell if node_max_ell is not None else 1
+        self.avg_num_neighbors = avg_num_neighbors or 10.0
+        self.envelope_exponent = envelope_exponent or 5
+        self.correlation = correlation if isinstance(correlation, int) \
+            else (correlation[0] if correlation else 2)
+        self.is_edge_model = True
+        super().__init__(edge_dim=edge_dim, **kwargs)
+
+    # ------------------------------------------------------------------
+    def _init_conv(self):
+        C = self.hidden_dim
+        self.node_embedding = nn.Linear(NUM_ELEMENTS, C, bias=False)
+        self.radial_embedding = RadialEmbeddingBlock(
+            self.r_max, self.num_bessel, self.envelope_exponent,
+            self.radial_type, self.distance_transform)
+        self.interactions = nn.ModuleList()
+        self.products = nn.ModuleList()
+        lmax_node = 0
+        for ilayer in range(self.num_conv_layers):
+            last = ilayer == self.num_conv_layers - 1
+            lmax_out = 0 if last else self.node_max_ell
+            # interaction always emits the full hidden tower; the
+            # product basis contracts to the layer's target (scalars
+            # only on the last layer — MACE convention)
+            self.interactions.append(RealAgnosticResidualInteractionBlock(
+                C, lmax_node, self.max_ell, self.node_max_ell,
+                self.radial_embedding.out_dim, self.avg_num_neighbors))
+            self.products.append(EquivariantProductBasisBlock(
+                C, self.node_max_ell, lmax_out,
+                self.correlation, NUM_ELEMENTS))
+            lmax_node = lmax_out
+
+    def _multihead(self):
+        """Per-layer readouts per head, summed over layers
+        (reference MACEStack.get_multihead_decoder, :573)."""
+        C = self.hidden_dim
+        self.graph_shared = nn.ModuleDict({})  # unused; kept for API
+        self.num_branches = 1
+        self.readouts = nn.ModuleList()
+        for ihead in range(self.num_heads):
+            per_layer = nn.ModuleList()
+            out_dim = self.head_dims[ihead] * (1 + self.var_output)
+            for ilayer in range(self.num_conv_layers):
+                last = ilayer == self.num_conv_layers - 1
+                if last:
+                    per_layer.append(NonLinearReadoutBlock(
+                        C, max(C // 2, out_dim), out_dim,
+                        self.activation_function))
+                else:
+                    per_layer.append(LinearReadoutBlock(C, out_dim))
+            self.readouts.append(per_layer)
+
+    # ------------------------------------------------------------------
+    def _node_elements(self, data) -> torch.Tensor:
+        z = data.get("z")
+        if z is None:
+            z = data.x[:, 0].long()
+        return z.clamp(min=0, max=NUM_ELEMENTS - 1).long()
+
+    def _embedding(self, data):
+        pos = data.pos
+        batch = data.get("batch")
+        if batch is None:
+            batch = torch.zeros(pos.shape[0], dtype=torch.long,
+                                device=pos.device)
+            data["batch"] = batch
+        # center positions per graph (keeps the autograd force path:
+        # centering is translation-invariant so forces are unaffected)
+        n_graphs = int(batch.max()) + 1
+        mean_pos = scatter(pos, batch, n_graphs, "mean")
+        pos_c = pos - mean_pos[batch]
+        vec, lengths = get_edge_vectors_and_lengths(
+            pos_c, data.edge_index, data.get("edge_shifts"))
+        edge_sh = spherical_harmonics(vec, self.max_ell, normalize=True)
+        edge_radial = self.radial_embedding(lengths)
+        elem = self._node_elements(data)
+        one_hot = torch.nn.functional.one_hot(
+            elem, NUM_ELEMENTS).to(self.node_embedding.weight.dtype)
+        h0 = self.node_embedding(one_hot)  # [N, C]
+        return h0, elem, edge_sh.to(h0.dtype), edge_radial.to(h0.dtype)
+
+    def forward(self, data):
+        h0, elem, edge_sh, edge_radial = self._embedding(data)
+        batch = data["batch"]
+        n = h0.shape[0]
+        C = self.hidden_dim
+        node_feats = h0.view(n, C, 1)
+        n_graphs = int(batch.max()) + 1
+
+        head_outputs = [None] * self.num_heads
+        for ilayer, (inter, prod) in enumerate(
+                zip(self.interactions, self.products)):
+            # pad features to the interaction's input tower
+            want = dim(inter.lmax_node)
+            if node_feats.shape[-1] < want:
+                node_feats = torch.nn.functional.pad(
+                    node_feats, (0, want - node_feats.shape[-1]))
+            m = inter(node_feats, data.edge_index, edge_sh, edge_radial)
+            node_feats = prod(m, elem, sc=m)
+            for ihead in range(self.num_heads):
+                r = self.readouts[ihead][ilayer](node_feats)
+                head_outputs[ihead] = r if head_outputs[ihead] is None \
+                    else head_outputs[ihead] + r
+
+        outputs = []
+        outputs_var = []
+        for ihead in range(self.num_heads):
+            out = head_outputs[ihead]
+            hd = self.head_dims[ihead]
+            if self.head_type[ihead] == "graph":
+                out = self.pool_fn(out, batch, n_graphs)
+            outputs.append(out[:, :hd])
+            outputs_var.append(out[:, hd:] ** 2 if self.var_output else None)
+        if self.var_output:
+            return outputs, outputs_var
+        return outputs
+
+    def __str__(self):
+        return "MACEStack"

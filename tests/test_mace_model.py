@@ -1,0 +1,106 @@
+"""MACE model-level checks: rotational invariance of predictions,
+equivariance of forces, force training (pattern: reference
+tests/test_equivariant_mace_integration.py / test_forces_equivariant*)."""
+
+import numpy as np
+import pytest
+import torch
+
+from hydragnn_amd.data import Batch
+from hydragnn_amd.models import create_model_config
+from hydragnn_amd.ops import radius_graph, scatter
+from hydragnn_amd.preprocess import create_dataloaders
+from hydragnn_amd.utils.config import update_config
+from hydragnn_amd.utils.datasets.synthetic import md17_shape_dataset
+
+from deterministic_graph_data import base_config
+
+
+def _mace_config(num_epoch=2, node_head=True):
+    config = base_config("MACE", heads=("node",) if node_head
+                         else ("graph",), num_epoch=num_epoch,
+                         hidden_dim=16, lr=0.005, batch_size=4)
+    arch = config["NeuralNetwork"]["Architecture"]
+    arch.update({
+        "max_ell": 2, "node_max_ell": 1, "correlation": 2,
+        "num_radial": 8, "radius": 7.0,
+        "enable_interatomic_potential": True,
+        "energy_weight": 1.0, "energy_peratom_weight": 1.0,
+        "force_weight": 10.0,
+    })
+    config["NeuralNetwork"]["Variables_of_interest"]["output_dim"] = [1]
+    return config
+
+
+def _build(config, dataset):
+    loaders = create_dataloaders(dataset, dataset, dataset, 4,
+                                 config=config)
+    config = update_config(config, *loaders)
+    model = create_model_config(config["NeuralNetwork"], use_gpu=False)
+    return model, config, loaders
+
+
+def _rand_rot(seed=0):
+    rng = np.random.default_rng(seed)
+    Q, _ = np.linalg.qr(rng.normal(size=(3, 3)))
+    if np.linalg.det(Q) < 0:
+        Q[:, 0] *= -1
+    return torch.from_numpy(Q).float()
+
+
+def test_mace_energy_rotation_invariant():
+    torch.manual_seed(0)
+    dataset = md17_shape_dataset(num_samples=4)
+    model, config, _ = _build(_mace_config(), dataset)
+    model.eval()
+    batch = Batch.from_data_list(dataset[:2])
+    with torch.no_grad():
+        e1 = model(batch)[0]
+    R = _rand_rot(1)
+    rot = [d.clone() for d in dataset[:2]]
+    for d in rot:
+        d.pos = d.pos @ R.T
+        d.edge_index = radius_graph(d.pos, 7.0, max_num_neighbors=30)
+    batch2 = Batch.from_data_list(rot)
+    with torch.no_grad():
+        e2 = model(batch2)[0]
+    assert torch.allclose(e1, e2, atol=1e-4), (
+        f"energy not rotation-invariant: {(e1 - e2).abs().max():.2e}")
+
+
+def test_mace_forces_rotation_equivariant():
+    torch.manual_seed(0)
+    dataset = md17_shape_dataset(num_samples=2)
+    model, config, _ = _build(_mace_config(), dataset)
+    model.eval()
+
+    def forces_of(data_list):
+        batch = Batch.from_data_list([d.clone() for d in data_list])
+        batch.pos.requires_grad_(True)
+        pred = model(batch)
+        E = scatter(pred[0], batch.batch, batch.num_graphs, "sum").sum()
+        return -torch.autograd.grad(E, batch.pos)[0]
+
+    f1 = forces_of(dataset[:2])
+    R = _rand_rot(2)
+    rot = [d.clone() for d in dataset[:2]]
+    for d in rot:
+        d.pos = (d.pos @ R.T).detach()
+    f2 = forces_of(rot)
+    assert torch.allclose(f1 @ R.T, f2, atol=1e-4), (
+        f"forces not equivariant: {(f1 @ R.T - f2).abs().max():.2e}")
+
+
+def test_mace_force_training_decreases():
+    torch.manual_seed(0)
+    from hydragnn_amd.train import train as train_fn
+    from hydragnn_amd.utils.optimizer import select_optimizer
+    dataset = md17_shape_dataset(num_samples=16)
+    model, config, loaders = _build(_mace_config(num_epoch=4), dataset)
+    opt = select_optimizer(model,
+                           config["NeuralNetwork"]["Training"]["Optimizer"])
+    errs = []
+    for _ in range(4):
+        err, _ = train_fn(loaders[0], model, opt, 0)
+        errs.append(float(err))
+    assert errs[-1] < errs[0], f"MACE force loss not decreasing: {errs}"
