@@ -1,0 +1,227 @@
+"""Flagship benchmark: MACE interatomic potential (energy + forces)
+training throughput on MD17-shape molecules (BASELINE.json configs[1]).
+
+Contract (driver): `python bench.py --gpus N --steps K --warmup W` runs
+W untimed warmup steps then times exactly K steps bracketed by
+barrier + torch.cuda.synchronize on both sides; rank 0 prints ONE JSON
+line with the whole-job aggregate graphs/sec.
+
+Launched for N>1 as: python -m torch.distributed.run --nnodes=1
+--nproc-per-node N --master-addr 127.0.0.1 bench.py --gpus N ...
+(one rank per GPU over RCCL).
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import time
+
+import torch
+import torch.distributed as dist
+
+
+# MACE MD17 headline config: bf16 compute, aspirin-shaped molecules.
+MODEL_CONFIG = {
+    "mpnn_type": "MACE",
+    "radius": 7.0,
+    "max_neighbours": 30,
+    "hidden_dim": 64,          # channels
+    "num_conv_layers": 2,
+    "max_ell": 2,
+    "node_max_ell": 1,
+    "correlation": 2,
+    "num_radial": 8,
+    "radial_type": "bessel",
+    "envelope_exponent": 5,
+    "avg_num_neighbors": 20.0,
+    "enable_interatomic_potential": True,
+    "energy_weight": 1.0,
+    "energy_peratom_weight": 1.0,
+    "force_weight": 100.0,
+    "output_heads": {
+        "node": [{"type": "branch-0",
+                  "architecture": {"num_headlayers": 2,
+                                   "dim_headlayers": [64, 64],
+                                   "type": "mlp"}}],
+    },
+    "task_weights": [1.0],
+    "output_dim": [1],
+    "output_type": ["node"],
+    "input_dim": 1,
+    "num_nodes": 21,
+    "pna_deg": None,
+    "edge_dim": None,
+    "pe_dim": 0,
+    "global_attn_engine": None,
+    "global_attn_type": None,
+    "global_attn_heads": 0,
+    "activation_function": "silu",
+    "freeze_conv_layers": False,
+    "initial_bias": None,
+    "graph_pooling": "mean",
+    "equivariance": True,
+    "basis_emb_size": None, "int_emb_size": None, "out_emb_size": None,
+    "num_gaussians": None, "num_filters": None, "num_before_skip": None,
+    "num_after_skip": None, "num_spherical": None,
+    "distance_transform": None,
+}
+
+LOCAL_BATCH = 32          # graphs per GPU per step (weak scaling)
+PRECISION = "bf16"
+
+
+def build_model_and_batch(device="cuda:0", local_batch=LOCAL_BATCH,
+                          precision=PRECISION, seed=17):
+    """Build the flagship model + one synthetic batch + a step()
+    closure (used by both bench.py and __graft_entry__.smoke)."""
+    from hydragnn_amd.data import Batch
+    from hydragnn_amd.models.create import create_model, resolve_precision
+    from hydragnn_amd.train import get_autocast_and_scaler
+    from hydragnn_amd.utils.datasets.synthetic import md17_shape_dataset
+    from hydragnn_amd.utils.optimizer import select_optimizer
+
+    torch.manual_seed(seed)
+    cfg = dict(MODEL_CONFIG)
+    model = create_model(
+        mpnn_type=cfg["mpnn_type"], input_dim=cfg["input_dim"],
+        hidden_dim=cfg["hidden_dim"], output_dim=cfg["output_dim"],
+        output_type=cfg["output_type"], output_heads=cfg["output_heads"],
+        activation_function=cfg["activation_function"],
+        loss_function_type="mse", task_weights=cfg["task_weights"],
+        num_conv_layers=cfg["num_conv_layers"],
+        num_nodes=cfg["num_nodes"], max_neighbours=cfg["max_neighbours"],
+        radius=cfg["radius"], num_radial=cfg["num_radial"],
+        radial_type=cfg["radial_type"],
+        envelope_exponent=cfg["envelope_exponent"],
+        max_ell=cfg["max_ell"], node_max_ell=cfg["node_max_ell"],
+        correlation=cfg["correlation"],
+        avg_num_neighbors=cfg["avg_num_neighbors"],
+        enable_interatomic_potential=True,
+        energy_weight=cfg["energy_weight"],
+        energy_peratom_weight=cfg["energy_peratom_weight"],
+        force_weight=cfg["force_weight"],
+        use_gpu=False,
+    )
+    _, param_dtype, _ = resolve_precision(precision)
+    model = model.to(device=device, dtype=param_dtype)
+
+    if dist.is_initialized() and dist.get_world_size() > 1:
+        model = torch.nn.parallel.DistributedDataParallel(
+            model, device_ids=[torch.device(device)]
+            if torch.device(device).type == "cuda" else None,
+            find_unused_parameters=True)
+        m = model.module
+    else:
+        m = model
+
+    rank = dist.get_rank() if dist.is_initialized() else 0
+    dataset = md17_shape_dataset(num_samples=local_batch,
+                                 radius=cfg["radius"],
+                                 max_neighbours=cfg["max_neighbours"],
+                                 seed=seed + rank)
+    batch = Batch.from_data_list(dataset).to(device)
+    for key in list(batch.keys()):
+        v = batch[key]
+        if torch.is_tensor(v) and torch.is_floating_point(v):
+            batch[key] = v.to(param_dtype)
+
+    optimizer = select_optimizer(model, {"type": "AdamW",
+                                         "learning_rate": 1e-3})
+    autocast, _ = get_autocast_and_scaler(precision)
+
+    def step():
+        optimizer.zero_grad(set_to_none=True)
+        batch.pos.requires_grad_(True)
+        with autocast:
+            pred = model(batch)
+            loss, _ = m.energy_force_loss(pred, batch, create_graph=True)
+        loss.backward()
+        optimizer.step()
+        return loss
+
+    return model, batch, step
+
+
+def main():
+    parser = argparse.ArgumentParser()
+    parser.add_argument("--gpus", type=int, default=1)
+    parser.add_argument("--steps", type=int, default=20)
+    parser.add_argument("--warmup", type=int, default=5)
+    parser.add_argument("--batch", type=int, default=LOCAL_BATCH)
+    args = parser.parse_args()
+
+    world_size = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    use_cuda = torch.cuda.is_available()
+    if world_size > 1:
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        dist.init_process_group("nccl" if use_cuda else "gloo")
+        if use_cuda:
+            torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", "0")))
+    device = (f"cuda:{int(os.environ.get('LOCAL_RANK', '0'))}"
+              if use_cuda else "cpu")
+
+    model, batch, step = build_model_and_batch(device=device,
+                                               local_batch=args.batch)
+
+    for _ in range(args.warmup):
+        step()
+
+    if dist.is_initialized():
+        dist.barrier()
+    if use_cuda:
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        step()
+    if use_cuda:
+        torch.cuda.synchronize()
+    if dist.is_initialized():
+        dist.barrier()
+    t1 = time.perf_counter()
+
+    elapsed = t1 - t0
+    # MAX over ranks
+    if dist.is_initialized():
+        t = torch.tensor([elapsed], device=device if use_cuda else "cpu")
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    n_gpus = world_size
+    graphs_per_step = args.batch * n_gpus
+    ms_per_step = elapsed / args.steps * 1000.0
+    value = graphs_per_step * args.steps / elapsed
+
+    if rank == 0:
+        print(json.dumps({
+            "metric": "graphs/sec training (MACE, MD17-shape)",
+            "value": value,
+            "unit": "graphs/s",
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": ms_per_step,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": PRECISION,
+            "data": "synthetic",
+            "config": {
+                "model": "MACE-MLIP energy+forces",
+                "global_batch": graphs_per_step,
+                "seq_len": 21,
+                "parallelism": f"dp{n_gpus}",
+                "hidden_dim": MODEL_CONFIG["hidden_dim"],
+                "num_conv_layers": MODEL_CONFIG["num_conv_layers"],
+                "max_ell": MODEL_CONFIG["max_ell"],
+                "correlation": MODEL_CONFIG["correlation"],
+            },
+        }))
+    if dist.is_initialized():
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()
