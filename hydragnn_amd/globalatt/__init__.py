@@ -1,0 +1,1 @@
+from .gps import HydraGPSConv, PerformerAttention, redraw_performer_projections

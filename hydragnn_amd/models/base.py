@@ -147,6 +147,23 @@ class Base(Module):
         self.feature_layers = ModuleList()
         self.heads_NN = ModuleList()
 
+        if self.use_global_attn:
+            # GPS positional/feature encoders (reference Base.py:215-260)
+            self.pos_emb = Linear(max(self.pe_dim, 1), self.hidden_dim)
+            if self.input_dim:
+                self.node_emb = Linear(self.input_dim, self.hidden_dim)
+                self.node_lin = Linear(2 * self.hidden_dim, self.hidden_dim)
+            if self.is_edge_model:
+                self.rel_pos_emb = Linear(max(self.pe_dim, 1),
+                                          self.hidden_dim)
+                if self.use_edge_attr:
+                    self.edge_emb = Linear(edge_dim, self.hidden_dim)
+                    self.edge_lin = Linear(2 * self.hidden_dim,
+                                           self.hidden_dim)
+            self.embed_dim = self.hidden_dim
+            if self.use_edge_attr or self.is_edge_model:
+                self.edge_embed_dim = self.hidden_dim
+
         self._init_conditioning()
         self._init_conv()
         self._multihead()
@@ -162,15 +179,26 @@ class Base(Module):
     def get_conv(self, input_dim, output_dim, edge_dim=None):
         raise NotImplementedError
 
+    def _apply_global_attn(self, conv):
+        """Wrap a local conv in a GPS layer when global attention is on
+        (reference Base.py:264)."""
+        if not self.use_global_attn:
+            return conv
+        from ..globalatt.gps import HydraGPSConv
+        return HydraGPSConv(
+            self.hidden_dim, conv, heads=self.global_attn_heads or 1,
+            dropout=self.dropout,
+            attn_type=(self.global_attn_type or "multihead"))
+
     def _init_conv(self):
-        self.graph_convs.append(
+        self.graph_convs.append(self._apply_global_attn(
             self.get_conv(self.embed_dim, self.hidden_dim,
-                          edge_dim=self.edge_embed_dim))
+                          edge_dim=self.edge_embed_dim)))
         self.feature_layers.append(BatchNormNode(self.hidden_dim))
         for _ in range(self.num_conv_layers - 1):
-            self.graph_convs.append(
+            self.graph_convs.append(self._apply_global_attn(
                 self.get_conv(self.hidden_dim, self.hidden_dim,
-                              edge_dim=self.edge_embed_dim))
+                              edge_dim=self.edge_embed_dim)))
             self.feature_layers.append(BatchNormNode(self.hidden_dim))
 
     def _init_conditioning(self):
@@ -325,6 +353,30 @@ class Base(Module):
                                              torch.bfloat16, torch.float16):
             x = x.float()
         pos = data.get("pos")
+        if self.use_global_attn:
+            conv_args["batch"] = data.get("batch")
+            dt = self.pos_emb.weight.dtype
+            pe = data.get("pe")
+            if pe is None:
+                raise ValueError(
+                    "GPS global attention requires positional encodings "
+                    "(data.pe); add them in preprocessing "
+                    "(hydragnn_amd.preprocess.add_laplacian_pe)")
+            h = self.pos_emb(pe.to(dt))
+            if self.input_dim:
+                h = self.node_lin(
+                    torch.cat([self.node_emb(x.to(dt)), h], dim=1))
+            if self.is_edge_model:
+                rel_pe = data.get("rel_pe")
+                if rel_pe is None:
+                    src, dst = data.edge_index[0], data.edge_index[1]
+                    rel_pe = (pe[dst] - pe[src]).abs()
+                e = self.rel_pos_emb(rel_pe.to(dt))
+                if self.use_edge_attr:
+                    e = self.edge_lin(torch.cat(
+                        [self.edge_emb(conv_args["edge_attr"]), e], dim=1))
+                conv_args["edge_attr"] = e
+            return h, pos, conv_args
         return x, pos, conv_args
 
     def _apply_graph_conditioning(self, x, batch, data):

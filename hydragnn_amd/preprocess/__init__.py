@@ -9,3 +9,4 @@ from .batch_sampler import (
     graph_node_costs,
 )
 from ..ops import radius_graph, radius_graph_pbc
+from .transforms import add_laplacian_pe, add_edge_lengths, normalize_rotation

@@ -176,6 +176,9 @@ def train(loader, model, opt, verbosity, precision="fp32", profiler=None):
     for ibatch, data in enumerate(iterate_tqdm(loader, verbosity)):
         if ibatch >= nbatch:
             break
+        from ..globalatt.gps import redraw_performer_projections
+        redraw_performer_projections(
+            model, int(os.environ.get("HYDRAGNN_ATTN_REDRAW", "1000")))
         tr.start("h2d")
         data = move_batch_to_device(data, param_dtype)
         tr.stop("h2d")
