@@ -93,23 +93,46 @@ class EdgeTensorProduct(nn.Module):
         self.paths = tp_paths(lmax_node, lmax_edge, lmax_out)
         self.lmax_out = lmax_out
         self.num_paths = len(self.paths)
-        for i, (l1, l2, l3) in enumerate(self.paths):
-            self.register_buffer(f"w3j_{i}", wigner_3j(l1, l2, l3).float())
+        self.d1 = dim(lmax_node)
+        self.d2 = dim(lmax_edge)
+        self.d3 = dim(lmax_out)
+        # One combined coupling matrix: rows = (m1, m2) outer pairs,
+        # cols = concatenated per-path m3 blocks.  The per-edge TP then
+        # collapses to outer-product + ONE GEMM + masked column
+        # reduction — batched-GEMM-shaped for MFMA instead of a kernel
+        # per path.
+        cols = sum(2 * l3 + 1 for (_, _, l3) in self.paths)
+        Wcat = torch.zeros(self.d1 * self.d2, cols)
+        path_of_col = torch.zeros(cols, dtype=torch.long)
+        m3_of_col = torch.zeros(cols, dtype=torch.long)
+        c0 = 0
+        for p, (l1, l2, l3) in enumerate(self.paths):
+            W = wigner_3j(l1, l2, l3).float()  # [2l1+1, 2l2+1, 2l3+1]
+            for a in range(2 * l1 + 1):
+                for b in range(2 * l2 + 1):
+                    row = (l1 * l1 + a) * self.d2 + (l2 * l2 + b)
+                    Wcat[row, c0:c0 + 2 * l3 + 1] = W[a, b]
+            path_of_col[c0:c0 + 2 * l3 + 1] = p
+            m3_of_col[c0:c0 + 2 * l3 + 1] = torch.arange(
+                l3 * l3, (l3 + 1) ** 2)
+            c0 += 2 * l3 + 1
+        self.register_buffer("Wcat", Wcat)
+        self.register_buffer("path_of_col", path_of_col)
+        self.register_buffer("m3_of_col", m3_of_col)
+        self.num_cols = cols
 
     def forward(self, x_src: torch.Tensor, Y: torch.Tensor,
                 weights: torch.Tensor) -> torch.Tensor:
         """x_src [E, C, D_node], Y [E, D_edge],
         weights [E, C, num_paths] -> [E, C, D_out]."""
         E, C, _ = x_src.shape
-        out = x_src.new_zeros(E, C, dim(self.lmax_out))
-        for i, (l1, l2, l3) in enumerate(self.paths):
-            W = getattr(self, f"w3j_{i}").to(x_src.dtype)
-            s1 = slice(l1 * l1, (l1 + 1) ** 2)
-            s2 = slice(l2 * l2, (l2 + 1) ** 2)
-            s3 = slice(l3 * l3, (l3 + 1) ** 2)
-            term = torch.einsum("abm,eca,eb->ecm", W, x_src[:, :, s1],
-                                Y[:, s2])
-            out[:, :, s3] += term * weights[:, :, i:i + 1]
+        z = (x_src.unsqueeze(-1) * Y.view(E, 1, 1, self.d2)).reshape(
+            E, C, self.d1 * self.d2)
+        raw = z @ self.Wcat.to(x_src.dtype)           # [E, C, cols]
+        w_exp = weights.index_select(2, self.path_of_col)
+        scaled = raw * w_exp
+        out = x_src.new_zeros(E, C, self.d3)
+        out.index_add_(2, self.m3_of_col, scaled)
         return out
 
 

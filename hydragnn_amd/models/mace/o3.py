@@ -119,21 +119,20 @@ class IrrepsLinear(nn.Module):
         self.weight = nn.Parameter(
             torch.randn(lmax + 1, c_in, c_out) / math.sqrt(c_in))
         self.bias = nn.Parameter(torch.zeros(c_out)) if bias else None
+        # per-m l index so the whole map is ONE bmm over D slices
+        lmap = torch.cat([torch.full((2 * l + 1,), l, dtype=torch.long)
+                          for l in range(lmax + 1)])
+        self.register_buffer("lmap", lmap)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        # x [N, C_in, D]
-        outs = []
-        for l in range(self.lmax + 1):
-            sl = slice(l * l, (l + 1) * (l + 1))
-            # [N, C_in, 2l+1] x [C_in, C_out] -> [N, C_out, 2l+1]
-            w = self.weight[l].to(x.dtype)
-            outs.append(torch.einsum("ncm,co->nom", x[:, :, sl], w))
-        out = torch.cat(outs, dim=-1)
+        # x [N, C_in, D] -> one batched GEMM [D](N,C_in)@(C_in,C_out)
+        W_m = self.weight.to(x.dtype)[self.lmap]  # [D, C_in, C_out]
+        out = torch.bmm(x.permute(2, 0, 1), W_m).permute(1, 2, 0)
         if self.bias is not None:
             out = torch.cat([
                 out[:, :, :1] + self.bias.to(x.dtype).view(1, -1, 1),
                 out[:, :, 1:]], dim=-1)
-        return out
+        return out.contiguous()
 
 
 def tp_paths(l_in_max: int, l_edge_max: int, l_out_max: int
