@@ -18,7 +18,9 @@ import torch
 import torch.distributed as dist
 
 from ..models.create import resolve_precision
-from ..utils.distributed import check_remaining_time, get_device
+from ..utils.distributed import (check_remaining_time, get_device,
+                                 is_fsdp2_enabled,
+                                 set_reshard_after_backward)
 from ..utils.model.model import Checkpoint, EarlyStopping, save_model
 from ..utils.print.print_utils import iterate_tqdm, log, print_distributed
 from ..utils.profiling_and_tracing import tracer as tr
@@ -172,6 +174,9 @@ def train(loader, model, opt, verbosity, precision="fp32", profiler=None):
     autocast, scaler = get_autocast_and_scaler(precision)
     nbatch = get_nbatch(loader)
     use_ia = _use_interatomic(m)
+    if use_ia and is_fsdp2_enabled():
+        # FSDP2 double-backward workaround (see set_reshard_after_backward)
+        set_reshard_after_backward(model, False)
 
     for ibatch, data in enumerate(iterate_tqdm(loader, verbosity)):
         if ibatch >= nbatch:
@@ -194,6 +199,9 @@ def train(loader, model, opt, verbosity, precision="fp32", profiler=None):
         tr.start("opt_step")
         opt.step()
         tr.stop("opt_step")
+        if use_ia and is_fsdp2_enabled():
+            set_reshard_after_backward(model, True)
+            set_reshard_after_backward(model, False)
         profiler.step()
         n = data.num_graphs
         total_error += loss.detach() * n

@@ -188,6 +188,31 @@ def get_distributed_model(model, verbosity: int = 0,
     return torch.nn.parallel.DistributedDataParallel(model, **ddp_kwargs)
 
 
+def is_fsdp2_enabled() -> bool:
+    return (bool(int(os.getenv("HYDRAGNN_USE_FSDP", "0")))
+            and int(os.getenv("HYDRAGNN_FSDP_VERSION", "2")) == 2)
+
+
+def set_reshard_after_backward(model, enabled: bool) -> bool:
+    """FSDP2 + double-backward force-training workaround (reference
+    train_validate_test.py:150-169): the force pass autograd.grad(E,
+    pos, create_graph=True) plus the loss backward traverse the graph
+    twice; resharding after the first pass leaves empty parameter
+    storage for the second.  Disable resharding around such steps."""
+    target = model.module if hasattr(model, "module") else model
+    done = False
+    setter = getattr(target, "set_reshard_after_backward", None)
+    if callable(setter):
+        setter(enabled)
+        done = True
+    for sub in getattr(target, "graph_convs", []) or []:
+        s = getattr(sub, "set_reshard_after_backward", None)
+        if callable(s):
+            s(enabled)
+            done = True
+    return done
+
+
 def distributed_model_wrapper(model, max_neighbours=None, verbosity: int = 0,
                               find_unused_parameters: bool = False):
     """Reference distributed.py:489: move to device and wrap."""
