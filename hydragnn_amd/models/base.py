@@ -425,7 +425,9 @@ class Base(Module):
         inv = self.activation_function(feat_layer(inv))
         return inv, equiv
 
-    def forward(self, data):
+    def encoder_forward(self, data):
+        """Embedding + conv stack + pooling (the shared encoder half
+        used by MultiTaskModelMP)."""
         inv_node_feat, equiv_node_feat, conv_args = self._embedding(data)
         batch = data.get("batch")
 
@@ -441,7 +443,10 @@ class Base(Module):
         n_graphs = int(batch.max()) + 1 if batch.numel() > 0 else 1
         x_graph = self.pool_fn(x, batch, n_graphs)
         x_graph = self._apply_graph_pool_conditioning(x_graph, data)
+        return x, x_graph, equiv_node_feat, conv_args
 
+    def forward(self, data):
+        x, x_graph, equiv_node_feat, conv_args = self.encoder_forward(data)
         return self._decode(x, x_graph, equiv_node_feat, data, conv_args)
 
     def _decode(self, x, x_graph, equiv_node_feat, data, conv_args):

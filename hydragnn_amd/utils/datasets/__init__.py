@@ -1,2 +1,6 @@
 from .abstractbasedataset import AbstractBaseDataset, dataset_name_to_id
 from .pickledataset import SimplePickleDataset, SimplePickleWriter
+from .abstractrawdataset import AbstractRawDataset
+from .rawloaders import LSMSDataset, XYZDataset, CFGDataset
+from .graphstore import GraphStoreWriter, GraphStoreDataset, DistDataset
+from .serializeddataset import SerializedDataset, SerializedWriter

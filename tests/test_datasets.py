@@ -1,0 +1,207 @@
+"""Dataset layer: graph store roundtrip, pickle roundtrip, raw-format
+parsers, cost-aware samplers (pattern: reference
+tests/test_datasetclass_inheritance.py, test_cost_aware_batch_sampler)."""
+
+import os
+
+import numpy as np
+import pytest
+import torch
+
+from hydragnn_amd.data import Batch, Data
+from hydragnn_amd.preprocess import (
+    CostAwareBatchSampler,
+    DistributedCostAwareBatchSampler,
+)
+from hydragnn_amd.utils.datasets.graphstore import (
+    DistDataset,
+    GraphStoreDataset,
+    GraphStoreWriter,
+)
+from hydragnn_amd.utils.datasets.pickledataset import (
+    SimplePickleDataset,
+    SimplePickleWriter,
+)
+from hydragnn_amd.utils.datasets.synthetic import lj_dataset
+
+
+def _samples(n=6):
+    return lj_dataset(num_samples=n, num_atoms=27, pbc=False)
+
+
+def test_graphstore_roundtrip(tmp_path):
+    ds = _samples()
+    w = GraphStoreWriter("total", str(tmp_path))
+    w.add(ds)
+    w.add_global("pna_deg", [0, 1, 2, 3])
+    w.add_global("minmax", np.array([[0.0], [1.0]]))
+    w.save()
+
+    r = GraphStoreDataset(str(tmp_path), "total")
+    assert len(r) == len(ds)
+    assert r.pna_deg == [0, 1, 2, 3]
+    for i in (0, len(ds) - 1):
+        a, b = ds[i], r[i]
+        assert torch.allclose(a.pos, b.pos)
+        assert torch.equal(a.edge_index, b.edge_index)
+        assert torch.allclose(a.forces, b.forces)
+    # preload + subset
+    r2 = GraphStoreDataset(str(tmp_path), "total", preload=True,
+                           subset=[1, 3])
+    assert len(r2) == 2
+    assert torch.allclose(r2[1].pos, ds[3].pos)
+    assert r.get_node_counts() == [27] * len(ds)
+
+
+def test_distdataset_shim(tmp_path):
+    ds = _samples(4)
+    w = GraphStoreWriter("total", str(tmp_path))
+    w.add(ds)
+    w.save()
+    dd = DistDataset(str(tmp_path))
+    dd.epoch_begin()
+    s = dd[2]
+    dd.epoch_end()
+    assert torch.allclose(s.pos, ds[2].pos)
+    assert dd.get_node_counts() == [27] * 4
+
+
+def test_pickle_roundtrip(tmp_path):
+    ds = _samples(4)
+    SimplePickleWriter(ds, str(tmp_path), "total",
+                       attrs={"pna_deg": [1, 2]})
+    r = SimplePickleDataset(str(tmp_path), "total")
+    assert len(r) == 4
+    assert torch.allclose(r[1].pos, ds[1].pos)
+    assert r.pna_deg == [1, 2]
+
+
+def test_cost_aware_sampler():
+    sizes = [5, 9, 3, 8, 2, 7, 6, 4]
+    ds = []
+    for s in sizes:
+        d = Data(x=torch.zeros(s, 1))
+        d.num_nodes = s
+        ds.append(d)
+    sampler = CostAwareBatchSampler(ds, max_nodes=10, shuffle=False)
+    batches = list(sampler)
+    for b in batches:
+        assert sum(sizes[i] for i in b) <= 10
+    assert sorted(i for b in batches for i in b) == list(range(len(sizes)))
+    # deterministic under seed+epoch
+    s2 = CostAwareBatchSampler(ds, max_nodes=10, shuffle=True, seed=1)
+    s2.set_epoch(3)
+    p1 = list(s2)
+    s3 = CostAwareBatchSampler(ds, max_nodes=10, shuffle=True, seed=1)
+    s3.set_epoch(3)
+    assert p1 == list(s3)
+    # oversized policies
+    d_big = Data(x=torch.zeros(20, 1)); d_big.num_nodes = 20
+    with pytest.raises(ValueError):
+        list(CostAwareBatchSampler(ds + [d_big], max_nodes=10,
+                                   shuffle=False))
+    ok = CostAwareBatchSampler(ds + [d_big], max_nodes=10, shuffle=False,
+                               oversized_policy="single")
+    assert [len(ds)] in list(ok)
+
+
+def test_distributed_cost_aware_identical_plan():
+    sizes = [5, 9, 3, 8, 2, 7, 6, 4, 10, 1]
+    ds = []
+    for s in sizes:
+        d = Data(x=torch.zeros(s, 1))
+        d.num_nodes = s
+        ds.append(d)
+    s0 = DistributedCostAwareBatchSampler(ds, max_nodes=12, shuffle=True,
+                                          seed=2, num_replicas=2, rank=0)
+    s1 = DistributedCostAwareBatchSampler(ds, max_nodes=12, shuffle=True,
+                                          seed=2, num_replicas=2, rank=1)
+    s0.set_epoch(1); s1.set_epoch(1)
+    b0, b1 = list(s0), list(s1)
+    assert len(b0) == len(b1)  # equal steps on all ranks
+
+
+def test_lsms_parser(tmp_path):
+    raw = tmp_path / "raw"
+    raw.mkdir()
+    # graph features: 1 value at col 0; nodes: Z at col 5, charge col 6
+    (raw / "sample1.txt").write_text(
+        "1.5 0.0\n"
+        "0 0 0.0 0.0 0.0 26 8.1\n"
+        "1 0 1.0 0.0 0.0 26 7.9\n"
+        "2 0 0.0 1.0 0.0 28 9.9\n")
+    config = {
+        "Dataset": {
+            "name": "t", "format": "LSMS",
+            "path": {"total": str(raw)},
+            "node_features": {"name": ["Z", "charge"], "dim": [1, 1],
+                              "column_index": [5, 6]},
+            "graph_features": {"name": ["e"], "dim": [1],
+                               "column_index": [0]},
+        },
+        "NeuralNetwork": {"Architecture": {
+            "radius": 2.0, "max_neighbours": 10,
+            "periodic_boundary_conditions": False}},
+    }
+    from hydragnn_amd.utils.datasets.rawloaders import LSMSDataset
+    ds = LSMSDataset(config)
+    assert len(ds) == 1
+    d = ds[0]
+    assert d.num_nodes == 3
+    assert d.num_edges > 0
+    assert d.get("edge_attr") is not None  # normalized edge lengths
+
+
+def test_extxyz_parser(tmp_path):
+    raw = tmp_path / "raw"
+    raw.mkdir()
+    (raw / "mol.xyz").write_text(
+        '3\nLattice="5 0 0 0 5 0 0 0 5" energy=-7.2\n'
+        "O 0.0 0.0 0.0\nH 0.96 0.0 0.0\nH 0.0 0.96 0.0\n")
+    config = {
+        "Dataset": {
+            "name": "t", "format": "XYZ",
+            "path": {"total": str(raw)},
+            "node_features": {"name": ["Z"], "dim": [1],
+                              "column_index": [0]},
+            "graph_features": {"name": ["energy"], "dim": [1],
+                               "column_index": [0]},
+        },
+        "NeuralNetwork": {"Architecture": {
+            "radius": 1.5, "max_neighbours": 10,
+            "periodic_boundary_conditions": False}},
+    }
+    from hydragnn_amd.utils.datasets.rawloaders import XYZDataset
+    ds = XYZDataset(config)
+    assert len(ds) == 1
+    assert ds[0].num_nodes == 3
+    assert ds[0].z.tolist() == [8, 1, 1]
+
+
+def test_cfg_parser(tmp_path):
+    raw = tmp_path / "raw"
+    raw.mkdir()
+    (raw / "c.cfg").write_text(
+        "BEGIN_CFG\nSize\n2\nSupercell\n4 0 0\n0 4 0\n0 0 4\n"
+        "AtomData: id type cartes_x cartes_y cartes_z fx fy fz\n"
+        "1 13 0.0 0.0 0.0 0.1 0.0 0.0\n"
+        "2 13 1.5 0.0 0.0 -0.1 0.0 0.0\n"
+        "Energy\n-3.4\nEND_CFG\n")
+    config = {
+        "Dataset": {
+            "name": "t", "format": "CFG",
+            "path": {"total": str(raw)},
+            "node_features": {"name": ["Z"], "dim": [1],
+                              "column_index": [1]},
+            "graph_features": {"name": ["energy"], "dim": [1],
+                               "column_index": [0]},
+        },
+        "NeuralNetwork": {"Architecture": {
+            "radius": 2.0, "max_neighbours": 10,
+            "periodic_boundary_conditions": False}},
+    }
+    from hydragnn_amd.utils.datasets.rawloaders import CFGDataset
+    ds = CFGDataset(config)
+    assert len(ds) == 1
+    assert ds[0].num_nodes == 2
+    assert ds[0].forces.shape == (2, 3)
