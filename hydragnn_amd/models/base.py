@@ -440,8 +440,10 @@ class Base(Module):
         if batch is None:
             batch = torch.zeros(x.shape[0], dtype=torch.long, device=x.device)
             data["batch"] = batch
-        n_graphs = int(batch.max()) + 1 if batch.numel() > 0 else 1
-        x_graph = self.pool_fn(x, batch, n_graphs)
+        n_graphs = data.get("num_graphs_")
+        if n_graphs is None:
+            n_graphs = int(batch.max()) + 1 if batch.numel() > 0 else 1
+        x_graph = self.pool_fn(x, batch, int(n_graphs))
         x_graph = self._apply_graph_pool_conditioning(x_graph, data)
         return x, x_graph, equiv_node_feat, conv_args
 
@@ -460,8 +462,15 @@ class Base(Module):
             data["dataset_name"] = dataset_name
         if dataset_name.dim() == 1:
             dataset_name = dataset_name.view(-1, 1)
-        datasetIDs = dataset_name.unique()
-        _, node_counts = torch.unique_consecutive(batch, return_counts=True)
+        if self.num_branches > 1:
+            # device->host sync is acceptable only on the multi-branch
+            # path; the single-branch hot path stays hipGraph-capturable
+            datasetIDs = dataset_name.unique()
+            _, node_counts = torch.unique_consecutive(
+                batch, return_counts=True)
+        else:
+            datasetIDs = None
+            node_counts = None
 
         for head_dim, headloc, type_head in zip(self.head_dims, self.heads_NN,
                                                 self.head_type):

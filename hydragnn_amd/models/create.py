@@ -289,10 +289,15 @@ class EnhancedModelWrapper(torch.nn.Module):
         assert data.pos.requires_grad, "data.pos must require grad"
         assert self.num_heads == 1, "Force training requires exactly one head"
 
+        n_graphs = data.get("num_graphs_")
+        if n_graphs is None:
+            n_graphs = int(data.batch.max()) + 1
+        n_graphs = int(n_graphs)
         if self.head_type[0] == "node":
             node_energy_pred = pred[0]
             graph_energy_pred = scatter(
-                node_energy_pred, data.batch, None, "sum").squeeze().float()
+                node_energy_pred, data.batch, n_graphs,
+                "sum").squeeze(-1).float()
         elif self.head_type[0] == "graph":
             if getattr(self.model, "graph_pooling", "mean") not in ("add",):
                 raise ValueError(
@@ -317,7 +322,9 @@ class EnhancedModelWrapper(torch.nn.Module):
         if self.energy_weight > 0:
             tot_loss = tot_loss + tasks_loss[0] * self.energy_weight
 
-        natoms = torch.bincount(data.batch)
+        natoms = scatter(torch.ones_like(data.batch,
+                                         dtype=graph_energy_pred.dtype),
+                         data.batch, n_graphs, "sum")
         e_pa_pred = graph_energy_pred / natoms
         e_pa_true = graph_energy_true / natoms
         pa_loss = loss_fn(e_pa_pred, e_pa_true)

@@ -121,7 +121,9 @@ class MACEStack(Base):
             data["batch"] = batch
         # center positions per graph (keeps the autograd force path:
         # centering is translation-invariant so forces are unaffected)
-        n_graphs = int(batch.max()) + 1
+        n_graphs = data.get("num_graphs_")
+        n_graphs = int(n_graphs) if n_graphs is not None else \
+            int(batch.max()) + 1
         mean_pos = scatter(pos, batch, n_graphs, "mean")
         pos_c = pos - mean_pos[batch]
         vec, lengths = get_edge_vectors_and_lengths(
@@ -140,7 +142,9 @@ class MACEStack(Base):
         n = h0.shape[0]
         C = self.hidden_dim
         node_feats = h0.view(n, C, 1)
-        n_graphs = int(batch.max()) + 1
+        n_graphs = data.get("num_graphs_")
+        n_graphs = int(n_graphs) if n_graphs is not None else \
+            int(batch.max()) + 1
 
         head_outputs = [None] * self.num_heads
         for ilayer, (inter, prod) in enumerate(
