@@ -120,17 +120,36 @@ class EdgeTensorProduct(nn.Module):
         self.register_buffer("path_of_col", path_of_col)
         self.register_buffer("m3_of_col", m3_of_col)
         self.num_cols = cols
+        # fused-kernel entry table (ops/etp.py): one entry per nonzero
+        # Wigner coefficient, (a=m1 slot, b=m2 slot, g=path, o=m3 slot)
+        from ...ops.etp import ETPTable
+        ents, cfs = [], []
+        for p, (l1, l2, l3) in enumerate(self.paths):
+            W = wigner_3j(l1, l2, l3).float()
+            nz = (W.abs() > 1e-12).nonzero()
+            for (m1, m2, m3) in nz.tolist():
+                ents.append((l1 * l1 + m1, l2 * l2 + m2, p,
+                             l3 * l3 + m3))
+                cfs.append(float(W[m1, m2, m3]))
+        self.etp_table = ETPTable(
+            torch.tensor(ents, dtype=torch.long),
+            torch.tensor(cfs), (self.d1, self.d2, self.num_paths,
+                                self.d3))
 
     def forward(self, x_src: torch.Tensor, Y: torch.Tensor,
                 weights: torch.Tensor) -> torch.Tensor:
         """x_src [E, C, D_node], Y [E, D_edge],
         weights [E, C, num_paths] -> [E, C, D_out].
 
-        Contraction order chosen for HBM traffic (the op is
-        memory-bound): fold Y into the coupling table first
-        (WY [E, D1, cols], no channel dim) then one batched GEMM over
-        edges — avoids materializing the E x C x D1 x D2 outer
-        product."""
+        GPU: one fused HIP kernel (ops/etp.py) computes the whole
+        contraction at the memory-bound roofline, including both
+        autograd passes of force training.  CPU/fp64: fold Y into the
+        coupling table first (WY [E, D1, cols], no channel dim) then a
+        batched GEMM over edges — avoids materializing the
+        E x C x D1 x D2 outer product."""
+        from ...ops.etp import _kernel_ok, etp_general
+        if _kernel_ok(self.etp_table, x_src, Y, weights):
+            return etp_general(x_src, Y, weights, self.etp_table)
         E, C, _ = x_src.shape
         W3 = self.Wcat.to(x_src.dtype).view(self.d1, self.d2,
                                             self.num_cols)

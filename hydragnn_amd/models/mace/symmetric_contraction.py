@@ -120,27 +120,36 @@ class Contraction(nn.Module):
                 torch.randn(num_elements, k, num_channels) / max(k, 1))
             self.weights[str(nu)] = w
 
+    def _cu(self, nu_key: int, node_elem, x_dtype):
+        """U[m] . w  ->  [n, c, o, i1..i_m] via ONE dense GEMM
+        (M = n*c, K = num_paths, N = flat) instead of tiny-batched
+        einsums."""
+        U = getattr(self, f"U{nu_key}").to(x_dtype)
+        w = self.weights[str(nu_key)][node_elem].to(x_dtype)  # [N,k,C]
+        n, k, c = w.shape
+        flat = U.reshape(-1, max(k, 1))  # [o*D^m, k]
+        wp = w.permute(0, 2, 1).reshape(n * c, k)
+        cu = wp @ flat.t()  # [n*c, o*D^m]
+        return cu.reshape((n, c) + U.shape[:-1])
+
     def forward(self, x: torch.Tensor, node_elem: torch.Tensor
                 ) -> torch.Tensor:
         """x [N, C, D]; node_elem [N] element ids -> [N, C, 2lout+1]."""
+        from ...ops.etp import fold_last
         nu = self.correlation
-        U = getattr(self, f"U{nu}").to(x.dtype)  # [o, i1..inu, k]
-        w = self.weights[str(nu)][node_elem].to(x.dtype)  # [N, k, C]
+        U = getattr(self, f"U{nu}")
         if U.shape[-1] == 0:
-            n, c = x.shape[0], x.shape[1]
-            out = x.new_zeros(n, c, 2 * self.lout + 1)
+            D = x.shape[-1]
+            out = x.new_zeros((x.shape[0], x.shape[1],
+                               2 * self.lout + 1) + (D,) * (nu - 1))
         else:
-            # contract weights and the last slot with x
-            # U [o, i1..inu, k] ; w [n,k,c] -> [n, c, o, i1..i_{nu-1}]
-            out = torch.einsum("...ik,nkc,nci->nc...", U, w, x)
+            cu = self._cu(nu, node_elem, x.dtype)
+            out = fold_last(cu, x)  # [n, c, o, i1..i_{nu-1}]
         for m in range(nu - 1, 0, -1):
-            Um = getattr(self, f"U{m}").to(x.dtype)
+            Um = getattr(self, f"U{m}")
             if Um.shape[-1] > 0:
-                wm = self.weights[str(m)][node_elem].to(x.dtype)
-                cu = torch.einsum("...ik,nkc->nc...i", Um, wm)
-                out = cu + out
-            # contract one slot with x
-            out = torch.einsum("nc...i,nci->nc...", out, x)
+                out = self._cu(m, node_elem, x.dtype) + out
+            out = fold_last(out, x)
         return out  # [N, C, 2lout+1]
 
 

@@ -536,7 +536,17 @@ std::vector<torch::Tensor> radius_pairs(torch::Tensor pos, torch::Tensor batch,
   return {src, dst, dist};
 }
 
+// defined in etp.hip
+torch::Tensor etp_general(torch::Tensor A, torch::Tensor B, torch::Tensor C,
+                          torch::Tensor entries, torch::Tensor coefs,
+                          torch::Tensor o_ranges, long do_);
+torch::Tensor etp_reduce(torch::Tensor A, torch::Tensor C, torch::Tensor D,
+                         torch::Tensor entries, torch::Tensor coefs,
+                         long db);
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("etp_general", &etp_general, "fused ETP contraction (HIP)");
+  m.def("etp_reduce", &etp_reduce, "fused ETP channel-reduce (HIP)");
   m.def("gather_fwd", &gather_fwd, "gather rows (HIP)");
   m.def("scatter_sum_fwd", &scatter_sum_fwd, "scatter-add (HIP)");
   m.def("scatter_mean_fwd", &scatter_mean_fwd, "scatter-mean (HIP)");

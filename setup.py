@@ -9,7 +9,8 @@ from torch.utils.cpp_extension import BuildExtension, CUDAExtension  # noqa: E40
 ext_modules = [
     CUDAExtension(
         name="hydragnn_amd.ops._hip_ops",
-        sources=["hydragnn_amd/ops/csrc/hip_ops.hip"],
+        sources=["hydragnn_amd/ops/csrc/hip_ops.hip",
+                 "hydragnn_amd/ops/csrc/etp.hip"],
         extra_compile_args={
             "cxx": ["-O3"],
             "nvcc": ["-O3", "--offload-arch=gfx950"],
