@@ -124,9 +124,14 @@ class Contraction(nn.Module):
         """U[m] . w  ->  [n, c, o, i1..i_m] via ONE dense GEMM
         (M = n*c, K = num_paths, N = flat) instead of tiny-batched
         einsums."""
+        from ...ops import gather
         U = getattr(self, f"U{nu_key}").to(x_dtype)
-        w = self.weights[str(nu_key)][node_elem].to(x_dtype)  # [N,k,C]
-        n, k, c = w.shape
+        wfull = self.weights[str(nu_key)].to(x_dtype)
+        nel, k, c = wfull.shape
+        # per-node element gather via our op (backward = fused
+        # scatter-add instead of torch's indexing_backward kernel)
+        w = gather(wfull.reshape(nel, k * c), node_elem).view(-1, k, c)
+        n = w.shape[0]
         flat = U.reshape(-1, max(k, 1))  # [o*D^m, k]
         wp = w.permute(0, 2, 1).reshape(n * c, k)
         cu = wp @ flat.t()  # [n*c, o*D^m]

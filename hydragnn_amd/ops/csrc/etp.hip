@@ -39,41 +39,60 @@ __device__ inline __hip_bfloat16 from_f32<__hip_bfloat16>(float v) {
   return __float2bfloat16(v);
 }
 
-// out[i,c,o] = sum over entries(coef, a, b, g) grouped by o of
+// out[i,c,o] = sum over entries(coef, a, b, g, o) of
 //              coef * A[i,c,a] * B[i,b] * C[i,c,g]
+//
+// One THREAD per (i, c): its A/B/C rows and do-accumulator live in a
+// per-thread LDS slice (runtime entry indices would force register
+// arrays to scratch otherwise).  Every thread walks the SAME entry
+// list in lockstep (entry words are wave-uniform LDS broadcasts), so
+// there is no divergence; global loads/stores are per-thread
+// contiguous rows -> coalesced across adjacent threads.  Slice stride
+// is padded to an odd word count to spread LDS banks.
 template <typename T>
 __global__ void etp_general_kernel(
     const T* __restrict__ A, const T* __restrict__ B,
     const T* __restrict__ C, T* __restrict__ out,
-    const int4* __restrict__ entries,   // (a, b, g, o) sorted by o
-    const float* __restrict__ coefs,
-    const int2* __restrict__ o_ranges,  // [do] (start, count)
+    const int4* __restrict__ entries,
+    const float* __restrict__ coefs, int n_ent,
     long NC, int nch, int da, int db, int dg, int do_) {
-  __shared__ float lds[kGroupsPerBlock][3 * kMaxDim];
-  int group = threadIdx.x / kGroup;
-  int lane = threadIdx.x % kGroup;
-  long i = (long)blockIdx.x * kGroupsPerBlock + group;
-  if (i >= NC) return;
-  long e = i / nch;
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  const int stride = ((da + db + dg + do_) | 1);  // odd word stride
+  float* slices = reinterpret_cast<float*>(smem);
+  int4* ent_lds = reinterpret_cast<int4*>(
+      smem + (size_t)blockDim.x * stride * 4);
+  float* coef_lds = reinterpret_cast<float*>(ent_lds + n_ent);
 
-  float* la = lds[group];
-  float* lb = la + kMaxDim;
-  float* lc = lb + kMaxDim;
-  // cooperative stage of the three rows
-  for (int k = lane; k < da; k += kGroup) la[k] = to_f32(A[i * da + k]);
-  for (int k = lane; k < db; k += kGroup) lb[k] = to_f32(B[e * db + k]);
-  for (int k = lane; k < dg; k += kGroup) lc[k] = to_f32(C[i * dg + k]);
-  __builtin_amdgcn_s_waitcnt(0);  // lgkmcnt(0): LDS writes visible
-  __builtin_amdgcn_wave_barrier();
+  // stage the entry table once per block
+  for (int k = threadIdx.x; k < n_ent; k += blockDim.x) {
+    ent_lds[k] = entries[k];
+    coef_lds[k] = coefs[k];
+  }
 
-  if (lane < do_) {
-    int2 r = o_ranges[lane];
-    float acc = 0.f;
-    for (int k = r.x; k < r.x + r.y; ++k) {
-      int4 q = entries[k];
-      acc += coefs[k] * la[q.x] * lb[q.y] * lc[q.z];
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  float* my = slices + (size_t)threadIdx.x * stride;
+  float* ma = my;
+  float* mb = ma + da;
+  float* mc = mb + db;
+  float* mo = mc + dg;
+  if (i < NC) {
+    long e = i / nch;
+    const T* ap = A + i * da;
+    const T* bp = B + e * db;
+    const T* cp = C + i * dg;
+    for (int k = 0; k < da; ++k) ma[k] = to_f32(ap[k]);
+    for (int k = 0; k < db; ++k) mb[k] = to_f32(bp[k]);
+    for (int k = 0; k < dg; ++k) mc[k] = to_f32(cp[k]);
+    for (int k = 0; k < do_; ++k) mo[k] = 0.f;
+  }
+  __syncthreads();
+  if (i < NC) {
+    for (int k = 0; k < n_ent; ++k) {
+      int4 q = ent_lds[k];
+      mo[q.w] += coef_lds[k] * ma[q.x] * mb[q.y] * mc[q.z];
     }
-    out[i * do_ + lane] = from_f32<T>(acc);
+    T* op = out + i * do_;
+    for (int k = 0; k < do_; ++k) op[k] = from_f32<T>(mo[k]);
   }
 }
 
@@ -138,19 +157,23 @@ torch::Tensor etp_general(torch::Tensor A, torch::Tensor B, torch::Tensor C,
               "etp dims exceed kernel limits");
   auto out = torch::empty({A.size(0), A.size(1), do_}, A.options());
   if (NC == 0) return out;
-  long blocks = (NC + kGroupsPerBlock - 1) / kGroupsPerBlock;
+  int n_ent = entries.size(0);
+  int block = 256;
+  int stride = (da + db + dg + (int)do_) | 1;
+  size_t lds_bytes = (size_t)block * stride * 4 + n_ent * 20;
+  TORCH_CHECK(lds_bytes <= 150 * 1024, "etp LDS budget exceeded");
+  long blocks = (NC + block - 1) / block;
   // fp64 is routed to the eager path in Python (float LDS staging here)
   AT_DISPATCH_FLOATING_TYPES_AND2(
       at::ScalarType::BFloat16, at::ScalarType::Half, A.scalar_type(),
       "etp_general", [&] {
         hipLaunchKernelGGL(
-            etp_general_kernel<scalar_t>, dim3(blocks),
-            dim3(kGroup * kGroupsPerBlock), 0, etp_stream(),
+            etp_general_kernel<scalar_t>, dim3(blocks), dim3(block),
+            lds_bytes, etp_stream(),
             A.data_ptr<scalar_t>(), B.data_ptr<scalar_t>(),
             C.data_ptr<scalar_t>(), out.data_ptr<scalar_t>(),
             reinterpret_cast<const int4*>(entries.data_ptr<int>()),
-            coefs.data_ptr<float>(),
-            reinterpret_cast<const int2*>(o_ranges.data_ptr<int>()),
+            coefs.data_ptr<float>(), n_ent,
             NC, nch, da, db, dg, (int)do_);
       });
   return out;
