@@ -174,14 +174,15 @@ class RealAgnosticResidualInteractionBlock(nn.Module):
         self.avg_num_neighbors = avg_num_neighbors
         self.linear_up = IrrepsLinear(num_channels, num_channels, lmax_node)
         self.conv_tp = EdgeTensorProduct(lmax_node, lmax_edge, lmax_out)
+        from ...ops.splitk_linear import SplitKLinear
         hidden = radial_mlp or [64, 64, 64]
         mods = []
         prev = radial_dim
         for h in hidden:
-            mods += [nn.Linear(prev, h), nn.SiLU()]
+            mods += [SplitKLinear(prev, h), nn.SiLU()]
             prev = h
         mods.append(
-            nn.Linear(prev, num_channels * self.conv_tp.num_paths))
+            SplitKLinear(prev, num_channels * self.conv_tp.num_paths))
         self.radial_mlp = nn.Sequential(*mods)
         self.num_channels = num_channels
         self.linear = IrrepsLinear(num_channels, num_channels, lmax_out)

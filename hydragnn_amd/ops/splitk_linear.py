@@ -1,0 +1,59 @@
+"""Linear layer with a split-K weight gradient.
+
+Per-edge MLPs (e.g. the MACE radial network) have tiny in/out widths
+but E ~ 10^5-10^6 rows: the weight-gradient GEMM dW = X^T g has
+M, N <= a few hundred and K = E, which hipBLASLt schedules on only
+(M/16)x(N/16) workgroups — a serial-K crawl on a 256-CU chip.  Here the
+K dimension is split into S chunks computed as a bmm (S x tiles
+workgroups) and summed — an order of magnitude faster at these shapes.
+
+backward() is built from differentiable torch ops, so the force
+training double-backward works unchanged.
+"""
+
+from __future__ import annotations
+
+import torch
+from torch import nn
+
+
+def _splitk_weight_grad(x2d: torch.Tensor, g2d: torch.Tensor,
+                        chunks: int) -> torch.Tensor:
+    E = x2d.shape[0]
+    S = min(chunks, max(1, E // 256))
+    if S <= 1:
+        return g2d.t() @ x2d
+    pad = (S - E % S) % S
+    if pad:
+        x2d = torch.nn.functional.pad(x2d, (0, 0, 0, pad))
+        g2d = torch.nn.functional.pad(g2d, (0, 0, 0, pad))
+    xs = x2d.view(S, -1, x2d.shape[1])
+    gs = g2d.view(S, -1, g2d.shape[1])
+    return torch.bmm(gs.transpose(1, 2), xs).sum(0)
+
+
+class _SplitKLinearFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, weight, bias):
+        ctx.save_for_backward(x, weight)
+        ctx.has_bias = bias is not None
+        return torch.nn.functional.linear(x, weight, bias)
+
+    @staticmethod
+    def backward(ctx, g):
+        x, weight = ctx.saved_tensors
+        g2d = g.reshape(-1, g.shape[-1])
+        x2d = x.reshape(-1, x.shape[-1])
+        gx = gw = gb = None
+        if ctx.needs_input_grad[0]:
+            gx = (g2d @ weight).view_as(x)
+        if ctx.needs_input_grad[1]:
+            gw = _splitk_weight_grad(x2d, g2d, 64)
+        if ctx.has_bias and ctx.needs_input_grad[2]:
+            gb = g2d.sum(0)
+        return gx, gw, gb
+
+
+class SplitKLinear(nn.Linear):
+    def forward(self, x):
+        return _SplitKLinearFn.apply(x, self.weight, self.bias)
