@@ -43,12 +43,12 @@ class _SplitKLinearFn(torch.autograd.Function):
     def backward(ctx, g):
         x, weight = ctx.saved_tensors
         g2d = g.reshape(-1, g.shape[-1])
-        x2d = x.reshape(-1, x.shape[-1])
+        x2d = x.reshape(-1, x.shape[-1]).to(g2d.dtype)
         gx = gw = gb = None
         if ctx.needs_input_grad[0]:
-            gx = (g2d @ weight).view_as(x)
+            gx = (g2d @ weight.to(g2d.dtype)).view_as(x).to(x.dtype)
         if ctx.needs_input_grad[1]:
-            gw = _splitk_weight_grad(x2d, g2d, 64)
+            gw = _splitk_weight_grad(x2d, g2d, 64).to(weight.dtype)
         if ctx.has_bias and ctx.needs_input_grad[2]:
             gb = g2d.sum(0)
         return gx, gw, gb
