@@ -153,7 +153,7 @@ torch::Tensor etp_general(torch::Tensor A, torch::Tensor B, torch::Tensor C,
   long NC = A.size(0) * A.size(1);
   int nch = A.size(1);
   int da = A.size(2), db = B.size(1), dg = C.size(2);
-  TORCH_CHECK(da <= 40 && db <= 40 && dg <= 40 && do_ <= 32,
+  TORCH_CHECK(da <= 192 && db <= 192 && dg <= 192 && do_ <= 192,
               "etp dims exceed kernel limits");
   auto out = torch::empty({A.size(0), A.size(1), do_}, A.options());
   if (NC == 0) return out;

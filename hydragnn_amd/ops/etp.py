@@ -89,7 +89,11 @@ def _dense_general(A, B, C, table: ETPTable):
 
 def _kernel_ok(table: ETPTable, *tensors) -> bool:
     da, db, dg, do = table.dims
-    if do > 32 or max(da, db, dg) > 40:
+    # real constraint is the per-block LDS budget (256 thread slices +
+    # staged entry table) — mirrors the check in csrc/etp.hip
+    lds_bytes = 256 * (da + db + dg + do + 1) * 4 + \
+        table.entries.shape[0] * 20
+    if lds_bytes > 150 * 1024 or max(da, db, dg, do) > 192:
         return False
     t = tensors[0]
     return (t.is_cuda and t.dtype in _KERNEL_DTYPES and not use_eager())
