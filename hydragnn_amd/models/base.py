@@ -78,9 +78,26 @@ class Base(Module):
         graph_attr_conditioning_mode: str = "concat_node",
         graph_attr_dim: int = 0,
         var_output: bool = False,
+        equivariant_attn_lmax: int = 1,
+        equivariant_attn_num_radial: int = 16,
+        equivariant_attn_feedforward_multiplier: int = 2,
+        equivariant_attn_allow_scalar_only: bool = False,
+        equivariant_attn_require_tensor_coupling: bool = True,
+        equivariant_attn_chunk_size: Optional[int] = 512,
+        equivariant_attn_coupling_mode: str = "parallel",
         **kwargs,
     ):
         super().__init__()
+        self.equivariant_attn_lmax = equivariant_attn_lmax
+        self.equivariant_attn_num_radial = equivariant_attn_num_radial
+        self.equivariant_attn_feedforward_multiplier = \
+            equivariant_attn_feedforward_multiplier
+        self.equivariant_attn_allow_scalar_only = \
+            equivariant_attn_allow_scalar_only
+        self.equivariant_attn_require_tensor_coupling = \
+            equivariant_attn_require_tensor_coupling
+        self.equivariant_attn_chunk_size = equivariant_attn_chunk_size
+        self.equivariant_attn_coupling_mode = equivariant_attn_coupling_mode
         self.input_dim = input_dim
         self.hidden_dim = hidden_dim
         self.head_dims = list(output_dim)
@@ -147,7 +164,9 @@ class Base(Module):
         self.feature_layers = ModuleList()
         self.heads_NN = ModuleList()
 
-        if self.use_global_attn:
+        self.is_equivariant_attn = (self.global_attn_engine or "") == \
+            "EquivariantTransformer"
+        if self.use_global_attn and not self.is_equivariant_attn:
             # GPS positional/feature encoders (reference Base.py:215-260)
             self.pos_emb = Linear(max(self.pe_dim, 1), self.hidden_dim)
             if self.input_dim:
@@ -180,10 +199,31 @@ class Base(Module):
         raise NotImplementedError
 
     def _apply_global_attn(self, conv):
-        """Wrap a local conv in a GPS layer when global attention is on
-        (reference Base.py:264)."""
+        """Wrap a local conv in a GPS or EquivariantTransformer layer
+        when global attention is on (reference Base.py:264)."""
         if not self.use_global_attn:
             return conv
+        if self.is_equivariant_attn:
+            from ..globalatt.equivariant import (
+                EquivariantLocalGlobalConv,
+                EquivariantTransformerLayer,
+                create_local_feature_adapter,
+            )
+            mpnn = str(self).replace("Stack", "").replace("SCF", "SchNet")
+            lmax = self.equivariant_attn_lmax
+            adapter = create_local_feature_adapter(
+                mpnn, self.hidden_dim, self.hidden_dim, lmax,
+                allow_scalar_only=self.equivariant_attn_allow_scalar_only)
+            layer = EquivariantTransformerLayer(
+                self.hidden_dim, lmax,
+                num_heads=self.global_attn_heads or 4,
+                num_radial=self.equivariant_attn_num_radial,
+                feedforward_multiplier=
+                self.equivariant_attn_feedforward_multiplier,
+                chunk_size=self.equivariant_attn_chunk_size)
+            return EquivariantLocalGlobalConv(
+                conv, adapter, layer,
+                mode=self.equivariant_attn_coupling_mode)
         from ..globalatt.gps import HydraGPSConv
         return HydraGPSConv(
             self.hidden_dim, conv, heads=self.global_attn_heads or 1,
@@ -353,6 +393,9 @@ class Base(Module):
                                              torch.bfloat16, torch.float16):
             x = x.float()
         pos = data.get("pos")
+        if self.use_global_attn and self.is_equivariant_attn:
+            conv_args["batch"] = data.get("batch")
+            return x, pos, conv_args
         if self.use_global_attn:
             conv_args["batch"] = data.get("batch")
             dt = self.pos_emb.weight.dtype

@@ -141,9 +141,9 @@ class PAINNStack(Base):
     def _init_conv(self):
         from .base import BatchNormNode
         for _ in range(self.num_conv_layers):
-            self.graph_convs.append(
+            self.graph_convs.append(self._apply_global_attn(
                 _PainnWrapper(_PainnConv(self.hidden_dim, self.hidden_dim,
-                                         self.num_radial, self.radius)))
+                                         self.num_radial, self.radius))))
             self.feature_layers.append(BatchNormNode(self.hidden_dim))
 
     def _embedding(self, data):
@@ -160,6 +160,8 @@ class PAINNStack(Base):
             "edge_dist": edge_dist,
             "vec_state": {},
         }
+        if self.use_global_attn:
+            conv_args["batch"] = data.get("batch")
         return x, data.pos, conv_args
 
     def __str__(self):
