@@ -1,0 +1,44 @@
+"""Example scripts run as subprocesses (pattern: reference
+tests/test_examples.py:18-90)."""
+
+import os
+import subprocess
+import sys
+
+import pytest
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def _run(script, *args, timeout=420):
+    env = dict(os.environ)
+    env["MASTER_ADDR"] = "127.0.0.1"
+    return subprocess.run(
+        [sys.executable, os.path.join(REPO, script), *args],
+        capture_output=True, text=True, timeout=timeout, cwd=REPO,
+        env=env)
+
+
+@pytest.mark.parametrize("mpnn_type", ["PNA", "SchNet"])
+def test_qm9_example(mpnn_type, tmp_path):
+    r = _run("examples/qm9/qm9.py", "--mpnn_type", mpnn_type,
+             "--num_epoch", "2", "--num_samples", "48")
+    assert r.returncode == 0, r.stderr[-2000:]
+
+
+@pytest.mark.parametrize("mpnn_type", ["MACE", "EGNN"])
+def test_md17_mlip_example(mpnn_type):
+    r = _run("examples/md17/md17_mlip.py", "--mpnn_type", mpnn_type,
+             "--num_epoch", "2", "--num_samples", "24")
+    assert r.returncode == 0, r.stderr[-2000:]
+
+
+def test_lennardjones_example():
+    r = _run("examples/LennardJones/LennardJones.py",
+             "--num_epoch", "2", "--num_samples", "16")
+    assert r.returncode == 0, r.stderr[-2000:]
+
+
+def test_multibranch_example():
+    r = _run("examples/multibranch/train.py", "--num_epoch", "2")
+    assert r.returncode == 0, r.stderr[-2000:]
