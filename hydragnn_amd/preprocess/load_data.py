@@ -62,10 +62,25 @@ def create_dataloaders(trainset, valset, testset, batch_size: int,
 
     use_dist = dist.is_initialized() and dist.get_world_size() > 1
 
+    use_custom = os.getenv("HYDRAGNN_CUSTOM_DATALOADER", "0") == "1"
+
     def make(ds, shuffle):
         if ds is None or len(ds) == 0:
             return TorchDataLoader([], batch_size=batch_size,
                                    collate_fn=_collate)
+        if use_custom:
+            from .dataloader import HydraDataLoader
+            bs = None
+            if batching is not None and \
+                    batching.get("mode") == "node_budget":
+                cls = (DistributedCostAwareBatchSampler if use_dist
+                       else CostAwareBatchSampler)
+                bs = cls(ds, max_nodes=batching["max_nodes"],
+                         shuffle=shuffle, seed=batching.get("seed", 0))
+            return HydraDataLoader(
+                ds, batch_size=batch_size, shuffle=shuffle,
+                batch_sampler=bs,
+                num_workers=max(num_workers, 2))
         if batching is not None and batching.get("mode") == "node_budget":
             max_nodes = batching["max_nodes"]
             if use_dist:

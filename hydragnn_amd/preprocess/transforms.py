@@ -78,3 +78,31 @@ def normalize_rotation(data: Data) -> Data:
     if data.get("forces") is not None:
         data.forces = (data.forces.double() @ v).to(data.forces.dtype)
     return data
+
+
+def pbc_as_tensor(pbc) -> "torch.Tensor":
+    """Normalize a pbc spec (bool | sequence of 3) to a bool [3] tensor
+    (reference graph_samples_checks_and_updates.py:506)."""
+    if isinstance(pbc, bool):
+        return torch.tensor([pbc] * 3)
+    t = torch.as_tensor(pbc).flatten().bool()
+    assert t.numel() == 3, "pbc must have 3 components"
+    return t
+
+
+def pbc_distance(data, norm: bool = False, max_length: float = 1.0):
+    """PBC-aware Distance transform: edge lengths using edge_shifts
+    (reference PBCDistance, graph_samples:439)."""
+    return add_edge_lengths(data, max_length=max_length if norm else 1.0)
+
+
+def pbc_local_cartesian(data):
+    """PBC-aware LocalCartesian: shift-corrected edge vectors as edge
+    attrs (reference PBCLocalCartesian, graph_samples:470)."""
+    vec, _ = get_edge_vectors_and_lengths(
+        data.pos, data.edge_index, data.get("edge_shifts"))
+    existing = data.get("edge_attr")
+    vec = vec.to(torch.float32)
+    data.edge_attr = torch.cat([existing, vec], dim=-1) \
+        if existing is not None else vec
+    return data

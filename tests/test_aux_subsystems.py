@@ -1,0 +1,138 @@
+"""Aux subsystem unit tests: energy regression, download safety, LSMS
+utils, descriptors, materials preprocessing, HPO helpers, custom
+dataloader, tracer."""
+
+import os
+
+import numpy as np
+import pytest
+import torch
+
+from hydragnn_amd.data import Data
+from hydragnn_amd.preprocess import HydraDataLoader, parse_omp_places
+from hydragnn_amd.preprocess.energy_linear_regression import (
+    energy_linear_regression,
+    shift_energies,
+)
+from hydragnn_amd.utils.datasets.download import safe_extract_tar, sha256_of
+from hydragnn_amd.utils.datasets.synthetic import lj_dataset
+from hydragnn_amd.utils.descriptors_and_embeddings import atomicdescriptors
+from hydragnn_amd.utils.hpo.deephyper import (
+    parse_slurm_nodelist,
+    run_random_search,
+)
+from hydragnn_amd.utils.lsms import (
+    compositional_histogram_cutoff,
+    convert_raw_data_energy_to_gibbs,
+)
+from hydragnn_amd.utils.materials.preprocessing import (
+    normalize_stress,
+    validate_atomistic_sample,
+    voigt_to_full,
+)
+
+
+def test_energy_linear_regression_recovers_reference_energies():
+    g = torch.Generator().manual_seed(0)
+    e_ref = {1: -13.6, 6: -1030.0, 8: -2040.0}
+    ds = []
+    for _ in range(30):
+        z = torch.cat([
+            torch.full((int(torch.randint(1, 5, (1,), generator=g)),), zz)
+            for zz in (1, 6, 8)]).long()
+        e = sum(e_ref[int(v)] for v in z) + float(
+            torch.randn(1, generator=g)) * 1e-3
+        d = Data(z=z, x=z.float().view(-1, 1),
+                 energy=torch.tensor([[e]]), y=torch.tensor([[e]]))
+        d.num_nodes = z.numel()
+        ds.append(d)
+    e_fit, present = energy_linear_regression(ds, distributed=False)
+    for zz, val in e_ref.items():
+        assert abs(e_fit[zz] - val) < 0.05, (zz, e_fit[zz])
+    shift_energies(ds, e_fit)
+    assert abs(float(ds[0].energy)) < 1.0
+
+
+def test_safe_tar_rejects_traversal(tmp_path):
+    import tarfile
+    import io
+    bad = tmp_path / "bad.tar"
+    with tarfile.open(bad, "w") as tar:
+        info = tarfile.TarInfo("../evil.txt")
+        data = b"x"
+        info.size = len(data)
+        tar.addfile(info, io.BytesIO(data))
+    with pytest.raises(ValueError):
+        safe_extract_tar(str(bad), str(tmp_path / "out"))
+
+
+def test_sha256(tmp_path):
+    p = tmp_path / "f.bin"
+    p.write_bytes(b"hello")
+    assert sha256_of(str(p)).startswith("2cf24dba")
+
+
+def test_lsms_utils():
+    ds = lj_dataset(num_samples=6, num_atoms=27, pbc=False)
+    pure = {13: float(ds[0].y) / 27}
+    convert_raw_data_energy_to_gibbs(ds, pure)
+    assert abs(float(ds[0].y)) < 1e-4  # formation energy of "pure" = 0
+    kept = compositional_histogram_cutoff(ds, 13, max_per_bin=2)
+    assert len(kept) == 2
+
+
+def test_atomic_descriptors():
+    ad = atomicdescriptors(element_types=[1, 6, 8])
+    f = ad.get_atom_features(torch.tensor([1, 6, 8]))
+    assert f.shape == (3, 3 + 5)
+    assert f[0, 0] == 1.0 and f[1, 1] == 1.0
+
+
+def test_materials_preprocessing():
+    s = voigt_to_full(torch.tensor([1., 2., 3., 4., 5., 6.]))
+    assert s[0, 0] == 1 and s[1, 2] == 4 and s[0, 1] == 6
+    n = normalize_stress(torch.eye(3), units="GPa")
+    assert abs(float(n[0, 0]) - 1 / 160.21766208) < 1e-9
+    d = lj_dataset(num_samples=1, num_atoms=27, pbc=False)[0]
+    validate_atomistic_sample(d, require_forces=True)
+    bad = d.clone()
+    bad.forces = bad.forces[:3]
+    with pytest.raises(ValueError, match="forces"):
+        validate_atomistic_sample(bad, require_forces=True)
+
+
+def test_hpo_helpers():
+    hosts = parse_slurm_nodelist("node[001-003,007]")
+    assert hosts == ["node001", "node002", "node003", "node007"]
+    best, val, hist = run_random_search(
+        lambda c: (c["lr"] - 0.01) ** 2,
+        {"lr": (1e-4, 1e-1, "log"), "dim": [16, 32]}, num_trials=20)
+    assert len(hist) == 20 and val < 0.01
+
+
+def test_parse_omp_places():
+    assert parse_omp_places("{0},{1},{2}") == [0, 1, 2]
+    assert parse_omp_places("{0:4}") == [0, 1, 2, 3]
+
+
+def test_hydra_dataloader_matches_default():
+    ds = lj_dataset(num_samples=10, num_atoms=8, pbc=False)
+    loader = HydraDataLoader(ds, batch_size=4, shuffle=False,
+                             num_workers=2)
+    batches = list(loader)
+    assert len(loader) == 3 and len(batches) == 3
+    assert batches[0].num_graphs == 4
+    assert batches[-1].num_graphs == 2
+    assert torch.allclose(batches[0].pos[:8], ds[0].pos)
+
+
+def test_tracer_roundtrip(tmp_path):
+    from hydragnn_amd.utils.profiling_and_tracing import tracer as tr
+    tr.reset()
+    tr.enable()
+    tr.start("region")
+    tr.stop("region")
+    tr.save(str(tmp_path))
+    assert (tmp_path / "gp_timing.p0").exists()
+    tr.disable()
+    tr.reset()
