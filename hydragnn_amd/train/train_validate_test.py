@@ -174,6 +174,9 @@ def train(loader, model, opt, verbosity, precision="fp32", profiler=None):
     autocast, scaler = get_autocast_and_scaler(precision)
     nbatch = get_nbatch(loader)
     use_ia = _use_interatomic(m)
+    dataset = getattr(loader, "dataset", None)
+    if dataset is not None and hasattr(dataset, "epoch_begin"):
+        dataset.epoch_begin()   # DDStore-style fetch window
     if use_ia and is_fsdp2_enabled():
         # FSDP2 double-backward workaround (see set_reshard_after_backward)
         set_reshard_after_backward(model, False)
@@ -209,6 +212,8 @@ def train(loader, model, opt, verbosity, precision="fp32", profiler=None):
             tasks_error[it] += tl.detach() * n
         num_samples_local += n
 
+    if dataset is not None and hasattr(dataset, "epoch_end"):
+        dataset.epoch_end()
     denom = max(num_samples_local, 1)
     train_error = reduce_values_ranks(total_error / denom)
     train_tasks_error = reduce_values_ranks(tasks_error / denom)
@@ -346,10 +351,14 @@ def train_validate_test(
         train_error, train_tasks_error = train(
             train_loader, model, optimizer, verbosity, precision, profiler)
         tr.stop("train")
-        val_error, val_tasks_error = validate(val_loader, model, verbosity,
-                                              precision)
-        test_error, test_tasks_error = _eval_pass(
-            test_loader, model, verbosity, precision)
+        if os.getenv("HYDRAGNN_VALTEST", "1") != "0":
+            val_error, val_tasks_error = validate(
+                val_loader, model, verbosity, precision)
+            test_error, test_tasks_error = _eval_pass(
+                test_loader, model, verbosity, precision)
+        else:
+            val_error = train_error.clone()
+            test_error = train_error.clone()
 
         if scheduler is not None:
             scheduler.step(val_error)

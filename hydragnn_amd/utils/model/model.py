@@ -34,22 +34,75 @@ def _unwrap(model):
     return model.module if hasattr(model, "module") else model
 
 
+def _full_state_dict(m):
+    """FSDP1 full-state-dict gather when applicable (reference
+    model.py:142)."""
+    try:
+        from torch.distributed.fsdp import (
+            FullyShardedDataParallel as FSDP, StateDictType,
+            FullStateDictConfig)
+        if isinstance(m, FSDP):
+            cfg = FullStateDictConfig(offload_to_cpu=True,
+                                      rank0_only=True)
+            with FSDP.state_dict_type(m, StateDictType.FULL_STATE_DICT,
+                                      cfg):
+                return m.state_dict()
+    except ImportError:
+        pass
+    sd = m.state_dict()
+    # FSDP2 DTensor shards -> full tensors
+    out = {}
+    for k, v in sd.items():
+        if hasattr(v, "full_tensor"):
+            v = v.full_tensor()
+        out[k] = v
+    return out
+
+
 def save_model(model, optimizer, name: str, epoch: Optional[int] = None,
                path: str = "./logs/") -> None:
     """torch.save {model_state_dict, optimizer_state_dict} to
-    logs/<name>/<name>[_epoch_E].pk + latest symlink."""
+    logs/<name>/<name>[_epoch_E].pk + latest symlink; MultiTaskModelMP
+    saves encoder plus a per-branch decoder file (branch-group rank0,
+    reference model.py:66-189)."""
     d = os.path.join(path, name)
     os.makedirs(d, exist_ok=True)
     if hasattr(optimizer, "consolidate_state_dict"):
         optimizer.consolidate_state_dict()
+    from ...models.multitask_mp import MultiTaskModelMP
+    if isinstance(_unwrap(model), MultiTaskModelMP) or \
+            isinstance(model, MultiTaskModelMP):
+        mt = model if isinstance(model, MultiTaskModelMP) else \
+            _unwrap(model)
+        enc = mt.encoder.module if hasattr(mt.encoder, "module") \
+            else mt.encoder
+        dec = mt.decoder.module if hasattr(mt.decoder, "module") \
+            else mt.decoder
+        if _rank() == 0:
+            torch.save({"model_state_dict": enc.state_dict(),
+                        "optimizer_state_dict":
+                        optimizer.state_dict()
+                        if optimizer is not None else {}},
+                       os.path.join(d, f"{name}.pk"))
+        branch_rank = dist.get_rank(mt.branch_group) \
+            if (dist.is_initialized() and mt.branch_group is not None) \
+            else 0
+        if branch_rank == 0:
+            torch.save({"model_state_dict": dec.state_dict()},
+                       os.path.join(
+                           d, f"{name}_branch{mt.branch_id}.pk"))
+        if dist.is_initialized():
+            dist.barrier()
+        return
     m = _unwrap(model)
     if _rank() == 0:
         fname = (f"{name}_epoch_{epoch}.pk" if epoch is not None
                  else f"{name}.pk")
         fpath = os.path.join(d, fname)
         torch.save({
-            "model_state_dict": m.state_dict(),
-            "optimizer_state_dict": optimizer.state_dict(),
+            "model_state_dict": _full_state_dict(m),
+            "optimizer_state_dict": optimizer.state_dict()
+            if optimizer is not None else {},
         }, fpath)
         latest = os.path.join(d, f"{name}.pk")
         if epoch is not None and fpath != latest:
