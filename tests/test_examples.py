@@ -42,3 +42,8 @@ def test_lennardjones_example():
 def test_multibranch_example():
     r = _run("examples/multibranch/train.py", "--num_epoch", "2")
     assert r.returncode == 0, r.stderr[-2000:]
+
+
+def test_multidataset_example():
+    r = _run("examples/multidataset/train.py", "--num_epoch", "2")
+    assert r.returncode == 0, r.stderr[-2000:]
