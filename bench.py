@@ -106,15 +106,14 @@ def build_model_and_batch(device="cuda:0", local_batch=LOCAL_BATCH,
     )
     _, param_dtype, _ = resolve_precision(precision)
     model = model.to(device=device, dtype=param_dtype)
+    m = model
 
-    if dist.is_initialized() and dist.get_world_size() > 1:
-        model = torch.nn.parallel.DistributedDataParallel(
-            model, device_ids=[torch.device(device)]
-            if torch.device(device).type == "cuda" else None,
-            find_unused_parameters=True)
-        m = model.module
-    else:
-        m = model
+    world = dist.get_world_size() if dist.is_initialized() else 1
+    if world > 1:
+        # replicate initial parameters across ranks
+        with torch.no_grad():
+            for p in model.parameters():
+                dist.broadcast(p.data, src=0)
 
     rank = dist.get_rank() if dist.is_initialized() else 0
     dataset = md17_shape_dataset(num_samples=local_batch,
@@ -128,21 +127,36 @@ def build_model_and_batch(device="cuda:0", local_batch=LOCAL_BATCH,
             batch[key] = v.to(param_dtype)
 
     use_graph = (torch.device(device).type == "cuda"
-                 and not (dist.is_initialized()
-                          and dist.get_world_size() > 1)
                  and os.environ.get("HYDRAGNN_HIPGRAPH", "1") == "1")
     optimizer = torch.optim.AdamW(model.parameters(), lr=1e-3,
                                   capturable=use_graph,
                                   foreach=True)
     autocast, _ = get_autocast_and_scaler(precision)
 
+    # Data parallelism as ONE flat-bucket all-reduce (SURVEY.md §2b:
+    # custom flat-bucket DP instead of DDP's per-bucket hooks): every
+    # parameter's .grad is a view into one flat buffer, backward
+    # accumulates into it, a single RCCL all-reduce syncs it — and the
+    # whole step (fwd + double-bwd + all-reduce + AdamW) is
+    # hipGraph-capturable, so multi-GPU keeps the captured fast path.
+    params = [p for p in model.parameters() if p.requires_grad]
+    total = sum(p.numel() for p in params)
+    flat_grad = torch.zeros(total, device=device, dtype=param_dtype)
+    off = 0
+    for p in params:
+        p.grad = flat_grad[off:off + p.numel()].view_as(p)
+        off += p.numel()
+
     def eager_step():
-        optimizer.zero_grad(set_to_none=not use_graph)
+        flat_grad.zero_()
         batch.pos.requires_grad_(True)
         with autocast:
             pred = model(batch)
             loss, _ = m.energy_force_loss(pred, batch, create_graph=True)
         loss.backward()
+        if world > 1:
+            dist.all_reduce(flat_grad)
+            flat_grad.div_(world)
         optimizer.step()
         return loss
 

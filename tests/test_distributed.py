@@ -170,3 +170,25 @@ def test_fsdp2_force_grad_regression():
         p.join(timeout=60)
     for rank, ok, info in results:
         assert ok, f"rank {rank}: {info}"
+
+
+def test_bench_two_rank_gloo():
+    """The exact driver launch path: torchrun 2 ranks, CPU/gloo."""
+    import subprocess
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    import sys
+    cmd = [sys.executable,
+         "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29541", os.path.join(repo, "bench.py"),
+         "--gpus", "2", "--steps", "2", "--warmup", "1",
+         "--batch", "4"]
+    r = subprocess.run(cmd, capture_output=True, text=True, timeout=600,
+                       cwd=repo)
+    assert r.returncode == 0, r.stderr[-3000:]
+    import json
+    line = [ln for ln in r.stdout.splitlines()
+            if ln.startswith("{")][-1]
+    d = json.loads(line)
+    assert d["n_gpus"] == 2
+    assert d["config"]["global_batch"] == 8
