@@ -241,6 +241,11 @@ class Batch(Data):
         batch["ptr"] = ptr
         batch.num_nodes = n_total
         batch["num_graphs_"] = len(data_list)
+        ei = batch.get("edge_index")
+        if ei is not None and ei.numel() > 0 and not ei.is_cuda:
+            dst = ei[1]
+            batch["edges_sorted_"] = bool(
+                (dst[1:] >= dst[:-1]).all())
         batch["edge_counts_"] = torch.as_tensor(edge_counts, device=device)
         return batch
 

@@ -44,7 +44,8 @@ def global_mean_pool(x, batch, size=None):
 
 
 def global_add_pool(x, batch, size=None):
-    return scatter(x, batch, size, "sum")
+    # the batch vector is sorted by construction -> CSR segment kernel
+    return scatter(x, batch, size, "sum", sorted_index=True)
 
 
 def global_max_pool(x, batch, size=None):

@@ -192,8 +192,8 @@ class RealAgnosticResidualInteractionBlock(nn.Module):
         self.lmax_node = lmax_node
 
     def forward(self, node_feats: torch.Tensor, edge_index: torch.Tensor,
-                edge_sh: torch.Tensor, edge_radial: torch.Tensor
-                ) -> torch.Tensor:
+                edge_sh: torch.Tensor, edge_radial: torch.Tensor,
+                edges_sorted: bool = False) -> torch.Tensor:
         src, dst = edge_index[0], edge_index[1]
         n, c, _ = node_feats.shape
         x = self.linear_up(node_feats)
@@ -201,7 +201,8 @@ class RealAgnosticResidualInteractionBlock(nn.Module):
         w = self.radial_mlp(edge_radial).view(
             -1, c, self.conv_tp.num_paths)
         mji = self.conv_tp(x_src, edge_sh, w)
-        m = scatter(mji.reshape(mji.shape[0], -1), dst, n, "sum")
+        m = scatter(mji.reshape(mji.shape[0], -1), dst, n, "sum",
+                    sorted_index=edges_sorted)
         m = m.view(n, c, -1) / self.avg_num_neighbors
         out = self.linear(m)
         # skip: per-l linear on the input, padded to lmax_out

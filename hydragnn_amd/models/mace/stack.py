@@ -154,7 +154,9 @@ class MACEStack(Base):
             if node_feats.shape[-1] < want:
                 node_feats = torch.nn.functional.pad(
                     node_feats, (0, want - node_feats.shape[-1]))
-            m = inter(node_feats, data.edge_index, edge_sh, edge_radial)
+            m = inter(node_feats, data.edge_index, edge_sh, edge_radial,
+                      edges_sorted=bool(data.get("edges_sorted_",
+                                                 False)))
             node_feats = prod(m, elem, sc=m)
             for ihead in range(self.num_heads):
                 r = self.readouts[ihead][ilayer](node_feats)

@@ -98,44 +98,53 @@ __global__ void etp_general_kernel(
 
 // out[e,b] = sum_c sum over entries(coef, a, b, g, o) of
 //            coef * A[e,c,a] * C[e,c,g] * D[e,c,o]
-// (the B-slot gradient: reduce over channels).
+// (the B-slot gradient: reduce over channels).  Same structure as
+// etp_general: one thread per (e, c), rows + db-accumulator in a
+// per-thread LDS slice, uniform entry walk; the channel reduction is
+// db fp32 atomicAdds per thread (db <= 12, light contention).
 template <typename T>
 __global__ void etp_reduce_kernel(
     const T* __restrict__ A, const T* __restrict__ C,
     const T* __restrict__ D, float* __restrict__ out,
-    const int4* __restrict__ entries,   // (a, b, g, o) any order
+    const int4* __restrict__ entries,
     const float* __restrict__ coefs, int n_ent,
     long E, int nch, int da, int db, int dg, int do_) {
-  // one wave (64 lanes) per edge, lane strides channels
-  __shared__ float lds[4][64 * 12];  // per-wave per-lane db-acc (db<=12)
-  int wave = threadIdx.x / 64;
-  int lane = threadIdx.x % 64;
-  long e = (long)blockIdx.x * 4 + wave;
-  bool active = e < E;
-  float* my = &lds[wave][lane * db];
-  for (int b = 0; b < db; ++b) my[b] = 0.f;
-  for (int c = lane; active && c < nch; c += 64) {
-    long i = e * nch + c;
-    const T* a = A + i * da;
-    const T* cc = C + i * dg;
-    const T* dd = D + i * do_;
-    for (int k = 0; k < n_ent; ++k) {
-      int4 q = entries[k];
-      my[q.y] += coefs[k] * to_f32(a[q.x]) * to_f32(cc[q.z])
-                 * to_f32(dd[q.w]);
-    }
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  const int stride = ((da + dg + do_ + db) | 1);
+  float* slices = reinterpret_cast<float*>(smem);
+  int4* ent_lds = reinterpret_cast<int4*>(
+      smem + (size_t)blockDim.x * stride * 4);
+  float* coef_lds = reinterpret_cast<float*>(ent_lds + n_ent);
+  for (int k = threadIdx.x; k < n_ent; k += blockDim.x) {
+    ent_lds[k] = entries[k];
+    coef_lds[k] = coefs[k];
+  }
+  long NC = E * nch;
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  float* my = slices + (size_t)threadIdx.x * stride;
+  float* ma = my;
+  float* mc = ma + da;
+  float* md = mc + dg;
+  float* mb = md + do_;
+  if (i < NC) {
+    const T* ap = A + i * da;
+    const T* cp = C + i * dg;
+    const T* dp = D + i * do_;
+    for (int k = 0; k < da; ++k) ma[k] = to_f32(ap[k]);
+    for (int k = 0; k < dg; ++k) mc[k] = to_f32(cp[k]);
+    for (int k = 0; k < do_; ++k) md[k] = to_f32(dp[k]);
+    for (int k = 0; k < db; ++k) mb[k] = 0.f;
   }
   __syncthreads();
-  // tree-reduce the 64 per-lane slices
-  for (int off = 32; off >= 1; off >>= 1) {
-    if (lane < off) {
-      float* other = &lds[wave][(lane + off) * db];
-      for (int b = 0; b < db; ++b) my[b] += other[b];
+  if (i < NC) {
+    for (int k = 0; k < n_ent; ++k) {
+      int4 q = ent_lds[k];
+      mb[q.y] += coef_lds[k] * ma[q.x] * mc[q.z] * md[q.w];
     }
-    __syncthreads();
-  }
-  if (active && lane == 0) {
-    for (int b = 0; b < db; ++b) out[e * db + b] = my[b];
+    long e = i / nch;
+    for (int b = 0; b < db; ++b) {
+      atomicAdd(&out[e * db + b], mb[b]);
+    }
   }
 }
 
@@ -189,14 +198,19 @@ torch::Tensor etp_reduce(torch::Tensor A, torch::Tensor C, torch::Tensor D,
   TORCH_CHECK(db <= 12, "etp_reduce db limit");
   auto out = torch::zeros({E, db}, A.options().dtype(torch::kFloat));
   if (E == 0) return out.to(A.scalar_type());
-  long blocks = (E + 3) / 4;
   int n_ent = entries.size(0);
+  int block = 256;
+  int stride = (da + dg + do_ + (int)db) | 1;
+  size_t lds_bytes = (size_t)block * stride * 4 + n_ent * 20;
+  TORCH_CHECK(lds_bytes <= 150 * 1024, "etp_reduce LDS budget exceeded");
+  long blocks = (E * nch + block - 1) / block;
   AT_DISPATCH_FLOATING_TYPES_AND2(
       at::ScalarType::BFloat16, at::ScalarType::Half, A.scalar_type(),
       "etp_reduce", [&] {
         hipLaunchKernelGGL(
-            etp_reduce_kernel<scalar_t>, dim3(blocks), dim3(256), 0,
-            etp_stream(), A.data_ptr<scalar_t>(), C.data_ptr<scalar_t>(),
+            etp_reduce_kernel<scalar_t>, dim3(blocks), dim3(block),
+            lds_bytes, etp_stream(), A.data_ptr<scalar_t>(),
+            C.data_ptr<scalar_t>(),
             D.data_ptr<scalar_t>(), out.data_ptr<float>(),
             reinterpret_cast<const int4*>(entries.data_ptr<int>()),
             coefs.data_ptr<float>(), n_ent, E, nch, da, (int)db, dg, do_);

@@ -220,6 +220,8 @@ def radius_graph_pbc(
             keep = pos_in_seg < max_num_neighbors
         src, dst, sh = src[keep], dst[keep], sh[keep]
 
+    order = np.lexsort((src, dst))  # dst-major for the CSR fast path
+    src, dst, sh = src[order], dst[order], sh[order]
     edge_index = torch.from_numpy(np.stack([src, dst])).long().to(device)
     edge_shifts = torch.from_numpy(sh).to(dtype).to(device)
     return edge_index, edge_shifts

@@ -57,12 +57,26 @@ def gather(src: torch.Tensor, index: torch.Tensor) -> torch.Tensor:
 # ---------------------------------------------------------------------------
 # scatter: out[i] = reduce_e{index[e]==i} src[e]
 # ---------------------------------------------------------------------------
+def _rowptr_from_sorted(index: torch.Tensor, dim_size: int):
+    """CSR row pointer from a sorted index vector, with no
+    device-to-host sync (hipGraph-capturable)."""
+    counts = torch.zeros(dim_size, dtype=torch.long, device=index.device)
+    counts.index_add_(0, index, torch.ones_like(index))
+    rowptr = torch.zeros(dim_size + 1, dtype=torch.long,
+                         device=index.device)
+    torch.cumsum(counts, 0, out=rowptr[1:])
+    return rowptr
+
+
 class _ScatterSum(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, src, index, dim_size):
+    def forward(ctx, src, index, dim_size, sorted_index=False):
         ctx.save_for_backward(index)
         if _use_hip(src):
             ext = get_extension(required=True)
+            if sorted_index and hasattr(ext, "segment_sum_csr"):
+                rowptr = _rowptr_from_sorted(index, dim_size)
+                return ext.segment_sum_csr(src.contiguous(), rowptr)
             return ext.scatter_sum_fwd(src.contiguous(), index, dim_size)
         out = src.new_zeros((dim_size,) + src.shape[1:])
         out.index_add_(0, index, src)
@@ -71,7 +85,7 @@ class _ScatterSum(torch.autograd.Function):
     @staticmethod
     def backward(ctx, grad_out):
         (index,) = ctx.saved_tensors
-        return gather(grad_out, index), None, None
+        return gather(grad_out, index), None, None, None
 
 
 class _ScatterMean(torch.autograd.Function):
@@ -160,6 +174,7 @@ def scatter(
     dim_size: Optional[int] = None,
     reduce: str = "sum",
     dim: int = 0,
+    sorted_index: bool = False,
 ) -> torch.Tensor:
     """Segment reduction along dim 0.
 
@@ -172,10 +187,11 @@ def scatter(
     if index.dtype != torch.long:
         index = index.long()
     if src.dim() == 1:
-        out = scatter(src.unsqueeze(1), index, dim_size, reduce)
+        out = scatter(src.unsqueeze(1), index, dim_size, reduce,
+                      sorted_index=sorted_index)
         return out.squeeze(1)
     if reduce in ("sum", "add"):
-        return _ScatterSum.apply(src, index, dim_size)
+        return _ScatterSum.apply(src, index, dim_size, sorted_index)
     if reduce == "mean":
         return _ScatterMean.apply(src, index, dim_size)
     if reduce in ("max", "amax"):
