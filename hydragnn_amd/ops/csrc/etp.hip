@@ -141,10 +141,20 @@ __global__ void etp_reduce_kernel(
       int4 q = ent_lds[k];
       mb[q.y] += coef_lds[k] * ma[q.x] * mc[q.z] * md[q.w];
     }
-    long e = i / nch;
+  }
+  long e = i / nch;
+  if (nch % 64 == 0) {
+    // a wave spans exactly one edge's channels: shuffle-reduce each b
+    // across the 64 lanes, then ONE atomic per b per wave
     for (int b = 0; b < db; ++b) {
-      atomicAdd(&out[e * db + b], mb[b]);
+      float v = (i < NC) ? mb[b] : 0.f;
+      for (int off = 32; off >= 1; off >>= 1)
+        v += __shfl_down(v, off, 64);
+      if ((threadIdx.x % 64) == 0 && i < NC)
+        atomicAdd(&out[e * db + b], v);
     }
+  } else if (i < NC) {
+    for (int b = 0; b < db; ++b) atomicAdd(&out[e * db + b], mb[b]);
   }
 }
 

@@ -74,7 +74,9 @@ class _ScatterSum(torch.autograd.Function):
         ctx.save_for_backward(index)
         if _use_hip(src):
             ext = get_extension(required=True)
-            if sorted_index and hasattr(ext, "segment_sum_csr"):
+            import os
+            if sorted_index and hasattr(ext, "segment_sum_csr") and \
+                    os.environ.get("HYDRAGNN_CSR_SCATTER", "1") == "1":
                 rowptr = _rowptr_from_sorted(index, dim_size)
                 return ext.segment_sum_csr(src.contiguous(), rowptr)
             return ext.scatter_sum_fwd(src.contiguous(), index, dim_size)
