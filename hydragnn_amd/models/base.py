@@ -398,30 +398,36 @@ class Base(Module):
             conv_args["batch"] = data.get("batch")
             return x, pos, conv_args
         if self.use_global_attn:
-            conv_args["batch"] = data.get("batch")
-            dt = self.pos_emb.weight.dtype
-            pe = data.get("pe")
-            if pe is None:
-                raise ValueError(
-                    "GPS global attention requires positional encodings "
-                    "(data.pe); add them in preprocessing "
-                    "(hydragnn_amd.preprocess.add_laplacian_pe)")
-            h = self.pos_emb(pe.to(dt))
-            if self.input_dim:
-                h = self.node_lin(
-                    torch.cat([self.node_emb(x.to(dt)), h], dim=1))
-            if self.is_edge_model:
-                rel_pe = data.get("rel_pe")
-                if rel_pe is None:
-                    src, dst = data.edge_index[0], data.edge_index[1]
-                    rel_pe = (pe[dst] - pe[src]).abs()
-                e = self.rel_pos_emb(rel_pe.to(dt))
-                if self.use_edge_attr:
-                    e = self.edge_lin(torch.cat(
-                        [self.edge_emb(conv_args["edge_attr"]), e], dim=1))
-                conv_args["edge_attr"] = e
+            h, conv_args = self._gps_encode(data, x, conv_args)
             return h, pos, conv_args
         return x, pos, conv_args
+
+    def _gps_encode(self, data, x, conv_args):
+        """GPS positional/feature encoders (reference Base.py:523-557);
+        reusable by stacks that override _embedding."""
+        conv_args["batch"] = data.get("batch")
+        dt = self.pos_emb.weight.dtype
+        pe = data.get("pe")
+        if pe is None:
+            raise ValueError(
+                "GPS global attention requires positional encodings "
+                "(data.pe); add them in preprocessing "
+                "(hydragnn_amd.preprocess.add_laplacian_pe)")
+        h = self.pos_emb(pe.to(dt))
+        if self.input_dim:
+            h = self.node_lin(
+                torch.cat([self.node_emb(x.to(dt)), h], dim=1))
+        if self.is_edge_model:
+            rel_pe = data.get("rel_pe")
+            if rel_pe is None:
+                src, dst = data.edge_index[0], data.edge_index[1]
+                rel_pe = (pe[dst] - pe[src]).abs()
+            e = self.rel_pos_emb(rel_pe.to(dt))
+            if self.use_edge_attr:
+                e = self.edge_lin(torch.cat(
+                    [self.edge_emb(conv_args["edge_attr"]), e], dim=1))
+            conv_args["edge_attr"] = e
+        return h, conv_args
 
     def _apply_graph_conditioning(self, x, batch, data):
         if not self.use_graph_attr_conditioning:

@@ -134,6 +134,8 @@ class SCFStack(Base):
         x = data.x
         if x is not None and not torch.is_floating_point(x):
             x = x.float()
+        if self.use_global_attn and not self.is_equivariant_attn:
+            x, conv_args = self._gps_encode(data, x, conv_args)
         return x, data.pos, conv_args
 
     def __str__(self):
