@@ -100,6 +100,14 @@ def create_dataloaders(trainset, valset, testset, batch_size: int,
             return TorchDataLoader(ds, batch_sampler=bs,
                                    collate_fn=_collate,
                                    num_workers=num_workers)
+        if oversampling and shuffle and not use_dist:
+            from torch.utils.data import RandomSampler
+            sampler = RandomSampler(
+                ds, replacement=True,
+                num_samples=num_samples or len(ds))
+            return TorchDataLoader(ds, batch_size=batch_size,
+                                   sampler=sampler, collate_fn=_collate,
+                                   num_workers=num_workers)
         if use_dist:
             sampler = DistributedSampler(ds, shuffle=shuffle and
                                          sampler_shuffle)

@@ -213,10 +213,32 @@ def set_reshard_after_backward(model, enabled: bool) -> bool:
     return done
 
 
+def deepspeed_model_wrapper(model, optimizer, config):
+    """DeepSpeed initialization (reference distributed.py:512-541).
+    deepspeed is optional; raises with guidance when absent — the
+    MI355X-native paths are DDP / FSDP over RCCL."""
+    try:
+        import deepspeed
+    except ImportError as e:
+        raise ImportError(
+            "DeepSpeed is not installed in this image; use the DDP/FSDP "
+            "paths (HYDRAGNN_USE_FSDP) which cover ZeRO-style sharding "
+            "natively over RCCL.") from e
+    from ..config.config_utils import parse_deepspeed_config
+    ds_config = parse_deepspeed_config(config)
+    engine, optimizer, _, _ = deepspeed.initialize(
+        model=model, optimizer=optimizer, config=ds_config)
+    return engine, optimizer
+
+
 def distributed_model_wrapper(model, max_neighbours=None, verbosity: int = 0,
-                              find_unused_parameters: bool = False):
+                              find_unused_parameters: bool = False,
+                              sync_batch_norm: bool = False):
     """Reference distributed.py:489: move to device and wrap."""
     device = get_device()
+    if sync_batch_norm and dist.is_initialized() and \
+            dist.get_world_size() > 1 and device.type == "cuda":
+        model = torch.nn.SyncBatchNorm.convert_sync_batchnorm(model)
     model = model.to(device)
     if dist.is_initialized() and dist.get_world_size() >= 1:
         # EnhancedModelWrapper's second autograd pass can leave params
