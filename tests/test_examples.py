@@ -47,3 +47,13 @@ def test_multibranch_example():
 def test_multidataset_example():
     r = _run("examples/multidataset/train.py", "--num_epoch", "2")
     assert r.returncode == 0, r.stderr[-2000:]
+
+
+def test_ising_example():
+    r = _run("examples/ising/ising.py", "--num_epoch", "2")
+    assert r.returncode == 0, r.stderr[-2000:]
+
+
+def test_eam_example():
+    r = _run("examples/eam/eam.py", "--num_epoch", "2")
+    assert r.returncode == 0, r.stderr[-2000:]
