@@ -10,5 +10,6 @@ postprocess, utils.
 
 from . import data, models, ops, postprocess, preprocess, train, utils  # noqa
 from .data import Batch, Data
+from .run import run_training, run_prediction
 
 __version__ = "0.1.0"

@@ -141,3 +141,23 @@ def dataset_loading_and_splitting(config):
         trainset, valset, testset,
         batch_size=config["NeuralNetwork"]["Training"]["batch_size"],
         config=config)
+
+
+def transform_raw_data_to_serialized(config):
+    """Legacy path (reference load_data.py:458): parse a raw dataset by
+    format and write the serialized pickle used by later runs."""
+    from ..utils.datasets.rawloaders import (CFGDataset, LSMSDataset,
+                                             XYZDataset)
+    from ..utils.datasets.serializeddataset import SerializedWriter
+    fmt = config["Dataset"].get("format", "LSMS").upper()
+    cls = {"LSMS": LSMSDataset, "XYZ": XYZDataset,
+           "CFG": CFGDataset}.get(fmt)
+    if cls is None:
+        raise ValueError(f"unknown raw format {fmt}")
+    ds = cls(config)
+    out = os.path.join(os.environ.get("SERIALIZED_DATA_PATH", "."),
+                       "serialized_dataset")
+    SerializedWriter(ds, out, config["Dataset"]["name"],
+                     minmax_node_feature=ds.minmax_node_feature,
+                     minmax_graph_feature=ds.minmax_graph_feature)
+    return ds

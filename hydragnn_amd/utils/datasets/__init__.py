@@ -4,3 +4,4 @@ from .abstractrawdataset import AbstractRawDataset
 from .rawloaders import LSMSDataset, XYZDataset, CFGDataset
 from .graphstore import GraphStoreWriter, GraphStoreDataset, DistDataset
 from .serializeddataset import SerializedDataset, SerializedWriter
+from .adios_compat import AdiosWriter, AdiosDataset
