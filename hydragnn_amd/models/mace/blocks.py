@@ -174,7 +174,7 @@ class RealAgnosticResidualInteractionBlock(nn.Module):
         self.avg_num_neighbors = avg_num_neighbors
         self.linear_up = IrrepsLinear(num_channels, num_channels, lmax_node)
         self.conv_tp = EdgeTensorProduct(lmax_node, lmax_edge, lmax_out)
-        from ...ops.splitk_linear import SplitKLinear
+        from ...ops.mfma_linear import MFMALinear as SplitKLinear
         hidden = radial_mlp or [64, 64, 64]
         mods = []
         prev = radial_dim

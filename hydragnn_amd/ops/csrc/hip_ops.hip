@@ -605,6 +605,10 @@ std::vector<torch::Tensor> radius_pairs(torch::Tensor pos, torch::Tensor batch,
   return {src, dst, dist};
 }
 
+// defined in mfma_linear.hip
+torch::Tensor mfma_linear(torch::Tensor A, torch::Tensor B,
+                          c10::optional<torch::Tensor> bias, bool trans_b);
+
 // defined in etp.hip
 torch::Tensor etp_general(torch::Tensor A, torch::Tensor B, torch::Tensor C,
                           torch::Tensor entries, torch::Tensor coefs,
@@ -615,6 +619,10 @@ torch::Tensor etp_reduce(torch::Tensor A, torch::Tensor C, torch::Tensor D,
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("etp_general", &etp_general, "fused ETP contraction (HIP)");
+  m.def("mfma_linear", &mfma_linear, "bf16 MFMA linear (HIP)",
+        pybind11::arg("A"), pybind11::arg("B"),
+        pybind11::arg("bias") = pybind11::none(),
+        pybind11::arg("trans_b") = true);
   m.def("etp_reduce", &etp_reduce, "fused ETP channel-reduce (HIP)");
   m.def("gather_fwd", &gather_fwd, "gather rows (HIP)");
   m.def("scatter_sum_fwd", &scatter_sum_fwd, "scatter-add (HIP)");

@@ -10,7 +10,8 @@ ext_modules = [
     CUDAExtension(
         name="hydragnn_amd.ops._hip_ops",
         sources=["hydragnn_amd/ops/csrc/hip_ops.hip",
-                 "hydragnn_amd/ops/csrc/etp.hip"],
+                 "hydragnn_amd/ops/csrc/etp.hip",
+                 "hydragnn_amd/ops/csrc/mfma_linear.hip"],
         extra_compile_args={
             "cxx": ["-O3"],
             "nvcc": ["-O3", "--offload-arch=gfx950"],
