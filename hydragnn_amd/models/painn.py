@@ -21,6 +21,7 @@ from ..ops import (
     scatter,
     sinc_basis,
 )
+from ..ops.mfma_linear import MFMALinear
 from .base import Base
 
 
@@ -31,8 +32,8 @@ class PainnMessage(nn.Module):
         self.cutoff = cutoff
         self.num_radial = num_radial
         self.scalar_message_mlp = nn.Sequential(
-            nn.Linear(node_size, node_size), nn.SiLU(),
-            nn.Linear(node_size, node_size * 3))
+            MFMALinear(node_size, node_size), nn.SiLU(),
+            MFMALinear(node_size, node_size * 3))
         self.filter_layer = nn.Linear(num_radial, node_size * 3)
 
     def forward(self, node_scalar, node_vector, edge_index, edge_diff,
@@ -69,8 +70,8 @@ class PainnUpdate(nn.Module):
         self.update_U = nn.Linear(node_size, node_size, bias=False)
         self.update_V = nn.Linear(node_size, node_size, bias=False)
         self.update_mlp = nn.Sequential(
-            nn.Linear(node_size * 2, node_size), nn.SiLU(),
-            nn.Linear(node_size, node_size * 3))
+            MFMALinear(node_size * 2, node_size), nn.SiLU(),
+            MFMALinear(node_size, node_size * 3))
 
     def forward(self, node_scalar, node_vector):
         Uv = self.update_U(node_vector)

@@ -20,6 +20,7 @@ from ..ops import (
     scatter,
     sinc_basis,
 )
+from ..ops.mfma_linear import MFMALinear
 from .base import Base, BatchNormNode
 from .layers import DegreeScalerAggregation
 from .painn import PainnUpdate
@@ -34,8 +35,8 @@ class PNAEqMessage(nn.Module):
         self.cutoff = cutoff
         self.num_radial = num_radial
         self.scalar_message_mlp = nn.Sequential(
-            nn.Linear(node_size, node_size), nn.SiLU(),
-            nn.Linear(node_size, node_size * 3))
+            MFMALinear(node_size, node_size), nn.SiLU(),
+            MFMALinear(node_size, node_size * 3))
         self.filter_layer = nn.Linear(num_radial, node_size * 3)
         self.aggr = DegreeScalerAggregation(aggregators, scalers, deg)
         self.scalar_proj = nn.Linear(
