@@ -71,7 +71,9 @@ def run_training(config: Union[str, dict], dataset=None,
     train_validate_test(model, optimizer, *loaders, writer=writer,
                         scheduler=scheduler,
                         config=config["NeuralNetwork"],
-                        log_name=log_name, verbosity=verbosity)
+                        log_name=log_name, verbosity=verbosity,
+                        create_plots=config.get("Visualization", {})
+                        .get("create_plots", False))
     save_model(model, optimizer, log_name)
     return model, config
 

@@ -390,4 +390,27 @@ def train_validate_test(
             break
 
     timer.stop()
+    # final eval + plots (reference train_validate_test.py:470-520)
+    if create_plots:
+        err, tasks_err, tv, pv = test(test_loader, model, verbosity,
+                                      precision)
+        rank = dist.get_rank() if dist.is_initialized() else 0
+        if rank == 0 and tv and tv[0].numel() > 0:
+            from ..postprocess.visualizer import Visualizer
+            viz = Visualizer(log_name,
+                             num_heads=len(tv))
+            viz.create_scatter_plots(tv, pv)
+            viz.create_error_histograms(tv, pv)
+    if os.getenv("HYDRAGNN_DUMP_TESTDATA"):
+        err, tasks_err, tv, pv = test(test_loader, model, verbosity,
+                                      precision)
+        rank = dist.get_rank() if dist.is_initialized() else 0
+        if rank == 0:
+            import numpy as np
+            os.makedirs(f"logs/{log_name}", exist_ok=True)
+            np.savez(f"logs/{log_name}/testdata.npz",
+                     **{f"true_{i}": t.cpu().numpy()
+                        for i, t in enumerate(tv)},
+                     **{f"pred_{i}": p_.cpu().numpy()
+                        for i, p_ in enumerate(pv)})
     tr.save(f"logs/{log_name}")
