@@ -94,3 +94,29 @@ def test_mfma_throughput_beats_blaslt():
     t_blas = bench(lambda: A @ W.t())
     print(f"mfma {t_mfma:.1f}us vs hipBLASLt {t_blas:.1f}us")
     assert t_mfma < t_blas * 1.5, (t_mfma, t_blas)
+
+
+def test_hipgraph_capture_matches_eager():
+    """Guard against silent capture corruption: the captured step's
+    losses track an eager run of the same seed/model/batch."""
+    import os
+    import sys
+    sys.path.insert(0, os.path.dirname(os.path.dirname(
+        os.path.abspath(__file__))))
+    import bench as bench_mod
+
+    losses = {}
+    for mode in ("eager", "graph"):
+        os.environ["HYDRAGNN_HIPGRAPH"] = "0" if mode == "eager" else "1"
+        torch.manual_seed(123)
+        model, batch, step = bench_mod.build_model_and_batch(
+            device="cuda:0", local_batch=16, seed=99)
+        ls = []
+        for _ in range(5):
+            ls.append(float(step().detach().float().cpu()))
+        losses[mode] = ls
+        del model, batch, step
+        torch.cuda.empty_cache()
+    os.environ.pop("HYDRAGNN_HIPGRAPH", None)
+    for a, b in zip(losses["eager"], losses["graph"]):
+        assert abs(a - b) / max(abs(a), 1e-6) < 0.05, losses
