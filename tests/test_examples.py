@@ -57,3 +57,14 @@ def test_ising_example():
 def test_eam_example():
     r = _run("examples/eam/eam.py", "--num_epoch", "2")
     assert r.returncode == 0, r.stderr[-2000:]
+
+
+def test_lsms_example():
+    r = _run("examples/lsms/lsms.py", "--num_epoch", "2")
+    assert r.returncode == 0, r.stderr[-2000:]
+
+
+def test_open_catalyst_example():
+    r = _run("examples/open_catalyst/open_catalyst.py",
+             "--num_epoch", "2", "--num_samples", "12")
+    assert r.returncode == 0, r.stderr[-2000:]
