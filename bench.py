@@ -147,23 +147,34 @@ def build_model_and_batch(device="cuda:0", local_batch=LOCAL_BATCH,
         p.grad = flat_grad[off:off + p.numel()].view_as(p)
         off += p.numel()
 
-    def eager_step():
+    def fwd_bwd():
         flat_grad.zero_()
         batch.pos.requires_grad_(True)
         with autocast:
             pred = model(batch)
             loss, _ = m.energy_force_loss(pred, batch, create_graph=True)
         loss.backward()
+        return loss
+
+    def tail():
+        # kept OUTSIDE the captured graph: collectives inside hipGraph
+        # capture are avoided by design (capture-safety over the last
+        # few launches), and the AdamW tail is a handful of foreach
+        # kernels.
         if world > 1:
             dist.all_reduce(flat_grad)
             flat_grad.div_(world)
         optimizer.step()
+
+    def eager_step():
+        loss = fwd_bwd()
+        tail()
         return loss
 
     step = eager_step
     if use_graph:
-        # Capture the whole fwd+double-bwd+opt step in a hipGraph:
-        # the MACE step is launch-bound at MD17 batch sizes, so replay
+        # Capture the fwd + double-backward in a hipGraph: the MACE
+        # step is launch-bound at MD17 molecule sizes, so replay
         # collapses thousands of small launches into one dispatch.
         try:
             s = torch.cuda.Stream()
@@ -174,10 +185,11 @@ def build_model_and_batch(device="cuda:0", local_batch=LOCAL_BATCH,
             torch.cuda.current_stream().wait_stream(s)
             g = torch.cuda.CUDAGraph()
             with torch.cuda.graph(g):
-                static_loss = eager_step()
+                static_loss = fwd_bwd()
 
             def graph_step():
                 g.replay()
+                tail()
                 return static_loss
 
             # sanity: replay twice and require a finite loss
