@@ -14,7 +14,7 @@ the HIP segment-reduction kernels can run without re-sorting per step.
 from __future__ import annotations
 
 import copy
-from typing import Any, Dict, Iterable, List, Optional, Sequence
+from typing import Any, Dict, Optional, Sequence
 
 import torch
 

@@ -24,7 +24,7 @@ from typing import Optional
 import torch
 from torch import nn
 
-from ..models.mace.o3 import IrrepsLinear, dim, wigner_3j
+from ..models.mace.o3 import IrrepsLinear, dim
 from ..models.mace.blocks import EdgeTensorProduct
 from ..ops import (
     bessel_basis,

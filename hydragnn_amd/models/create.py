@@ -13,7 +13,6 @@ import torch
 
 from ..ops import scatter
 from ..utils.distributed import get_device
-from .base import Base
 from .stacks import CGCNNStack, GATStack, GINStack, MFCStack, PNAStack, SAGEStack
 
 PRECISION_MAP = {

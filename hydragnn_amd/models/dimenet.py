@@ -18,7 +18,7 @@ import torch
 from torch import nn
 
 from ..ops import gather, get_edge_vectors_and_lengths, scatter
-from .base import Base, BatchNormNode
+from .base import Base
 from .pna_plus import BesselBasisLayer
 
 

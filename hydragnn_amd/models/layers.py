@@ -12,7 +12,6 @@ Uniform stack-facing wrapper signature (see base.Base):
 
 from __future__ import annotations
 
-import math
 from typing import List, Optional
 
 import torch

@@ -28,7 +28,7 @@ from .blocks import (
     RadialEmbeddingBlock,
     RealAgnosticResidualInteractionBlock,
 )
-from .o3 import IrrepsLinear, dim
+from .o3 import dim
 
 NUM_ELEMENTS = 118
 

@@ -11,8 +11,6 @@ decoder over the branch group — RCCL over xGMI on MI355X.
 
 from __future__ import annotations
 
-from typing import Optional
-
 import torch
 import torch.distributed as dist
 from torch import nn

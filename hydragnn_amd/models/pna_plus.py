@@ -13,8 +13,8 @@ from typing import List, Optional
 import torch
 from torch import nn
 
-from ..ops import bessel_basis, gather, get_edge_vectors_and_lengths
-from .base import Base, BatchNormNode
+from ..ops import gather, get_edge_vectors_and_lengths
+from .base import Base
 from .layers import DegreeScalerAggregation
 
 

@@ -14,7 +14,7 @@ dense-einsum path that autograd differentiates directly.
 
 from __future__ import annotations
 
-from typing import Dict, Optional, Tuple
+from typing import Dict, Tuple
 
 import torch
 

@@ -73,12 +73,19 @@ class _ScatterSum(torch.autograd.Function):
     def forward(ctx, src, index, dim_size, sorted_index=False):
         ctx.save_for_backward(index)
         if _use_hip(src):
-            ext = get_extension(required=True)
             import os
+            ext = get_extension(required=True)
             if sorted_index and hasattr(ext, "segment_sum_csr") and \
                     os.environ.get("HYDRAGNN_CSR_SCATTER", "1") == "1":
                 rowptr = _rowptr_from_sorted(index, dim_size)
                 return ext.segment_sum_csr(src.contiguous(), rowptr)
+            if os.environ.get("HYDRAGNN_DETERMINISTIC", "0") == "1":
+                # order-independent accumulation: sort once, CSR reduce
+                # (atomicAdd float accumulation is order-dependent)
+                perm = torch.argsort(index, stable=True)
+                rowptr = _rowptr_from_sorted(index[perm], dim_size)
+                return ext.segment_sum_csr(
+                    src.index_select(0, perm).contiguous(), rowptr)
             return ext.scatter_sum_fwd(src.contiguous(), index, dim_size)
         out = src.new_zeros((dim_size,) + src.shape[1:])
         out.index_add_(0, index, src)

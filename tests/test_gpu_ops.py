@@ -105,3 +105,22 @@ def test_training_step_gpu():
     model, config, loaders = run_training(
         "GIN", heads=("graph",), num_samples=16, num_epoch=2, use_gpu=True)
     assert next(model.parameters()).is_cuda
+
+
+def test_deterministic_scatter_mode():
+    """HYDRAGNN_DETERMINISTIC=1: repeated unsorted scatters are
+    bitwise identical (survey hard-part: deterministic force accum)."""
+    import os
+    torch.manual_seed(0)
+    E, N, F = 50000, 500, 64
+    src = torch.randn(E, F, device="cuda")
+    idx = torch.randint(0, N, (E,), device="cuda")
+    os.environ["HYDRAGNN_DETERMINISTIC"] = "1"
+    try:
+        outs = [scatter(src, idx, N, "sum") for _ in range(3)]
+        assert torch.equal(outs[0], outs[1])
+        assert torch.equal(outs[0], outs[2])
+        ref = _cpu_ref_scatter(src, idx, N, "sum")
+        assert torch.allclose(outs[0].cpu(), ref, atol=1e-4)
+    finally:
+        os.environ.pop("HYDRAGNN_DETERMINISTIC")
