@@ -68,7 +68,7 @@ MODEL_CONFIG = {
     "distance_transform": None,
 }
 
-LOCAL_BATCH = 512        # graphs per GPU per step (weak scaling)
+LOCAL_BATCH = int(os.environ.get("HYDRAGNN_BENCH_BATCH", "512"))
 PRECISION = "bf16"
 
 

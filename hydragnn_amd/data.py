@@ -7,8 +7,9 @@ We keep the same attribute conventions (``x``, ``pos``, ``edge_index``,
 ``edge_attr``, ``y``, ``batch``) so configs and user code translate
 directly, but the implementation is self-contained: no PyG dependency,
 collation is a single pass with pinned-memory-friendly contiguous
-tensors, and batches carry a CSR ``rowptr`` over destination nodes so
-the HIP segment-reduction kernels can run without re-sorting per step.
+tensors, and collation records whether edges are destination-sorted
+(``edges_sorted_``) so the deterministic CSR segment-reduction kernels
+can be used without re-sorting per step.
 """
 
 from __future__ import annotations
@@ -178,11 +179,10 @@ class Batch(Data):
     """A batch of graphs collated into one big disconnected graph.
 
     Adds:
-      batch   [N]    graph id per node
-      ptr     [B+1]  node offsets per graph
-      rowptr  [N+1]  CSR row pointer over *destination* nodes of
-                     edge_index (edges sorted by dst at collation) for
-                     the HIP segment-reduce kernels.
+      batch          [N]    graph id per node
+      ptr            [B+1]  node offsets per graph
+      edges_sorted_  bool   whether edge_index is dst-sorted (enables
+                            the CSR segment-reduce fast path)
     """
 
     @classmethod

@@ -116,10 +116,21 @@ def _head_indices_node_or_mixed(m, data):
 # ---------------------------------------------------------------------------
 @torch.no_grad()
 def reduce_values_ranks(local_tensor):
-    if dist.is_initialized() and dist.get_world_size() > 1:
-        dist.all_reduce(local_tensor, op=dist.ReduceOp.SUM)
-        local_tensor = local_tensor / dist.get_world_size()
-    return local_tensor
+    """Cross-rank mean; HYDRAGNN_AGGR_BACKEND=mpi routes the tiny
+    metric reductions through MPI instead of RCCL (the reference's
+    at-scale tuning) when mpi4py is available."""
+    if not (dist.is_initialized() and dist.get_world_size() > 1):
+        return local_tensor
+    if os.getenv("HYDRAGNN_AGGR_BACKEND") == "mpi":
+        try:
+            from mpi4py import MPI
+            out = MPI.COMM_WORLD.allreduce(
+                local_tensor.detach().cpu().numpy(), op=MPI.SUM)
+            return torch.as_tensor(out) / dist.get_world_size()
+        except ImportError:
+            pass
+    dist.all_reduce(local_tensor, op=dist.ReduceOp.SUM)
+    return local_tensor / dist.get_world_size()
 
 
 @torch.no_grad()
