@@ -193,17 +193,27 @@ class RealAgnosticResidualInteractionBlock(nn.Module):
 
     def forward(self, node_feats: torch.Tensor, edge_index: torch.Tensor,
                 edge_sh: torch.Tensor, edge_radial: torch.Tensor,
-                edges_sorted: bool = False) -> torch.Tensor:
+                edges_sorted: bool = False,
+                etp_meta=None) -> torch.Tensor:
+        from ...ops.etp import _kernel_ok, etp_indexed
         src, dst = edge_index[0], edge_index[1]
         n, c, _ = node_feats.shape
         x = self.linear_up(node_feats)
-        x_src = gather(x.reshape(n, -1), src).view(-1, c, x.shape[-1])
         w = self.radial_mlp(edge_radial).view(
             -1, c, self.conv_tp.num_paths)
-        mji = self.conv_tp(x_src, edge_sh, w)
-        m = scatter(mji.reshape(mji.shape[0], -1), dst, n, "sum",
-                    sorted_index=edges_sorted)
-        m = m.view(n, c, -1) / self.avg_num_neighbors
+        if etp_meta is not None and _kernel_ok(
+                self.conv_tp.etp_table, x, edge_sh, w):
+            # fused gather + tensor product + segment sum (one kernel;
+            # gradients stay in the fused family — ops/etp.py)
+            m = etp_indexed(x, edge_sh, w, self.conv_tp.etp_table,
+                            etp_meta)
+        else:
+            x_src = gather(x.reshape(n, -1), src).view(-1, c,
+                                                       x.shape[-1])
+            mji = self.conv_tp(x_src, edge_sh, w)
+            m = scatter(mji.reshape(mji.shape[0], -1), dst, n, "sum",
+                        sorted_index=edges_sorted).view(n, c, -1)
+        m = m / self.avg_num_neighbors
         out = self.linear(m)
         # skip: per-l linear on the input, padded to lmax_out
         sc = self.skip_linear(node_feats[:, :, :dim(min(self.lmax_node,

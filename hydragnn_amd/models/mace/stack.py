@@ -136,9 +136,30 @@ class MACEStack(Base):
         h0 = self.node_embedding(one_hot)  # [N, C]
         return h0, elem, edge_sh.to(h0.dtype), edge_radial.to(h0.dtype)
 
+    def _edge_struct(self, data):
+        """Per-batch ETP metadata for the fused gather+TP+sum kernel
+        (capture-safe index arithmetic, built once per batch)."""
+        if not data.pos.is_cuda:
+            return None
+        key = "_etp_meta_"
+        cached = data.get(key)
+        if cached is not None:
+            return cached
+        from ...ops.etp import ETPMeta
+        from ...ops.scatter import _rowptr_from_sorted
+        src, dst = data.edge_index[0], data.edge_index[1]
+        n = data.pos.shape[0]
+        eid_d = torch.argsort(dst, stable=True)
+        rowptr_dst = _rowptr_from_sorted(dst[eid_d], n)
+        meta = ETPMeta(src.numel(), ai=src[eid_d], bi=eid_d, ci=eid_d,
+                       rowptr=rowptr_dst, n_a_rows=n)
+        data[key] = meta
+        return meta
+
     def forward(self, data):
         h0, elem, edge_sh, edge_radial = self._embedding(data)
         batch = data["batch"]
+        etp_meta = self._edge_struct(data)
         n = h0.shape[0]
         C = self.hidden_dim
         node_feats = h0.view(n, C, 1)
@@ -156,7 +177,8 @@ class MACEStack(Base):
                     node_feats, (0, want - node_feats.shape[-1]))
             m = inter(node_feats, data.edge_index, edge_sh, edge_radial,
                       edges_sorted=bool(data.get("edges_sorted_",
-                                                 False)))
+                                                 False)),
+                      etp_meta=etp_meta)
             node_feats = prod(m, elem, sc=m)
             for ihead in range(self.num_heads):
                 r = self.readouts[ihead][ilayer](node_feats)

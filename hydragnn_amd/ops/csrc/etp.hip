@@ -55,7 +55,9 @@ __global__ void etp_general_kernel(
     const T* __restrict__ C, T* __restrict__ out,
     const int4* __restrict__ entries,
     const float* __restrict__ coefs, int n_ent,
-    long NC, int nch, int da, int db, int dg, int do_) {
+    long NC, int nch, int da, int db, int dg, int do_,
+    const long* __restrict__ ai, const long* __restrict__ bi,
+    const long* __restrict__ ci) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   const int stride = ((da + db + dg + do_) | 1);  // odd word stride
   float* slices = reinterpret_cast<float*>(smem);
@@ -77,9 +79,13 @@ __global__ void etp_general_kernel(
   float* mo = mc + dg;
   if (i < NC) {
     long e = i / nch;
-    const T* ap = A + i * da;
-    const T* bp = B + e * db;
-    const T* cp = C + i * dg;
+    int c = (int)(i - e * nch);
+    long ea = ai ? ai[e] : e;
+    long eb = bi ? bi[e] : e;
+    long ec = ci ? ci[e] : e;
+    const T* ap = A + (ea * nch + c) * da;
+    const T* bp = B + eb * db;
+    const T* cp = C + (ec * nch + c) * dg;
     for (int k = 0; k < da; ++k) ma[k] = to_f32(ap[k]);
     for (int k = 0; k < db; ++k) mb[k] = to_f32(bp[k]);
     for (int k = 0; k < dg; ++k) mc[k] = to_f32(cp[k]);
@@ -158,23 +164,153 @@ __global__ void etp_reduce_kernel(
   }
 }
 
+// Fused gather + tensor product + segment sum:
+//   out[r, c, o] = sum_{e in rowptr[r]..rowptr[r+1]} sum_k coef_k
+//                  A[ai[e], c, a] B[bi[e], b] C[ci[e], c, g]
+// One thread per (out row, channel); the edge loop restages each
+// edge's rows into the thread's LDS slice — removes the materialized
+// per-edge message tensor, its scatter pass, and the standalone node
+// gather of the unfused pipeline.
+template <typename T>
+__global__ void etp_nodesum_kernel(
+    const T* __restrict__ A, const T* __restrict__ B,
+    const T* __restrict__ C, T* __restrict__ out,
+    const int4* __restrict__ entries,
+    const float* __restrict__ coefs, int n_ent,
+    const long* __restrict__ rowptr, long R, int nch,
+    int da, int db, int dg, int do_,
+    const long* __restrict__ ai, const long* __restrict__ bi,
+    const long* __restrict__ ci) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  const int stride = ((da + db + dg + do_) | 1);
+  float* slices = reinterpret_cast<float*>(smem);
+  int4* ent_lds = reinterpret_cast<int4*>(
+      smem + (size_t)blockDim.x * stride * 4);
+  float* coef_lds = reinterpret_cast<float*>(ent_lds + n_ent);
+  for (int k = threadIdx.x; k < n_ent; k += blockDim.x) {
+    ent_lds[k] = entries[k];
+    coef_lds[k] = coefs[k];
+  }
+  long RC = R * nch;
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  float* my = slices + (size_t)threadIdx.x * stride;
+  float* ma = my;
+  float* mb = ma + da;
+  float* mc = mb + db;
+  float* mo = mc + dg;
+  __syncthreads();
+  if (i >= RC) return;
+  long r = i / nch;
+  int c = (int)(i - r * nch);
+  for (int k = 0; k < do_; ++k) mo[k] = 0.f;
+  long lo = rowptr[r], hi = rowptr[r + 1];
+  for (long e = lo; e < hi; ++e) {
+    long ea = ai ? ai[e] : e;
+    long eb = bi ? bi[e] : e;
+    long ec = ci ? ci[e] : e;
+    const T* ap = A + (ea * nch + c) * da;
+    const T* bp = B + eb * db;
+    const T* cp = C + (ec * nch + c) * dg;
+    for (int k = 0; k < da; ++k) ma[k] = to_f32(ap[k]);
+    for (int k = 0; k < db; ++k) mb[k] = to_f32(bp[k]);
+    for (int k = 0; k < dg; ++k) mc[k] = to_f32(cp[k]);
+    for (int k = 0; k < n_ent; ++k) {
+      int4 q = ent_lds[k];
+      mo[q.w] += coef_lds[k] * ma[q.x] * mb[q.y] * mc[q.z];
+    }
+  }
+  T* op = out + i * do_;
+  for (int k = 0; k < do_; ++k) op[k] = from_f32<T>(mo[k]);
+}
+
+// etp_reduce with per-slot row indices
+template <typename T>
+__global__ void etp_reduce_idx_kernel(
+    const T* __restrict__ A, const T* __restrict__ C,
+    const T* __restrict__ D, float* __restrict__ out,
+    const int4* __restrict__ entries,
+    const float* __restrict__ coefs, int n_ent,
+    long E, int nch, int da, int db, int dg, int do_,
+    const long* __restrict__ ai, const long* __restrict__ ci,
+    const long* __restrict__ di) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  const int stride = ((da + dg + do_ + db) | 1);
+  float* slices = reinterpret_cast<float*>(smem);
+  int4* ent_lds = reinterpret_cast<int4*>(
+      smem + (size_t)blockDim.x * stride * 4);
+  float* coef_lds = reinterpret_cast<float*>(ent_lds + n_ent);
+  for (int k = threadIdx.x; k < n_ent; k += blockDim.x) {
+    ent_lds[k] = entries[k];
+    coef_lds[k] = coefs[k];
+  }
+  long NC = E * nch;
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  float* my = slices + (size_t)threadIdx.x * stride;
+  float* ma = my;
+  float* mc = ma + da;
+  float* md = mc + dg;
+  float* mb = md + do_;
+  if (i < NC) {
+    long e = i / nch;
+    int c = (int)(i - e * nch);
+    long ea = ai ? ai[e] : e;
+    long ec = ci ? ci[e] : e;
+    long ed = di ? di[e] : e;
+    const T* ap = A + (ea * nch + c) * da;
+    const T* cp = C + (ec * nch + c) * dg;
+    const T* dp = D + (ed * nch + c) * do_;
+    for (int k = 0; k < da; ++k) ma[k] = to_f32(ap[k]);
+    for (int k = 0; k < dg; ++k) mc[k] = to_f32(cp[k]);
+    for (int k = 0; k < do_; ++k) md[k] = to_f32(dp[k]);
+    for (int k = 0; k < db; ++k) mb[k] = 0.f;
+  }
+  __syncthreads();
+  if (i < NC) {
+    for (int k = 0; k < n_ent; ++k) {
+      int4 q = ent_lds[k];
+      mb[q.y] += coef_lds[k] * ma[q.x] * mc[q.z] * md[q.w];
+    }
+  }
+  long e = i / nch;
+  if (nch % 64 == 0) {
+    for (int b = 0; b < db; ++b) {
+      float v = (i < NC) ? mb[b] : 0.f;
+      for (int off = 32; off >= 1; off >>= 1)
+        v += __shfl_down(v, off, 64);
+      if ((threadIdx.x % 64) == 0 && i < NC)
+        atomicAdd(&out[e * db + b], v);
+    }
+  } else if (i < NC) {
+    for (int b = 0; b < db; ++b) atomicAdd(&out[e * db + b], mb[b]);
+  }
+}
+
 }  // namespace
 
 static hipStream_t etp_stream() {
   return at::hip::getCurrentHIPStream().stream();
 }
 
+static const long* idx_ptr(const c10::optional<torch::Tensor>& t) {
+  return t.has_value() ? t->data_ptr<long>() : nullptr;
+}
+
 torch::Tensor etp_general(torch::Tensor A, torch::Tensor B, torch::Tensor C,
                           torch::Tensor entries, torch::Tensor coefs,
-                          torch::Tensor o_ranges, long do_) {
+                          torch::Tensor o_ranges, long do_,
+                          c10::optional<torch::Tensor> ai,
+                          c10::optional<torch::Tensor> bi,
+                          c10::optional<torch::Tensor> ci,
+                          long n_rows) {
   TORCH_CHECK(A.is_cuda() && A.is_contiguous());
   TORCH_CHECK(B.is_contiguous() && C.is_contiguous());
-  long NC = A.size(0) * A.size(1);
+  long E = n_rows > 0 ? n_rows : A.size(0);
+  long NC = E * A.size(1);
   int nch = A.size(1);
   int da = A.size(2), db = B.size(1), dg = C.size(2);
   TORCH_CHECK(da <= 192 && db <= 192 && dg <= 192 && do_ <= 192,
               "etp dims exceed kernel limits");
-  auto out = torch::empty({A.size(0), A.size(1), do_}, A.options());
+  auto out = torch::empty({E, A.size(1), do_}, A.options());
   if (NC == 0) return out;
   int n_ent = entries.size(0);
   int block = 256;
@@ -193,16 +329,58 @@ torch::Tensor etp_general(torch::Tensor A, torch::Tensor B, torch::Tensor C,
             C.data_ptr<scalar_t>(), out.data_ptr<scalar_t>(),
             reinterpret_cast<const int4*>(entries.data_ptr<int>()),
             coefs.data_ptr<float>(), n_ent,
-            NC, nch, da, db, dg, (int)do_);
+            NC, nch, da, db, dg, (int)do_,
+            idx_ptr(ai), idx_ptr(bi), idx_ptr(ci));
+      });
+  return out;
+}
+
+torch::Tensor etp_nodesum(torch::Tensor A, torch::Tensor B,
+                          torch::Tensor C, torch::Tensor entries,
+                          torch::Tensor coefs, long do_,
+                          torch::Tensor rowptr,
+                          c10::optional<torch::Tensor> ai,
+                          c10::optional<torch::Tensor> bi,
+                          c10::optional<torch::Tensor> ci) {
+  TORCH_CHECK(A.is_cuda() && A.is_contiguous());
+  TORCH_CHECK(B.is_contiguous() && C.is_contiguous());
+  long R = rowptr.numel() - 1;
+  int nch = A.size(1);
+  int da = A.size(2), db = B.size(1), dg = C.size(2);
+  auto out = torch::empty({R, (long)nch, do_}, A.options());
+  long RC = R * nch;
+  if (RC == 0) return out;
+  int n_ent = entries.size(0);
+  int block = 256;
+  int stride = (da + db + dg + (int)do_) | 1;
+  size_t lds_bytes = (size_t)block * stride * 4 + n_ent * 20;
+  TORCH_CHECK(lds_bytes <= 150 * 1024, "etp LDS budget exceeded");
+  long blocks = (RC + block - 1) / block;
+  AT_DISPATCH_FLOATING_TYPES_AND2(
+      at::ScalarType::BFloat16, at::ScalarType::Half, A.scalar_type(),
+      "etp_nodesum", [&] {
+        hipLaunchKernelGGL(
+            etp_nodesum_kernel<scalar_t>, dim3(blocks), dim3(block),
+            lds_bytes, etp_stream(),
+            A.data_ptr<scalar_t>(), B.data_ptr<scalar_t>(),
+            C.data_ptr<scalar_t>(), out.data_ptr<scalar_t>(),
+            reinterpret_cast<const int4*>(entries.data_ptr<int>()),
+            coefs.data_ptr<float>(), n_ent,
+            rowptr.data_ptr<long>(), R, nch, da, db, dg, (int)do_,
+            idx_ptr(ai), idx_ptr(bi), idx_ptr(ci));
       });
   return out;
 }
 
 torch::Tensor etp_reduce(torch::Tensor A, torch::Tensor C, torch::Tensor D,
                          torch::Tensor entries, torch::Tensor coefs,
-                         long db) {
+                         long db,
+                         c10::optional<torch::Tensor> ai,
+                         c10::optional<torch::Tensor> ci,
+                         c10::optional<torch::Tensor> di,
+                         long n_rows) {
   TORCH_CHECK(A.is_cuda() && A.is_contiguous());
-  long E = A.size(0);
+  long E = n_rows > 0 ? n_rows : A.size(0);
   int nch = A.size(1);
   int da = A.size(2), dg = C.size(2), do_ = D.size(2);
   TORCH_CHECK(db <= 12, "etp_reduce db limit");
@@ -218,12 +396,13 @@ torch::Tensor etp_reduce(torch::Tensor A, torch::Tensor C, torch::Tensor D,
       at::ScalarType::BFloat16, at::ScalarType::Half, A.scalar_type(),
       "etp_reduce", [&] {
         hipLaunchKernelGGL(
-            etp_reduce_kernel<scalar_t>, dim3(blocks), dim3(block),
+            etp_reduce_idx_kernel<scalar_t>, dim3(blocks), dim3(block),
             lds_bytes, etp_stream(), A.data_ptr<scalar_t>(),
             C.data_ptr<scalar_t>(),
             D.data_ptr<scalar_t>(), out.data_ptr<float>(),
             reinterpret_cast<const int4*>(entries.data_ptr<int>()),
-            coefs.data_ptr<float>(), n_ent, E, nch, da, (int)db, dg, do_);
+            coefs.data_ptr<float>(), n_ent, E, nch, da, (int)db, dg, do_,
+            idx_ptr(ai), idx_ptr(ci), idx_ptr(di));
       });
   return out.to(A.scalar_type());
 }

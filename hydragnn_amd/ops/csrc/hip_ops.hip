@@ -612,18 +612,52 @@ torch::Tensor mfma_linear(torch::Tensor A, torch::Tensor B,
 // defined in etp.hip
 torch::Tensor etp_general(torch::Tensor A, torch::Tensor B, torch::Tensor C,
                           torch::Tensor entries, torch::Tensor coefs,
-                          torch::Tensor o_ranges, long do_);
+                          torch::Tensor o_ranges, long do_,
+                          c10::optional<torch::Tensor> ai,
+                          c10::optional<torch::Tensor> bi,
+                          c10::optional<torch::Tensor> ci, long n_rows);
+torch::Tensor etp_nodesum(torch::Tensor A, torch::Tensor B,
+                          torch::Tensor C, torch::Tensor entries,
+                          torch::Tensor coefs, long do_,
+                          torch::Tensor rowptr,
+                          c10::optional<torch::Tensor> ai,
+                          c10::optional<torch::Tensor> bi,
+                          c10::optional<torch::Tensor> ci);
 torch::Tensor etp_reduce(torch::Tensor A, torch::Tensor C, torch::Tensor D,
                          torch::Tensor entries, torch::Tensor coefs,
-                         long db);
+                         long db,
+                         c10::optional<torch::Tensor> ai,
+                         c10::optional<torch::Tensor> ci,
+                         c10::optional<torch::Tensor> di, long n_rows);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
-  m.def("etp_general", &etp_general, "fused ETP contraction (HIP)");
+  m.def("etp_general", &etp_general, "fused ETP contraction (HIP)",
+        pybind11::arg("A"), pybind11::arg("B"), pybind11::arg("C"),
+        pybind11::arg("entries"), pybind11::arg("coefs"),
+        pybind11::arg("o_ranges"), pybind11::arg("do_"),
+        pybind11::arg("ai") = pybind11::none(),
+        pybind11::arg("bi") = pybind11::none(),
+        pybind11::arg("ci") = pybind11::none(),
+        pybind11::arg("n_rows") = 0);
+  m.def("etp_nodesum", &etp_nodesum, "fused gather+TP+segment sum (HIP)",
+        pybind11::arg("A"), pybind11::arg("B"), pybind11::arg("C"),
+        pybind11::arg("entries"), pybind11::arg("coefs"),
+        pybind11::arg("do_"), pybind11::arg("rowptr"),
+        pybind11::arg("ai") = pybind11::none(),
+        pybind11::arg("bi") = pybind11::none(),
+        pybind11::arg("ci") = pybind11::none());
   m.def("mfma_linear", &mfma_linear, "bf16 MFMA linear (HIP)",
         pybind11::arg("A"), pybind11::arg("B"),
         pybind11::arg("bias") = pybind11::none(),
         pybind11::arg("trans_b") = true);
-  m.def("etp_reduce", &etp_reduce, "fused ETP channel-reduce (HIP)");
+  m.def("etp_reduce", &etp_reduce, "fused ETP channel-reduce (HIP)",
+        pybind11::arg("A"), pybind11::arg("C"), pybind11::arg("D"),
+        pybind11::arg("entries"), pybind11::arg("coefs"),
+        pybind11::arg("db"),
+        pybind11::arg("ai") = pybind11::none(),
+        pybind11::arg("ci") = pybind11::none(),
+        pybind11::arg("di") = pybind11::none(),
+        pybind11::arg("n_rows") = 0);
   m.def("gather_fwd", &gather_fwd, "gather rows (HIP)");
   m.def("scatter_sum_fwd", &scatter_sum_fwd, "scatter-add (HIP)");
   m.def("segment_sum_csr", &segment_sum_csr, "CSR segment sum (HIP)");

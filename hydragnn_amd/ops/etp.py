@@ -197,3 +197,205 @@ def etp_reduce(A: torch.Tensor, C: torch.Tensor, D: torch.Tensor,
         return _ETPReduce.apply(A, C, D, table)
     W = table.dense(A.device, torch.float32).to(A.dtype)
     return torch.einsum("eca,ecg,eco,abgo->eb", A, C, D, W)
+
+
+# ---------------------------------------------------------------------------
+# Indexed / CSR-fused family: gather + TP + segment-sum in one kernel.
+#
+# Primitive (positions e walk a chosen edge ordering):
+#   OUT[r, c, o] = sum_{e in rowptr[r]..rowptr[r+1]} sum_k coef_k
+#                  A[ai[e], c, a] B[bi[e], b] C[ci[e], c, g]
+# Gradients are instances of the SAME primitive with re-sorted positions
+# (see ETPMeta.grad_meta_a), so force training stays fused end to end.
+# ---------------------------------------------------------------------------
+class ETPMeta:
+    """Edge-ordering metadata: per-slot row indices + optional output
+    CSR.  bi/ci must be bijections (or None = identity); ai may be a
+    many-to-one node map."""
+
+    def __init__(self, n_positions, ai=None, bi=None, ci=None,
+                 rowptr=None, n_a_rows=None):
+        self.n_positions = int(n_positions)
+        self.ai = ai
+        self.bi = bi
+        self.ci = ci
+        self.rowptr = rowptr
+        self.n_a_rows = n_a_rows  # rows of A (for the gA output)
+        self._r_of_pos = None
+        self._grad_meta_a = None
+
+    def r_of_pos(self):
+        """Output row of each position (identity without CSR)."""
+        if self.rowptr is None:
+            return None
+        if self._r_of_pos is None:
+            counts = self.rowptr[1:] - self.rowptr[:-1]
+            self._r_of_pos = torch.repeat_interleave(
+                torch.arange(counts.numel(),
+                             device=self.rowptr.device), counts)
+        return self._r_of_pos
+
+    def grad_meta_a(self):
+        """Meta for the A-slot gradient: positions re-sorted by ai so
+        the output CSR accumulates over A's rows."""
+        if self._grad_meta_a is None:
+            assert self.ai is not None and self.n_a_rows is not None, \
+                "A-slot gradient needs ai + n_a_rows"
+            rho = torch.argsort(self.ai, stable=True)
+            rowptr_a = _rowptr_from_sorted(self.ai[rho], self.n_a_rows)
+            r = self.r_of_pos()
+            ai_new = r[rho] if r is not None else rho
+            bi_new = self.bi[rho] if self.bi is not None else rho
+            ci_new = self.ci[rho] if self.ci is not None else rho
+            m = ETPMeta(self.n_positions, ai=ai_new, bi=bi_new,
+                        ci=ci_new, rowptr=rowptr_a, n_a_rows=None)
+            self._grad_meta_a = (rho, m)
+        return self._grad_meta_a
+
+
+def _pos_index(t, idx, E):
+    if idx is None:
+        return t[:E]
+    return t.index_select(0, idx)
+
+
+def _etp_indexed_dense(A, B, C, table, meta):
+    """Differentiable fallback (CPU / fp64): gather positions, dense
+    einsum, CSR accumulate."""
+    E = meta.n_positions
+    Apos = _pos_index(A, meta.ai, E)
+    Bpos = _pos_index(B, meta.bi, E)
+    Cpos = _pos_index(C, meta.ci, E)
+    out_pos = _dense_general(Apos, Bpos, Cpos, table)
+    if meta.rowptr is None:
+        return out_pos
+    r = meta.r_of_pos()
+    R = meta.rowptr.numel() - 1
+    out = out_pos.new_zeros(R, out_pos.shape[1], out_pos.shape[2])
+    out.index_add_(0, r, out_pos)
+    return out
+
+
+class _ETPIndexed(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, A, B, C, table, meta):
+        ctx.save_for_backward(A, B, C)
+        ctx.table = table
+        ctx.meta = meta
+        ext = get_extension(required=True)
+        ent, coefs, o_ranges = table.device_tensors(A.device)
+        if meta.rowptr is not None:
+            return ext.etp_nodesum(A.contiguous(), B.contiguous(),
+                                   C.contiguous(), ent, coefs,
+                                   table.dims[3], meta.rowptr,
+                                   meta.ai, meta.bi, meta.ci)
+        return ext.etp_general(A.contiguous(), B.contiguous(),
+                               C.contiguous(), ent, coefs, o_ranges,
+                               table.dims[3], meta.ai, meta.bi,
+                               meta.ci, meta.n_positions)
+
+    @staticmethod
+    def backward(ctx, gout):
+        A, B, C = ctx.saved_tensors
+        table = ctx.table
+        meta = ctx.meta
+        gout = gout.contiguous()
+        E = meta.n_positions
+        gA = gB = gC = None
+        if ctx.needs_input_grad[0]:
+            rho, meta_a = meta.grad_meta_a()
+            gA = etp_indexed(gout, B, C, table.perm("obga"), meta_a)
+        if ctx.needs_input_grad[1]:
+            # per-position reduce then un-permute via bi (bijection)
+            gB_pos = _etp_reduce_idx(A, C, gout, table, meta)
+            if meta.bi is not None:
+                gB = gB_pos.new_zeros(B.shape[0], gB_pos.shape[1])
+                gB.index_copy_(0, meta.bi, gB_pos)
+            else:
+                gB = gB_pos
+        if ctx.needs_input_grad[2]:
+            r = meta.r_of_pos()
+            meta_c = ETPMeta(E, ai=meta.ai, bi=meta.bi,
+                             ci=r if r is not None else None,
+                             rowptr=None)
+            gC_pos = etp_indexed(A, B, gout, table.perm("abog"), meta_c)
+            if meta.ci is not None:
+                gC = gC_pos.new_zeros(C.shape)
+                gC.index_copy_(0, meta.ci, gC_pos)
+            else:
+                gC = gC_pos
+        return gA, gB, gC, None, None
+
+
+class _ETPReduceIdx(torch.autograd.Function):
+    """Per-position channel-reduced contraction with row indices:
+    out[e, b] = sum_c sum_k coef A[ai,c,a] C[ci,c,g] D[di,c,o]."""
+
+    @staticmethod
+    def forward(ctx, A, C, D, table, ai, ci, di, E):
+        ctx.save_for_backward(A, C, D)
+        ctx.table = table
+        ctx.idx = (ai, ci, di)
+        ctx.E = E
+        ext = get_extension(required=True)
+        ent, coefs, _ = table.device_tensors(A.device)
+        return ext.etp_reduce(A.contiguous(), C.contiguous(),
+                              D.contiguous(), ent, coefs,
+                              table.dims[1], ai, ci, di, E)
+
+    @staticmethod
+    def backward(ctx, gout):
+        A, C, D = ctx.saved_tensors
+        ai, ci, di = ctx.idx
+        E = ctx.E
+        table = ctx.table
+        gout = gout.contiguous()
+
+        def scatter_rows(per_pos, idx, rows):
+            if idx is None:
+                return per_pos
+            out = per_pos.new_zeros((rows,) + per_pos.shape[1:])
+            # idx may be many-to-one (node map): accumulate
+            return out.index_add_(
+                0, idx, per_pos) if idx.numel() == per_pos.shape[0] \
+                else out
+        gA = gC = gD = None
+        if ctx.needs_input_grad[0]:
+            # gA[ai,c,a] += coef gout[b] C[g] D[o]
+            m = ETPMeta(E, ai=ci, bi=None, ci=di, rowptr=None)
+            per = etp_indexed(C, gout, D, table.perm("gboa"), m)
+            gA = scatter_rows(per, ai, A.shape[0]) if ai is not None \
+                else per
+        if ctx.needs_input_grad[1]:
+            m = ETPMeta(E, ai=ai, bi=None, ci=di, rowptr=None)
+            per = etp_indexed(A, gout, D, table.perm("abog"), m)
+            gC = scatter_rows(per, ci, C.shape[0]) if ci is not None \
+                else per
+        if ctx.needs_input_grad[2]:
+            m = ETPMeta(E, ai=ai, bi=None, ci=ci, rowptr=None)
+            per = etp_indexed(A, gout, C, table.perm("abgo"), m)
+            gD = scatter_rows(per, di, D.shape[0]) if di is not None \
+                else per
+        return gA, gC, gD, None, None, None, None, None
+
+
+def _etp_reduce_idx(A, C, D, table, meta):
+    r = meta.r_of_pos()
+    di = r if r is not None else None
+    if (A.is_cuda and A.dtype in _KERNEL_DTYPES and not use_eager()
+            and table.dims[1] <= 12):
+        return _ETPReduceIdx.apply(A, C, D, table, meta.ai, meta.ci,
+                                   di, meta.n_positions)
+    E = meta.n_positions
+    Apos = _pos_index(A, meta.ai, E)
+    Cpos = _pos_index(C, meta.ci, E)
+    Dpos = D if di is None else D.index_select(0, di)
+    W = table.dense(A.device, torch.float32).to(A.dtype)
+    return torch.einsum("eca,ecg,eco,abgo->eb", Apos, Cpos, Dpos, W)
+
+
+def etp_indexed(A, B, C, table, meta: ETPMeta) -> torch.Tensor:
+    """Indexed / CSR-fused contraction (see ETPMeta)."""
+    if _kernel_ok(table, A, B, C):
+        return _ETPIndexed.apply(A, B, C, table, meta)
+    return _etp_indexed_dense(A, B, C, table, meta)
