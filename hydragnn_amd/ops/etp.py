@@ -19,6 +19,7 @@ from typing import Dict, Tuple
 import torch
 
 from ._extension import get_extension, use_eager
+from .scatter import _rowptr_from_sorted
 
 _KERNEL_DTYPES = (torch.float32, torch.bfloat16, torch.float16)
 
