@@ -248,8 +248,15 @@ class ETPMeta:
             ai_new = r[rho] if r is not None else rho
             bi_new = self.bi[rho] if self.bi is not None else rho
             ci_new = self.ci[rho] if self.ci is not None else rho
+            # the gradient instance's A is the original OUT tensor:
+            # its row count (for second-order CSR) is R (CSR mode) or
+            # the position count (per-edge mode)
+            n_rows_gA = (self.rowptr.numel() - 1
+                         if self.rowptr is not None
+                         else self.n_positions)
             m = ETPMeta(self.n_positions, ai=ai_new, bi=bi_new,
-                        ci=ci_new, rowptr=rowptr_a, n_a_rows=None)
+                        ci=ci_new, rowptr=rowptr_a,
+                        n_a_rows=n_rows_gA)
             self._grad_meta_a = (rho, m)
         return self._grad_meta_a
 
@@ -318,7 +325,7 @@ class _ETPIndexed(torch.autograd.Function):
             r = meta.r_of_pos()
             meta_c = ETPMeta(E, ai=meta.ai, bi=meta.bi,
                              ci=r if r is not None else None,
-                             rowptr=None)
+                             rowptr=None, n_a_rows=A.shape[0])
             gC_pos = etp_indexed(A, B, gout, table.perm("abog"), meta_c)
             if meta.ci is not None:
                 gC = gC_pos.new_zeros(C.shape)
@@ -363,17 +370,20 @@ class _ETPReduceIdx(torch.autograd.Function):
         gA = gC = gD = None
         if ctx.needs_input_grad[0]:
             # gA[ai,c,a] += coef gout[b] C[g] D[o]
-            m = ETPMeta(E, ai=ci, bi=None, ci=di, rowptr=None)
+            m = ETPMeta(E, ai=ci, bi=None, ci=di, rowptr=None,
+                        n_a_rows=C.shape[0])
             per = etp_indexed(C, gout, D, table.perm("gboa"), m)
             gA = scatter_rows(per, ai, A.shape[0]) if ai is not None \
                 else per
         if ctx.needs_input_grad[1]:
-            m = ETPMeta(E, ai=ai, bi=None, ci=di, rowptr=None)
+            m = ETPMeta(E, ai=ai, bi=None, ci=di, rowptr=None,
+                        n_a_rows=A.shape[0])
             per = etp_indexed(A, gout, D, table.perm("abog"), m)
             gC = scatter_rows(per, ci, C.shape[0]) if ci is not None \
                 else per
         if ctx.needs_input_grad[2]:
-            m = ETPMeta(E, ai=ai, bi=None, ci=ci, rowptr=None)
+            m = ETPMeta(E, ai=ai, bi=None, ci=ci, rowptr=None,
+                        n_a_rows=A.shape[0])
             per = etp_indexed(A, gout, C, table.perm("abgo"), m)
             gD = scatter_rows(per, di, D.shape[0]) if di is not None \
                 else per
