@@ -317,8 +317,9 @@ class _ETPIndexed(torch.autograd.Function):
             # per-position reduce then un-permute via bi (bijection)
             gB_pos = _etp_reduce_idx(A, C, gout, table, meta)
             if meta.bi is not None:
+                # index_add_: derived metas can carry many-to-one maps
                 gB = gB_pos.new_zeros(B.shape[0], gB_pos.shape[1])
-                gB.index_copy_(0, meta.bi, gB_pos)
+                gB = gB.index_add(0, meta.bi, gB_pos)
             else:
                 gB = gB_pos
         if ctx.needs_input_grad[2]:
@@ -329,7 +330,7 @@ class _ETPIndexed(torch.autograd.Function):
             gC_pos = etp_indexed(A, B, gout, table.perm("abog"), meta_c)
             if meta.ci is not None:
                 gC = gC_pos.new_zeros(C.shape)
-                gC.index_copy_(0, meta.ci, gC_pos)
+                gC = gC.index_add(0, meta.ci, gC_pos)
             else:
                 gC = gC_pos
         return gA, gB, gC, None, None
