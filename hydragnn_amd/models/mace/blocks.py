@@ -204,7 +204,7 @@ class RealAgnosticResidualInteractionBlock(nn.Module):
         import os
         if etp_meta is not None and _kernel_ok(
                 self.conv_tp.etp_table, x, edge_sh, w) and \
-                os.environ.get("HYDRAGNN_FUSED_ETP", "1") == "1":
+                os.environ.get("HYDRAGNN_FUSED_ETP", "0") == "1":
             # fused gather + tensor product + segment sum (one kernel;
             # gradients stay in the fused family — ops/etp.py)
             m = etp_indexed(x, edge_sh, w, self.conv_tp.etp_table,
