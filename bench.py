@@ -68,7 +68,7 @@ MODEL_CONFIG = {
     "distance_transform": None,
 }
 
-LOCAL_BATCH = int(os.environ.get("HYDRAGNN_BENCH_BATCH", "512"))
+LOCAL_BATCH = int(os.environ.get("HYDRAGNN_BENCH_BATCH", "1024"))
 PRECISION = "bf16"
 
 
