@@ -5,4 +5,5 @@ from .config_utils import (
     get_log_name_config,
     update_multibranch_heads,
     parse_deepspeed_config,
+    validate_equivariant_transformer_config,
 )

@@ -114,6 +114,8 @@ def update_config(config, train_loader, val_loader, test_loader):
     arch.setdefault("equivariant_attn_chunk_size", 512)
     arch.setdefault("equivariant_attn_coupling_mode", "parallel")
 
+    validate_equivariant_transformer_config(arch)
+
     batching = training.get("Batching")
     if batching is not None:
         mode = batching.get("mode", "fixed")
@@ -210,6 +212,35 @@ def update_config(config, train_loader, val_loader, test_loader):
     training.setdefault("Optimizer", {"type": "AdamW", "learning_rate": 1e-3})
     training.setdefault("precision", "fp32")
     return config
+
+
+def validate_equivariant_transformer_config(arch) -> None:
+    """Engine-specific option validation (reference
+    config_utils.py:187-250)."""
+    if arch.get("global_attn_engine") != "EquivariantTransformer":
+        return
+    mpnn_type = arch.get("mpnn_type")
+    if mpnn_type in ("SchNet", "DimeNet"):
+        if arch.get("equivariant_attn_require_tensor_coupling", True):
+            raise ValueError(
+                f"{mpnn_type} cannot provide tensor-valued local/global "
+                "coupling; set equivariant_attn_require_tensor_coupling="
+                "false")
+        if not arch.get("equivariant_attn_allow_scalar_only", False):
+            raise ValueError(
+                f"{mpnn_type} requires "
+                "equivariant_attn_allow_scalar_only=true")
+    if mpnn_type == "SchNet" and arch.get("equivariance"):
+        raise ValueError(
+            "SchNet with EquivariantTransformer cannot use coordinate "
+            "updates; set Architecture.equivariance=false")
+    if mpnn_type == "MACE" and arch.get("num_conv_layers", 0) < 2:
+        raise ValueError(
+            "MACE with EquivariantTransformer requires at least two "
+            "convolution layers (the final MACE layer is scalar-only)")
+    lmax = arch.get("equivariant_attn_lmax", 1)
+    if not (0 <= int(lmax) <= 3):
+        raise ValueError("equivariant_attn_lmax must be in [0, 3]")
 
 
 def save_config(config, log_name: str, path: str = "./logs/") -> None:

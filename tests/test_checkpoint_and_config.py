@@ -120,3 +120,27 @@ def test_head_indices_multihead():
     # indices are disjoint and cover y
     allidx = torch.cat(hidx).sort().values
     assert torch.equal(allidx, torch.arange(batch.y.shape[0]))
+
+
+def test_equivariant_transformer_config_validation():
+    from hydragnn_amd.utils.config import (
+        validate_equivariant_transformer_config)
+    # SchNet needs scalar-only opt-in
+    arch = {"global_attn_engine": "EquivariantTransformer",
+            "mpnn_type": "SchNet",
+            "equivariant_attn_require_tensor_coupling": True}
+    with pytest.raises(ValueError, match="tensor-valued"):
+        validate_equivariant_transformer_config(arch)
+    arch["equivariant_attn_require_tensor_coupling"] = False
+    with pytest.raises(ValueError, match="scalar_only"):
+        validate_equivariant_transformer_config(arch)
+    arch["equivariant_attn_allow_scalar_only"] = True
+    validate_equivariant_transformer_config(arch)  # ok now
+    # MACE needs >= 2 layers
+    with pytest.raises(ValueError, match="two"):
+        validate_equivariant_transformer_config(
+            {"global_attn_engine": "EquivariantTransformer",
+             "mpnn_type": "MACE", "num_conv_layers": 1})
+    # non-ET engine: no-op
+    validate_equivariant_transformer_config(
+        {"global_attn_engine": "gps", "mpnn_type": "SchNet"})
