@@ -205,3 +205,44 @@ def test_cfg_parser(tmp_path):
     assert len(ds) == 1
     assert ds[0].num_nodes == 2
     assert ds[0].forces.shape == (2, 3)
+
+
+def test_graphstore_two_rank_write(tmp_path):
+    """2-rank gloo write -> single reader sees both shards."""
+    import subprocess, sys, textwrap
+    script = tmp_path / "writer.py"
+    script.write_text(textwrap.dedent(f"""
+        import os, sys, torch, torch.distributed as dist
+        sys.path.insert(0, {str(os.getcwd())!r})
+        dist.init_process_group("gloo")
+        rank = dist.get_rank()
+        from hydragnn_amd.utils.datasets.graphstore import GraphStoreWriter
+        from hydragnn_amd.utils.datasets.synthetic import lj_dataset
+        ds = lj_dataset(num_samples=3, num_atoms=27, pbc=False,
+                        seed=100 + rank)
+        w = GraphStoreWriter("total", {str(tmp_path)!r})
+        w.add(ds)
+        w.save()
+        dist.destroy_process_group()
+    """))
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29551", str(script)],
+        capture_output=True, text=True, timeout=300, cwd=os.getcwd())
+    assert r.returncode == 0, r.stderr[-1500:]
+    reader = GraphStoreDataset(str(tmp_path), "total")
+    assert len(reader) == 6
+    # shard 0 and shard 1 differ (different seeds)
+    assert not torch.allclose(reader[0].pos, reader[3].pos)
+
+
+def test_scatter_max_cpu_argmax_backward():
+    """CPU scatter-max backward routes gradient to the argmax edge."""
+    src = torch.tensor([[1.0], [5.0], [3.0], [2.0]], requires_grad=True)
+    idx = torch.tensor([0, 0, 1, 1])
+    from hydragnn_amd.ops import scatter
+    out = scatter(src, idx, 2, "max")
+    assert out.flatten().tolist() == [5.0, 3.0]
+    out.sum().backward()
+    assert src.grad.flatten().tolist() == [0.0, 1.0, 1.0, 0.0]
