@@ -311,7 +311,7 @@ class _ETPIndexed(torch.autograd.Function):
         E = meta.n_positions
         gA = gB = gC = None
         if ctx.needs_input_grad[0]:
-            rho, meta_a = meta.grad_meta_a()
+            _, meta_a = meta.grad_meta_a()
             gA = etp_indexed(gout, B, C, table.perm("obga"), meta_a)
         if ctx.needs_input_grad[1]:
             # per-position reduce then un-permute via bi (bijection)

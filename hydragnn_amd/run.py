@@ -22,7 +22,8 @@ from .preprocess.load_data import dataset_loading_and_splitting
 from .train import train_validate_test, test as test_fn
 from .utils.config import get_log_name_config, save_config, update_config
 from .utils.distributed import distributed_model_wrapper, setup_ddp
-from .utils.model import get_summary_writer, load_existing_model, save_model
+from .utils.model import (get_summary_writer, load_existing_model,
+                          load_existing_model_config, save_model)
 from .utils.optimizer import select_optimizer
 from .utils.print.print_utils import setup_log
 
@@ -61,9 +62,12 @@ def run_training(config: Union[str, dict], dataset=None,
 
     model = create_model_config(config["NeuralNetwork"],
                                 verbosity=verbosity, use_gpu=use_gpu)
-    model = distributed_model_wrapper(model, verbosity=verbosity)
     optimizer = select_optimizer(
         model, config["NeuralNetwork"]["Training"]["Optimizer"])
+    # restart support (reference model.py:204-211)
+    load_existing_model_config(
+        model, config["NeuralNetwork"]["Training"], optimizer=optimizer)
+    model = distributed_model_wrapper(model, verbosity=verbosity)
     scheduler = torch.optim.lr_scheduler.ReduceLROnPlateau(
         optimizer, mode="min", factor=0.5, patience=5)
     writer = get_summary_writer(log_name)

@@ -8,5 +8,6 @@ from .train_validate_test import (
     gather_tensor_ranks,
     move_batch_to_device,
     get_autocast_and_scaler,
+    get_nbatch,
 )
 from ..models.create import resolve_precision

@@ -144,3 +144,37 @@ def test_equivariant_transformer_config_validation():
     # non-ET engine: no-op
     validate_equivariant_transformer_config(
         {"global_attn_engine": "gps", "mpnn_type": "SchNet"})
+
+
+def test_continue_startfrom(tmp_path, monkeypatch):
+    """Training.continue/startfrom restart path (reference
+    model.py:204-211)."""
+    monkeypatch.chdir(tmp_path)
+    torch.manual_seed(0)
+    import hydragnn_amd
+    cfg = base_config("GIN", heads=("graph",), num_epoch=2)
+    ds = make_deterministic_dataset(num_samples=24, num_heads_node=0)
+    model, out_cfg = hydragnn_amd.run_training(cfg, dataset=ds,
+                                               use_gpu=False)
+    from hydragnn_amd.utils.config import get_log_name_config
+    name = get_log_name_config(out_cfg)
+    cfg2 = base_config("GIN", heads=("graph",), num_epoch=1)
+    cfg2["NeuralNetwork"]["Training"]["continue"] = 1
+    cfg2["NeuralNetwork"]["Training"]["startfrom"] = name
+    # resumed run starts from the checkpointed weights
+    model2, _ = hydragnn_amd.run_training(cfg2, dataset=ds,
+                                          use_gpu=False)
+    assert model2 is not None
+
+
+def test_evalonly_and_max_batch(monkeypatch, tmp_path):
+    monkeypatch.chdir(tmp_path)
+    monkeypatch.setenv("HYDRAGNN_EVALONLY", "1")
+    model, config, loaders = run_training(
+        "GIN", heads=("graph",), num_samples=16, num_epoch=5)
+    # eval-only: no training happened -> params at init
+    monkeypatch.delenv("HYDRAGNN_EVALONLY")
+    monkeypatch.setenv("HYDRAGNN_MAX_NUM_BATCH", "1")
+    from hydragnn_amd.train import get_nbatch
+    assert get_nbatch(loaders[0]) == 1
+    monkeypatch.delenv("HYDRAGNN_MAX_NUM_BATCH")
