@@ -291,6 +291,15 @@ static hipStream_t etp_stream() {
   return at::hip::getCurrentHIPStream().stream();
 }
 
+static int etp_block_size() {
+  static int b = []() {
+    const char* e = getenv("HYDRAGNN_ETP_BLOCK");
+    int v = e ? atoi(e) : 256;
+    return (v == 64 || v == 128 || v == 256 || v == 512) ? v : 256;
+  }();
+  return b;
+}
+
 static const long* idx_ptr(const c10::optional<torch::Tensor>& t) {
   return t.has_value() ? t->data_ptr<long>() : nullptr;
 }
@@ -313,7 +322,7 @@ torch::Tensor etp_general(torch::Tensor A, torch::Tensor B, torch::Tensor C,
   auto out = torch::empty({E, A.size(1), do_}, A.options());
   if (NC == 0) return out;
   int n_ent = entries.size(0);
-  int block = 256;
+  int block = etp_block_size();
   int stride = (da + db + dg + (int)do_) | 1;
   size_t lds_bytes = (size_t)block * stride * 4 + n_ent * 20;
   TORCH_CHECK(lds_bytes <= 150 * 1024, "etp LDS budget exceeded");
@@ -351,7 +360,7 @@ torch::Tensor etp_nodesum(torch::Tensor A, torch::Tensor B,
   long RC = R * nch;
   if (RC == 0) return out;
   int n_ent = entries.size(0);
-  int block = 256;
+  int block = etp_block_size();
   int stride = (da + db + dg + (int)do_) | 1;
   size_t lds_bytes = (size_t)block * stride * 4 + n_ent * 20;
   TORCH_CHECK(lds_bytes <= 150 * 1024, "etp LDS budget exceeded");
@@ -387,7 +396,7 @@ torch::Tensor etp_reduce(torch::Tensor A, torch::Tensor C, torch::Tensor D,
   auto out = torch::zeros({E, db}, A.options().dtype(torch::kFloat));
   if (E == 0) return out.to(A.scalar_type());
   int n_ent = entries.size(0);
-  int block = 256;
+  int block = etp_block_size();
   int stride = (da + dg + do_ + (int)db) | 1;
   size_t lds_bytes = (size_t)block * stride * 4 + n_ent * 20;
   TORCH_CHECK(lds_bytes <= 150 * 1024, "etp_reduce LDS budget exceeded");
