@@ -93,3 +93,29 @@ def test_bf16_training_step_gpu():
         loss, _ = model.energy_force_loss(pred, batch, create_graph=True)
     loss.backward()
     assert torch.isfinite(loss)
+
+
+@pytest.mark.parametrize("mpnn_type", ALL_STACKS)
+def test_stack_gpu_matches_cpu(mpnn_type):
+    """HIP-path outputs match the CPU eager reference (fp32)."""
+    torch.manual_seed(0)
+    config = base_config(mpnn_type, heads=("graph",), num_epoch=1,
+                         hidden_dim=16)
+    arch = config["NeuralNetwork"]["Architecture"]
+    if mpnn_type == "MACE":
+        arch.update({"max_ell": 2, "node_max_ell": 1, "correlation": 2,
+                     "num_radial": 8})
+    ds = make_deterministic_dataset(num_samples=4, num_heads_node=0)
+    loaders = create_dataloaders(ds, ds, ds, 4, config=config)
+    config = update_config(config, *loaders)
+    model = create_model_config(config["NeuralNetwork"], use_gpu=False)
+    model.eval()
+    batch_cpu = Batch.from_data_list([d.clone() for d in ds])
+    with torch.no_grad():
+        out_cpu = model(batch_cpu)[0]
+    model_gpu = model.to("cuda")
+    batch_gpu = Batch.from_data_list([d.clone() for d in ds]).to("cuda")
+    with torch.no_grad():
+        out_gpu = model_gpu(batch_gpu)[0]
+    err = (out_cpu - out_gpu.cpu()).abs().max()
+    assert err < 1e-3, f"{mpnn_type}: CPU/GPU mismatch {err:.2e}"
