@@ -254,6 +254,12 @@ def main():
     ms_per_step = elapsed / args.steps * 1000.0
     value = graphs_per_step * args.steps / elapsed
 
+    if rank == 0 and use_cuda and \
+            os.environ.get("HYDRAGNN_BENCH_MEMSTATS"):
+        import sys
+        print(f"[bench] peak GPU memory: "
+              f"{torch.cuda.max_memory_allocated() / 2**30:.1f} GiB",
+              file=sys.stderr)
     if rank == 0:
         print(json.dumps({
             "metric": "graphs/sec training (MACE, MD17-shape)",
