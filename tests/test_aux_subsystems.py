@@ -136,3 +136,74 @@ def test_tracer_roundtrip(tmp_path):
     assert (tmp_path / "gp_timing.p0").exists()
     tr.disable()
     tr.reset()
+
+
+def test_download_with_local_http_server(tmp_path):
+    """Resumable sha256-verified download against a local HTTP server
+    (reference tests/test_dataset_download* pattern)."""
+    import http.server
+    import threading
+    from functools import partial
+    from hydragnn_amd.utils.datasets.download import download, sha256_of
+
+    src = tmp_path / "serve"
+    src.mkdir()
+    payload = bytes(range(256)) * 4096  # 1 MiB
+    (src / "data.bin").write_bytes(payload)
+    handler = partial(http.server.SimpleHTTPRequestHandler,
+                      directory=str(src))
+    httpd = http.server.ThreadingHTTPServer(("127.0.0.1", 0), handler)
+    port = httpd.server_address[1]
+    t = threading.Thread(target=httpd.serve_forever, daemon=True)
+    t.start()
+    try:
+        dest = tmp_path / "out" / "data.bin"
+        url = f"http://127.0.0.1:{port}/data.bin"
+        sha = sha256_of(str(src / "data.bin"))
+        download(url, str(dest), sha)
+        assert dest.read_bytes() == payload
+        # resume: truncate to a .part and re-download (Range request)
+        dest.unlink()
+        part = tmp_path / "out" / "data.bin.part"
+        part.write_bytes(payload[: len(payload) // 2])
+        download(url, str(dest), sha)
+        assert dest.read_bytes() == payload
+        # wrong checksum rejected
+        dest.unlink()
+        with pytest.raises(ValueError, match="sha256"):
+            download(url, str(dest), "0" * 64)
+    finally:
+        httpd.shutdown()
+
+
+def test_radial_bases():
+    """Radial basis properties (reference test_radial_transforms
+    pattern): cutoff zeros, limits, shapes."""
+    from hydragnn_amd.ops import (bessel_basis, chebyshev_basis,
+                                  cosine_cutoff, gaussian_basis,
+                                  polynomial_cutoff, sinc_basis)
+    import math
+    r = torch.linspace(0.01, 6.0, 50).view(-1, 1)
+    w = torch.arange(1, 9).float() * math.pi / 5.0
+    b = bessel_basis(r, 5.0, w)
+    assert b.shape == (50, 8)
+    # bessel bases vanish at r_max (sin(n*pi) = 0)
+    at_rmax = bessel_basis(torch.tensor([[5.0]]), 5.0, w)
+    assert at_rmax.abs().max() < 1e-5
+    # polynomial cutoff: 1 at 0, 0 beyond r_max, monotone-ish
+    pc = polynomial_cutoff(r, 5.0)
+    assert float(polynomial_cutoff(torch.tensor([[0.0]]), 5.0)) == pytest.approx(1.0)
+    assert float(polynomial_cutoff(torch.tensor([[5.5]]), 5.0)) == 0.0
+    cc = cosine_cutoff(r.squeeze(-1), 5.0)
+    assert float(cosine_cutoff(torch.tensor([0.0]), 5.0)) == pytest.approx(1.0)
+    assert float(cosine_cutoff(torch.tensor([5.1]), 5.0)) == 0.0
+    # sinc basis finite at r -> 0 with the analytic limit n*pi/rc
+    s0 = sinc_basis(torch.tensor([[1e-12]]), 5.0, 4)
+    assert torch.allclose(
+        s0.flatten(),
+        torch.arange(1, 5).float() * math.pi / 5.0, atol=1e-4)
+    ch = chebyshev_basis(r, 6.0, 5)
+    assert ch.shape == (50, 5)
+    assert (ch.abs() <= 1.0 + 1e-6).all()
+    g = gaussian_basis(r, torch.linspace(0, 5, 10).view(1, -1), -0.5)
+    assert g.shape == (50, 10) and (g <= 1.0).all()
