@@ -6,9 +6,11 @@ MultiheadAttention or Performer/FAVOR+ linear attention) + MLP, with
 residuals and three norms; Performer projection redraw is driven by the
 training loop, not forward (safe under activation checkpointing).
 
-On MI355X the dense-batch SDPA path runs through torch's ROCm
-scaled_dot_product_attention (AOTriton flash kernels); the
-segment-varlen HIP kernel is the planned upgrade for large graphs.
+On MI355X the multihead path runs the segment-varlen HIP kernel
+(ops/csrc/varlen_attn.hip): one (graph, head) workgroup with K/V in
+LDS and an online softmax over exactly the graph's nodes — no dense
+padding.  Batches outside its envelope (head_dim > 32 or a graph with
+> 256 nodes) and the Performer path fall back to dense-batch SDPA.
 """
 
 from __future__ import annotations
@@ -122,6 +124,30 @@ class HydraGPSConv(nn.Module):
             return True
         return False
 
+    def _try_varlen_attention(self, x, batch):
+        """Run nn.MultiheadAttention's math through the segment-varlen
+        HIP kernel (no dense padding).  Returns None when the batch
+        does not fit the kernel envelope (falls back to dense SDPA)."""
+        from ..ops.scatter import _rowptr_from_sorted
+        from ..ops.varlen_attn import varlen_attention, varlen_eligible
+        attn = self.attn
+        head_dim = self.channels // self.heads
+        if not varlen_eligible(head_dim, 0, x.device):
+            return None
+        if attn.in_proj_weight is None or not attn.batch_first:
+            return None
+        num_graphs = int(batch[-1]) + 1 if batch.numel() else 0
+        ptr = _rowptr_from_sorted(batch, num_graphs)
+        max_seg = int((ptr[1:] - ptr[:-1]).max()) if num_graphs else 0
+        if not varlen_eligible(head_dim, max_seg, x.device):
+            return None
+        qkv = nn.functional.linear(x, attn.in_proj_weight,
+                                   attn.in_proj_bias)
+        q, k, v = qkv.view(x.shape[0], 3, self.heads,
+                           head_dim).unbind(dim=1)
+        out = varlen_attention(q, k, v, ptr, batch)
+        return attn.out_proj(out.reshape(x.shape[0], self.channels))
+
     def forward(self, inv_node_feat, equiv_node_feat, batch=None, **kwargs):
         hs = []
         equiv_out = equiv_node_feat
@@ -135,15 +161,20 @@ class HydraGPSConv(nn.Module):
             h_local = self.norm1(h_local)
             hs.append(h_local)
 
-        # global attention over dense batch
-        x_dense, mask = to_dense_batch(inv_node_feat, batch)
-        if self.attn_type == "multihead":
-            h_attn, _ = self.attn(x_dense, x_dense, x_dense,
-                                  key_padding_mask=~mask,
-                                  need_weights=False)
-        else:
-            h_attn = self.attn(x_dense, mask=mask)
-        h_attn = h_attn[mask]
+        # global attention: segment-varlen HIP kernel when the batch
+        # fits its envelope (molecular graphs), dense batch otherwise
+        h_attn = None
+        if self.attn_type == "multihead" and batch is not None:
+            h_attn = self._try_varlen_attention(inv_node_feat, batch)
+        if h_attn is None:
+            x_dense, mask = to_dense_batch(inv_node_feat, batch)
+            if self.attn_type == "multihead":
+                h_attn, _ = self.attn(x_dense, x_dense, x_dense,
+                                      key_padding_mask=~mask,
+                                      need_weights=False)
+            else:
+                h_attn = self.attn(x_dense, mask=mask)
+            h_attn = h_attn[mask]
         h_attn = nn.functional.dropout(h_attn, self.dropout, self.training)
         h_attn = h_attn + inv_node_feat
         h_attn = self.norm2(h_attn)

@@ -630,6 +630,10 @@ torch::Tensor etp_reduce(torch::Tensor A, torch::Tensor C, torch::Tensor D,
                          c10::optional<torch::Tensor> ci,
                          c10::optional<torch::Tensor> di, long n_rows);
 
+// defined in varlen_attn.hip
+torch::Tensor varlen_attention(torch::Tensor Q, torch::Tensor K,
+                               torch::Tensor V, torch::Tensor ptr);
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("etp_general", &etp_general, "fused ETP contraction (HIP)",
         pybind11::arg("A"), pybind11::arg("B"), pybind11::arg("C"),
@@ -664,4 +668,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("scatter_mean_fwd", &scatter_mean_fwd, "scatter-mean (HIP)");
   m.def("scatter_minmax_fwd", &scatter_minmax_fwd, "scatter-min/max (HIP)");
   m.def("radius_pairs", &radius_pairs, "radius pair enumeration (HIP)");
+  m.def("varlen_attention", &varlen_attention,
+        "segment-varlen attention (HIP)");
 }

@@ -11,7 +11,8 @@ ext_modules = [
         name="hydragnn_amd.ops._hip_ops",
         sources=["hydragnn_amd/ops/csrc/hip_ops.hip",
                  "hydragnn_amd/ops/csrc/etp.hip",
-                 "hydragnn_amd/ops/csrc/mfma_linear.hip"],
+                 "hydragnn_amd/ops/csrc/mfma_linear.hip",
+                 "hydragnn_amd/ops/csrc/varlen_attn.hip"],
         extra_compile_args={
             "cxx": ["-O3"],
             "nvcc": ["-O3", "--offload-arch=gfx950"],

@@ -66,3 +66,24 @@ def test_laplacian_pe_deterministic():
     add_laplacian_pe(d2, 3)
     assert torch.allclose(d1.pe, d2.pe)
     assert d1.pe.shape == (10, 3)
+
+
+def test_torch_varlen_attention_reference():
+    """The varlen reference (used as the HIP kernel's numerics baseline
+    and recompute backward) equals naive per-graph attention."""
+    from hydragnn_amd.ops.varlen_attn import torch_varlen_attention
+    torch.manual_seed(0)
+    sizes = [3, 1, 9, 6]
+    batch = torch.repeat_interleave(torch.arange(4), torch.tensor(sizes))
+    N, H, dh = sum(sizes), 2, 8
+    q, k, v = (torch.randn(N, H, dh) for _ in range(3))
+    out = torch_varlen_attention(q, k, v, batch)
+    lo = 0
+    for n in sizes:
+        qs = q[lo:lo + n].transpose(0, 1)  # [H, n, dh]
+        ks = k[lo:lo + n].transpose(0, 1)
+        vs = v[lo:lo + n].transpose(0, 1)
+        a = torch.softmax(qs @ ks.transpose(-1, -2) / dh ** 0.5, dim=-1)
+        ref = (a @ vs).transpose(0, 1)
+        assert torch.allclose(out[lo:lo + n], ref, atol=1e-5)
+        lo += n
