@@ -15,6 +15,8 @@ headline bench.
 
 from __future__ import annotations
 
+import math
+
 import torch
 
 from ._extension import get_extension
@@ -33,9 +35,12 @@ def torch_varlen_attention(q: torch.Tensor, k: torch.Tensor,
     xd, mask = to_dense_batch(x, batch)
     B, Nmax = mask.shape
     qd, kd, vd = xd.view(B, Nmax, 3, H, dh).permute(2, 0, 3, 1, 4)
-    attn_mask = mask.view(B, 1, 1, Nmax).expand(B, H, Nmax, Nmax)
-    out = torch.nn.functional.scaled_dot_product_attention(
-        qd, kd, vd, attn_mask=attn_mask)
+    # explicit softmax attention (not SDPA): double-differentiable on
+    # every backend, which the recompute backward relies on; finite
+    # mask bias keeps padded rows NaN-free
+    bias = torch.where(mask.view(B, 1, 1, Nmax), 0.0, -1e9).to(q.dtype)
+    logits = qd @ kd.transpose(-1, -2) / math.sqrt(dh) + bias
+    out = torch.softmax(logits, dim=-1) @ vd
     out = out.permute(0, 2, 1, 3)  # [B, Nmax, H, dh]
     return out[mask]
 

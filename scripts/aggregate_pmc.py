@@ -2,10 +2,10 @@
 
 Usage: python scripts/aggregate_pmc.py <csv> [<out_csv>]
 
-Groups by (kernel, counter) and prints total counter values plus
-derived figures: MFMA utilisation proxy (SQ_INSTS_MFMA / SQ_BUSY_CYCLES)
-and HBM read bytes (FETCH_SIZE is reported in 32-byte TCC lines on
-gfx9).
+Groups by (kernel, counter) and prints total counter values (units as
+rocprofv3 reports them; FETCH_SIZE is kilobytes fetched past TCC).
+SQ_INSTS_MFMA / SQ_BUSY_CYCLES gives an MFMA-issue-density proxy per
+kernel.
 """
 
 import csv
@@ -27,7 +27,8 @@ def main():
                 kcol = keys.get("kernel_name")
                 ccol = keys.get("counter_name")
                 vcol = keys.get("counter_value")
-            name = row[kcol].split("(")[0][:80]
+            name = row[kcol].replace("(anonymous namespace)::", "")
+            name = name.split("(")[0][:80]
             sums[(name, row[ccol])] += float(row[vcol])
             if row[ccol].endswith("WAVES"):
                 dispatches[name] += 1
