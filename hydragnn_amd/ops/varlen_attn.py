@@ -56,15 +56,19 @@ class _VarlenAttn(torch.autograd.Function):
 
     @staticmethod
     def backward(ctx, g):
+        # recompute through the torch reference, differentiating w.r.t.
+        # the SAVED tensors (not detached copies) so the second-order
+        # graph stays connected for force-style double backward
         q, k, v, batch = ctx.saved_tensors
+        need = ctx.needs_input_grad[:3]
         with torch.enable_grad():
-            qq, kk, vv = (t.detach().requires_grad_(True)
-                          for t in (q, k, v))
-            out = torch_varlen_attention(qq, kk, vv, batch)
-            grads = torch.autograd.grad(
-                out, (qq, kk, vv), g,
-                create_graph=torch.is_grad_enabled())
-        return grads[0], grads[1], grads[2], None, None
+            out = torch_varlen_attention(q, k, v, batch)
+            inputs = [t for t, n in zip((q, k, v), need) if n]
+            grads = iter(torch.autograd.grad(
+                out, inputs, g, create_graph=torch.is_grad_enabled(),
+                allow_unused=True))
+        res = [next(grads) if n else None for n in need]
+        return res[0], res[1], res[2], None, None
 
 
 def varlen_attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
