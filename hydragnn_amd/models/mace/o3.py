@@ -125,7 +125,15 @@ class IrrepsLinear(nn.Module):
         self.register_buffer("lmap", lmap)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        # x [N, C_in, D] -> one batched GEMM [D](N,C_in)@(C_in,C_out)
+        from ...ops.irreps_linear import (irreps_linear,
+                                          irreps_linear_eligible)
+        if irreps_linear_eligible(x, self.weight):
+            # single-pass MFMA kernel (csrc/irreps_linear.hip): x and
+            # the whole weight stack LDS-resident, no permute copies
+            return irreps_linear(x.to(torch.bfloat16).contiguous(),
+                                 self.weight.to(torch.bfloat16),
+                                 self.lmap, self.bias)
+        # fallback: one batched GEMM [D](N,C_in)@(C_in,C_out)
         W_m = self.weight.to(x.dtype)[self.lmap]  # [D, C_in, C_out]
         out = torch.bmm(x.permute(2, 0, 1), W_m).permute(1, 2, 0)
         if self.bias is not None:

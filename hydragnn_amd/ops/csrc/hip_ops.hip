@@ -634,6 +634,12 @@ torch::Tensor etp_reduce(torch::Tensor A, torch::Tensor C, torch::Tensor D,
 torch::Tensor varlen_attention(torch::Tensor Q, torch::Tensor K,
                                torch::Tensor V, torch::Tensor ptr);
 
+// defined in irreps_linear.hip
+torch::Tensor irreps_linear(torch::Tensor X, torch::Tensor W,
+                            torch::Tensor lmap,
+                            c10::optional<torch::Tensor> bias,
+                            bool trans_w);
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("etp_general", &etp_general, "fused ETP contraction (HIP)",
         pybind11::arg("A"), pybind11::arg("B"), pybind11::arg("C"),
@@ -670,4 +676,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("radius_pairs", &radius_pairs, "radius pair enumeration (HIP)");
   m.def("varlen_attention", &varlen_attention,
         "segment-varlen attention (HIP)");
+  m.def("irreps_linear", &irreps_linear,
+        "per-l channel-mixing MFMA linear (HIP)",
+        pybind11::arg("X"), pybind11::arg("W"), pybind11::arg("lmap"),
+        pybind11::arg("bias") = pybind11::none(),
+        pybind11::arg("trans_w") = false);
 }

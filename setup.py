@@ -12,7 +12,8 @@ ext_modules = [
         sources=["hydragnn_amd/ops/csrc/hip_ops.hip",
                  "hydragnn_amd/ops/csrc/etp.hip",
                  "hydragnn_amd/ops/csrc/mfma_linear.hip",
-                 "hydragnn_amd/ops/csrc/varlen_attn.hip"],
+                 "hydragnn_amd/ops/csrc/varlen_attn.hip",
+                 "hydragnn_amd/ops/csrc/irreps_linear.hip"],
         extra_compile_args={
             "cxx": ["-O3"],
             "nvcc": ["-O3", "--offload-arch=gfx950"],
