@@ -7,7 +7,7 @@
 //
 // The torch path (bmm over D slices) costs 3x HBM traffic: permute
 // copy -> D skinny hipBLASLt GEMMs (measured 1.4-3% MFMA issue
-// density) -> permute-back copy.  Here one workgroup owns a 16-row
+// density) -> permute-back copy.  Here one workgroup owns a BM-row
 // n-tile: it reads x[n0:n0+16, :, :] once at full coalescing,
 // transposes it into LDS as [m][n][ci], keeps the whole weight stack
 // [L, Cin, Cout] in LDS, runs all D 16xCout GEMMs on
@@ -28,7 +28,7 @@ namespace {
 using bf16x8 = __attribute__((ext_vector_type(8))) __bf16;
 using f32x4 = __attribute__((ext_vector_type(4))) float;
 
-constexpr int BM = 16;       // n rows per workgroup
+constexpr int BM = 32;       // n rows per workgroup
 constexpr int MAXD = 16;     // (lmax+1)^2, lmax <= 3
 constexpr int PAD = 8;       // bf16 elems of row padding (16 B)
 
@@ -99,15 +99,19 @@ __global__ __launch_bounds__(256) void irreps_linear_kernel(
   }
   __syncthreads();
 
-  // each wave owns co-tiles wave, wave+4, ... of 16 columns
-  for (int ct = wave; ct < Cout / 16; ct += 4) {
+  // waves round-robin the (row-tile, co-tile) grid of 16x16 sub-tiles
+  const int nct = Cout / 16;
+  for (int t = wave; t < (BM / 16) * nct; t += 4) {
+    const int rt = t / nct;
+    const int ct = t - rt * nct;
     f32x4 acc[MAXD];
     for (int m = 0; m < D; ++m) acc[m] = {0.f, 0.f, 0.f, 0.f};
     for (int k0 = 0; k0 < Cin; k0 += 32) {
       for (int m = 0; m < D; ++m) {
-        // A fragment: lane holds lA[m][lane&15][k0 + (lane>>4)*8 + j]
+        // A fragment: lane holds lA[m][row][k0 + (lane>>4)*8 + j]
         const __hip_bfloat16* ap =
-            &lA[(m * BM + (lane & 15)) * CP + k0 + (lane >> 4) * 8];
+            &lA[(m * BM + rt * 16 + (lane & 15)) * CP + k0 +
+                (lane >> 4) * 8];
         bf16x8 afrag = *reinterpret_cast<const bf16x8*>(ap);
         // B fragment: lane holds lW[l][k0+(lane>>4)*8+q][ct*16+(lane&15)]
         const __hip_bfloat16* bp =
@@ -123,7 +127,7 @@ __global__ __launch_bounds__(256) void irreps_linear_kernel(
     int col = ct * 16 + (lane & 15);
     float b = bias != nullptr ? bias[col] : 0.f;
     for (int reg = 0; reg < 4; ++reg) {
-      long row = n0 + (lane >> 4) * 4 + reg;
+      long row = n0 + rt * 16 + (lane >> 4) * 4 + reg;
       if (row < N) {
         __hip_bfloat16* op = out + (row * Cout + col) * D;
         for (int m = 0; m < D; ++m)
@@ -153,7 +157,7 @@ torch::Tensor irreps_linear(torch::Tensor X, torch::Tensor W,
   TORCH_CHECK(Cin % 32 == 0 && Cout % 64 == 0,
               "irreps_linear needs Cin % 32 == 0, Cout % 64 == 0");
   TORCH_CHECK(D <= 16 && lmap.numel() == D);
-  size_t lds_bytes = ((size_t)D * 16 * (Cin + 8) +
+  size_t lds_bytes = ((size_t)D * BM * (Cin + 8) +
                       (size_t)L * Cin * (Cout + 8)) * 2;
   TORCH_CHECK(lds_bytes <= 160 * 1024, "irreps_linear LDS budget");
   auto out = torch::empty({N, (long)Cout, (long)D}, X.options());

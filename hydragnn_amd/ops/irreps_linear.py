@@ -29,7 +29,7 @@ def irreps_kernel_ok(n: int, c_in: int, c_out: int, d: int,
                      n_l: int) -> bool:
     if c_in % 32 != 0 or c_out % 64 != 0 or d > 16:
         return False
-    lds = (d * 16 * (c_in + 8) + n_l * c_in * (c_out + 8)) * 2
+    lds = (d * 32 * (c_in + 8) + n_l * c_in * (c_out + 8)) * 2
     return lds <= _MAX_LDS and n >= 64
 
 
@@ -55,16 +55,15 @@ class _IrrepsLinearFn(torch.autograd.Function):
         if ctx.needs_input_grad[0]:
             gx = _IrrepsLinearFn.apply(g, W, lmap, None, not trans_w)
         if ctx.needs_input_grad[1]:
-            # lmap is the canonical l-block layout (lo = l^2), so the
-            # slices are static — no device sync, capture-safe
-            parts = []
-            d = x.shape[2]
-            for l in range(W.shape[0]):
-                lo, hi = l * l, min((l + 1) * (l + 1), d)
-                gw_l = torch.einsum("nim,nom->io", x[:, :, lo:hi],
-                                    g[:, :, lo:hi])
-                parts.append(gw_l.t() if trans_w else gw_l)
-            gw = torch.stack(parts).to(W.dtype)
+            # per-m weight grads as ONE batched GEMM (good hipBLASLt
+            # split-K tiles), then fold m -> l with index_add
+            # (capture-safe: lmap stays on device)
+            gw_m = torch.bmm(x.permute(2, 1, 0), g.permute(2, 0, 1))
+            gw = gw_m.new_zeros(W.shape[0], x.shape[1], g.shape[1])
+            gw.index_add_(0, lmap, gw_m)
+            if trans_w:
+                gw = gw.transpose(1, 2)
+            gw = gw.to(W.dtype)
         if bias is not None and ctx.needs_input_grad[3]:
             gb = g[:, :, 0].sum(0).to(bias.dtype)
         return gx, gw, None, gb, None
