@@ -639,6 +639,9 @@ torch::Tensor irreps_linear(torch::Tensor X, torch::Tensor W,
                             torch::Tensor lmap,
                             c10::optional<torch::Tensor> bias,
                             bool trans_w);
+torch::Tensor irreps_linear_gw(torch::Tensor X, torch::Tensor G,
+                               torch::Tensor lmap, long L,
+                               long nblocks);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("etp_general", &etp_general, "fused ETP contraction (HIP)",
@@ -681,4 +684,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         pybind11::arg("X"), pybind11::arg("W"), pybind11::arg("lmap"),
         pybind11::arg("bias") = pybind11::none(),
         pybind11::arg("trans_w") = false);
+  m.def("irreps_linear_gw", &irreps_linear_gw,
+        "irreps-linear weight-grad partials (HIP)",
+        pybind11::arg("X"), pybind11::arg("G"), pybind11::arg("lmap"),
+        pybind11::arg("L"), pybind11::arg("nblocks") = 512);
 }

@@ -137,6 +137,110 @@ __global__ __launch_bounds__(256) void irreps_linear_kernel(
   }
 }
 
+// Weight gradient: gW[l, ci, co] = sum_{n, m in l} x[n,ci,m]*g[n,co,m].
+// The bmm route needs permuted COPIES of both x and g (50 MB/call at
+// bench shapes); here persistent blocks stride the n-tiles, stage x
+// and g through LDS with the same transpose as the forward, run the
+// contraction over n on MFMA (k = n-chunk of 32), and write per-block
+// fp32 partials (summed by the caller) — deterministic, no atomics,
+// single-pass traffic.
+__global__ __launch_bounds__(256) void irreps_linear_gw_kernel(
+    const __hip_bfloat16* __restrict__ X,   // [N, Cin, D]
+    const __hip_bfloat16* __restrict__ G,   // [N, Cout, D]
+    float* __restrict__ partials,           // [nblocks, L, Cin, Cout]
+    const long* __restrict__ lmap,          // [D]
+    long N, int Cin, int Cout, int D, int L) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  const int CP = Cin + PAD;
+  const int GP = Cout + PAD;
+  __hip_bfloat16* lA = reinterpret_cast<__hip_bfloat16*>(smem);  // [D][BM][CP]
+  __hip_bfloat16* lG = lA + D * BM * CP;                         // [D][BM][GP]
+  __shared__ int lmap_s[MAXD];
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  if (tid < D) lmap_s[tid] = (int)lmap[tid];
+
+  // each wave owns (Cin/16 * Cout/16 / 4) sub-tiles, acc per (tile, l)
+  const int nct = Cout / 16;
+  const int ntiles = (Cin / 16) * nct;
+  const int mytiles = ntiles / 4;             // Cin,Cout >= 32 -> >= 1
+  f32x4 acc[16];                              // [mytile * L], L*mytiles<=16
+  for (int i = 0; i < 16; ++i) acc[i] = {0.f, 0.f, 0.f, 0.f};
+
+  const long tiles_n = (N + BM - 1) / BM;
+  for (long tb = blockIdx.x; tb < tiles_n; tb += gridDim.x) {
+    const long n0 = tb * BM;
+    __syncthreads();
+    {  // stage x and g tiles (transpose to [m][n][c])
+      const int rex = Cin * D, reg_ = Cout * D;
+      const int cx = BM * rex / 8, cg = BM * reg_ / 8;
+      for (int idx = tid; idx < cx + cg; idx += 256) {
+        bool isx = idx < cx;
+        int id2 = isx ? idx : idx - cx;
+        int re = isx ? rex : reg_;
+        int C = isx ? Cin : Cout;
+        int P = isx ? CP : GP;
+        __hip_bfloat16* dst = isx ? lA : lG;
+        const __hip_bfloat16* src = isx ? X : G;
+        int r = id2 / (re / 8);
+        int f = (id2 - r * (re / 8)) * 8;
+        long gn = n0 + r;
+        if (gn < N) {
+          bf16x8 v = *reinterpret_cast<const bf16x8*>(
+              src + gn * re + f);
+          for (int t = 0; t < 8; ++t) {
+            int c = (f + t) / D, m = (f + t) - ((f + t) / D) * D;
+            dst[(m * BM + r) * P + c] = __hip_bfloat16(v[t]);
+          }
+        } else {
+          for (int t = 0; t < 8; ++t) {
+            int c = (f + t) / D, m = (f + t) - ((f + t) / D) * D;
+            dst[(m * BM + r) * P + c] = __hip_bfloat16(0.f);
+          }
+        }
+      }
+    }
+    __syncthreads();
+    for (int mt = 0; mt < mytiles; ++mt) {
+      int t = wave * mytiles + mt;
+      int cit = t / nct, cot = t - (t / nct) * nct;
+      for (int m = 0; m < D; ++m) {
+        int l = lmap_s[m];
+        // k = n over the whole BM=32 tile in ONE mfma: lane group
+        // (lane>>4) covers k = (lane>>4)*8 + q, q = 0..7
+        // A[ci][k=n] from lA[m][n][ci] (stride-CP gather)
+        const __hip_bfloat16* ap =
+            &lA[(m * BM + (lane >> 4) * 8) * CP + cit * 16 + (lane & 15)];
+        bf16x8 afrag;
+        for (int q = 0; q < 8; ++q) afrag[q] = (__bf16)ap[q * CP];
+        // B[k=n][co] from lG[m][n][co]
+        const __hip_bfloat16* bp =
+            &lG[(m * BM + (lane >> 4) * 8) * GP + cot * 16 + (lane & 15)];
+        bf16x8 bfrag;
+        for (int q = 0; q < 8; ++q) bfrag[q] = (__bf16)bp[q * GP];
+        acc[mt * 4 + l] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            afrag, bfrag, acc[mt * 4 + l], 0, 0, 0);
+      }
+    }
+  }
+  // flush: C layout col = lane&15 (co), row = (lane>>4)*4 + reg (ci)
+  for (int mt = 0; mt < mytiles; ++mt) {
+    int t = wave * mytiles + mt;
+    int cit = t / nct, cot = t - (t / nct) * nct;
+    for (int l = 0; l < L; ++l) {
+      f32x4 a = acc[mt * 4 + l];
+      for (int reg_ = 0; reg_ < 4; ++reg_) {
+        int ci = cit * 16 + (lane >> 4) * 4 + reg_;
+        int co = cot * 16 + (lane & 15);
+        partials[(((long)blockIdx.x * L + l) * Cin + ci) * Cout + co] =
+            a[reg_];
+      }
+    }
+  }
+}
+
 }  // namespace
 
 torch::Tensor irreps_linear(torch::Tensor X, torch::Tensor W,
@@ -183,3 +287,37 @@ torch::Tensor irreps_linear(torch::Tensor X, torch::Tensor W,
   else launch(irreps_linear_kernel<false>);
   return out;
 }
+
+torch::Tensor irreps_linear_gw(torch::Tensor X, torch::Tensor G,
+                               torch::Tensor lmap, long L,
+                               long nblocks) {
+  TORCH_CHECK(X.is_cuda() && X.is_contiguous());
+  TORCH_CHECK(G.is_cuda() && G.is_contiguous());
+  TORCH_CHECK(X.scalar_type() == at::ScalarType::BFloat16 &&
+              G.scalar_type() == at::ScalarType::BFloat16);
+  long N = X.size(0);
+  int Cin = X.size(1), D = X.size(2);
+  int Cout = G.size(1);
+  TORCH_CHECK(G.size(0) == N && G.size(2) == D);
+  int tiles = (Cin / 16) * (Cout / 16);
+  TORCH_CHECK(Cin % 16 == 0 && Cout % 16 == 0 && tiles % 4 == 0 &&
+              tiles <= 16 && L <= 4 && D <= 16,
+              "irreps_linear_gw shape envelope");
+  size_t lds_bytes = ((size_t)D * BM * (Cin + 8) +
+                      (size_t)D * BM * (Cout + 8)) * 2;
+  TORCH_CHECK(lds_bytes <= 160 * 1024, "irreps_linear_gw LDS budget");
+  auto partials = torch::empty(
+      {nblocks, L, Cin, Cout},
+      X.options().dtype(torch::kFloat));
+  if (N == 0) return partials.zero_();
+  auto lmap_c = lmap.contiguous();
+  auto stream = at::hip::getCurrentHIPStream().stream();
+  hipLaunchKernelGGL(irreps_linear_gw_kernel, dim3(nblocks), dim3(256),
+                     lds_bytes, stream,
+                     reinterpret_cast<const __hip_bfloat16*>(X.data_ptr()),
+                     reinterpret_cast<const __hip_bfloat16*>(G.data_ptr()),
+                     partials.data_ptr<float>(),
+                     lmap_c.data_ptr<long>(), N, Cin, Cout, D, (int)L);
+  return partials;
+}
+
