@@ -11,9 +11,14 @@ v_mfma_f32_16x16x32_bf16, registers holding every m until the single
 output write.
 
 Gradient closure: gX is the same kernel with the weight read
-transposed (trans_w flipped); gW is a small per-l reduction GEMM done
-with torch einsum (output is only [Cin, Cout] per l).  First and
-second order both stay functional, so force training works.
+transposed (trans_w flipped); gW is a second HIP kernel
+(irreps_linear_gw_kernel) contracting over n with MFMA from the same
+LDS staging — copy-free and deterministic (per-block fp32 partials,
+summed on the stream).  An earlier bmm/einsum gW was measured SLOWER
+than the torch path end-to-end (permuted copies of x and g); the
+kernel version flips the A/B to +1.3% (32.7k vs 32.3k g/s, default
+bench).  First and second order both stay on the kernels, so force
+training works.
 """
 
 from __future__ import annotations
@@ -90,12 +95,7 @@ def irreps_linear_eligible(x: torch.Tensor, W: torch.Tensor) -> bool:
     import os
     if use_eager() or not x.is_cuda:
         return False
-    # default OFF: the kernel wins on forward traffic, but the weight
-    # gradient needs permuted copies of x and g that the bmm path's
-    # autograd gets for free from its saved forward permute — measured
-    # 31.1k vs 32.2k g/s end-to-end on the default bench (A/B in
-    # profiles/README.md).  Flip on with HYDRAGNN_IRREPS_MFMA=1.
-    if os.environ.get("HYDRAGNN_IRREPS_MFMA", "0") != "1":
+    if os.environ.get("HYDRAGNN_IRREPS_MFMA", "1") == "0":
         return False
     if not (x.dtype == torch.bfloat16 or torch.is_autocast_enabled()):
         return False
