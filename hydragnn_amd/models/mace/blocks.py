@@ -210,8 +210,11 @@ class RealAgnosticResidualInteractionBlock(nn.Module):
             m = etp_indexed(x, edge_sh, w, self.conv_tp.etp_table,
                             etp_meta)
         else:
-            x_src = gather(x.reshape(n, -1), src).view(-1, c,
-                                                       x.shape[-1])
+            src_csr = getattr(etp_meta, "src_csr", None) \
+                if etp_meta is not None else None
+            x_src = gather(x.reshape(n, -1), src,
+                           backward_csr=src_csr).view(-1, c,
+                                                      x.shape[-1])
             mji = self.conv_tp(x_src, edge_sh, w)
             m = scatter(mji.reshape(mji.shape[0], -1), dst, n, "sum",
                         sorted_index=edges_sorted).view(n, c, -1)

@@ -153,6 +153,11 @@ class MACEStack(Base):
         rowptr_dst = _rowptr_from_sorted(dst[eid_d], n)
         meta = ETPMeta(src.numel(), ai=src[eid_d], bi=eid_d, ci=eid_d,
                        rowptr=rowptr_dst, n_a_rows=n)
+        # src-sort for gather's backward scatter (src is unsorted in a
+        # dst-sorted batch): contention-free indexed CSR instead of
+        # atomics
+        perm_s = torch.argsort(src, stable=True)
+        meta.src_csr = (perm_s, _rowptr_from_sorted(src[perm_s], n))
         data[key] = meta
         return meta
 

@@ -141,6 +141,45 @@ __global__ void segment_sum_csr_kernel<__hip_bfloat16, float>(
   }
 }
 
+// Indexed CSR segment sum: out[r] = sum_{e in row r} src[perm[e]].
+// Used as gather's backward when the caller precomputed an index-sort
+// (scatter-by-unsorted-index without atomics: deterministic, no L2
+// contention on hot rows).
+template <typename T, typename ACC>
+__global__ void segment_sum_csr_idx_kernel(
+    const T* __restrict__ src, const long* __restrict__ rowptr,
+    const long* __restrict__ perm, T* __restrict__ out, long N,
+    long F) {
+  long total = N * F;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x) {
+    long n = i / F;
+    long f = i - n * F;
+    ACC acc = (ACC)0;
+    long lo = rowptr[n], hi = rowptr[n + 1];
+    for (long e = lo; e < hi; ++e) acc += (ACC)src[perm[e] * F + f];
+    out[i] = (T)acc;
+  }
+}
+
+template <>
+__global__ void segment_sum_csr_idx_kernel<__hip_bfloat16, float>(
+    const __hip_bfloat16* __restrict__ src,
+    const long* __restrict__ rowptr, const long* __restrict__ perm,
+    __hip_bfloat16* __restrict__ out, long N, long F) {
+  long total = N * F;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x) {
+    long n = i / F;
+    long f = i - n * F;
+    float acc = 0.f;
+    long lo = rowptr[n], hi = rowptr[n + 1];
+    for (long e = lo; e < hi; ++e)
+      acc += __bfloat162float(src[perm[e] * F + f]);
+    out[i] = __float2bfloat16(acc);
+  }
+}
+
 __global__ void count_kernel(const long* __restrict__ index,
                              float* __restrict__ count, long E) {
   for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < E;
@@ -421,7 +460,8 @@ torch::Tensor scatter_sum_fwd(torch::Tensor src, torch::Tensor index,
   return out;
 }
 
-torch::Tensor segment_sum_csr(torch::Tensor src, torch::Tensor rowptr) {
+torch::Tensor segment_sum_csr(torch::Tensor src, torch::Tensor rowptr,
+                              c10::optional<torch::Tensor> perm) {
   CHECK_CUDA(src); CHECK_CONTIG(src);
   long N = rowptr.numel() - 1;
   long F = src.numel() / std::max<long>(src.size(0), 1);
@@ -430,23 +470,47 @@ torch::Tensor segment_sum_csr(torch::Tensor src, torch::Tensor rowptr) {
   auto out = torch::empty(sizes, src.options());
   auto rp = rowptr.contiguous();
   if (N == 0) return out;
+  torch::Tensor pc;
+  const long* perm_ptr = nullptr;
+  if (perm.has_value()) {
+    pc = perm->contiguous();
+    perm_ptr = pc.data_ptr<long>();
+  }
   if (src.scalar_type() == at::ScalarType::BFloat16) {
-    hipLaunchKernelGGL((segment_sum_csr_kernel<__hip_bfloat16, float>),
-                       dim3(n_blocks(N * F, kBlock, 8192)), dim3(kBlock),
-                       0, cur_stream(),
-                       reinterpret_cast<const __hip_bfloat16*>(src.data_ptr()),
-                       rp.data_ptr<long>(),
-                       reinterpret_cast<__hip_bfloat16*>(out.data_ptr()),
-                       N, F);
+    if (perm_ptr) {
+      hipLaunchKernelGGL(
+          (segment_sum_csr_idx_kernel<__hip_bfloat16, float>),
+          dim3(n_blocks(N * F, kBlock, 8192)), dim3(kBlock), 0,
+          cur_stream(),
+          reinterpret_cast<const __hip_bfloat16*>(src.data_ptr()),
+          rp.data_ptr<long>(), perm_ptr,
+          reinterpret_cast<__hip_bfloat16*>(out.data_ptr()), N, F);
+    } else {
+      hipLaunchKernelGGL(
+          (segment_sum_csr_kernel<__hip_bfloat16, float>),
+          dim3(n_blocks(N * F, kBlock, 8192)), dim3(kBlock), 0,
+          cur_stream(),
+          reinterpret_cast<const __hip_bfloat16*>(src.data_ptr()),
+          rp.data_ptr<long>(),
+          reinterpret_cast<__hip_bfloat16*>(out.data_ptr()), N, F);
+    }
     return out;
   }
   AT_DISPATCH_FLOATING_TYPES_AND(at::ScalarType::Half, src.scalar_type(),
                                  "segment_sum_csr", [&] {
-    hipLaunchKernelGGL((segment_sum_csr_kernel<scalar_t, scalar_t>),
-                       dim3(n_blocks(N * F, kBlock, 8192)), dim3(kBlock),
-                       0, cur_stream(), src.data_ptr<scalar_t>(),
-                       rp.data_ptr<long>(), out.data_ptr<scalar_t>(), N,
-                       F);
+    if (perm_ptr) {
+      hipLaunchKernelGGL((segment_sum_csr_idx_kernel<scalar_t, scalar_t>),
+                         dim3(n_blocks(N * F, kBlock, 8192)), dim3(kBlock),
+                         0, cur_stream(), src.data_ptr<scalar_t>(),
+                         rp.data_ptr<long>(), perm_ptr,
+                         out.data_ptr<scalar_t>(), N, F);
+    } else {
+      hipLaunchKernelGGL((segment_sum_csr_kernel<scalar_t, scalar_t>),
+                         dim3(n_blocks(N * F, kBlock, 8192)), dim3(kBlock),
+                         0, cur_stream(), src.data_ptr<scalar_t>(),
+                         rp.data_ptr<long>(), out.data_ptr<scalar_t>(), N,
+                         F);
+    }
   });
   return out;
 }
@@ -673,7 +737,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         pybind11::arg("n_rows") = 0);
   m.def("gather_fwd", &gather_fwd, "gather rows (HIP)");
   m.def("scatter_sum_fwd", &scatter_sum_fwd, "scatter-add (HIP)");
-  m.def("segment_sum_csr", &segment_sum_csr, "CSR segment sum (HIP)");
+  m.def("segment_sum_csr", &segment_sum_csr, "CSR segment sum (HIP)",
+        pybind11::arg("src"), pybind11::arg("rowptr"),
+        pybind11::arg("perm") = pybind11::none());
   m.def("scatter_mean_fwd", &scatter_mean_fwd, "scatter-mean (HIP)");
   m.def("scatter_minmax_fwd", &scatter_minmax_fwd, "scatter-min/max (HIP)");
   m.def("radius_pairs", &radius_pairs, "radius pair enumeration (HIP)");

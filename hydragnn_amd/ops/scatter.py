@@ -34,9 +34,11 @@ def _use_hip(t: torch.Tensor) -> bool:
 # ---------------------------------------------------------------------------
 class _Gather(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, src: torch.Tensor, index: torch.Tensor, dim_size: int):
+    def forward(ctx, src: torch.Tensor, index: torch.Tensor,
+                dim_size: int, backward_csr=None):
         ctx.save_for_backward(index)
         ctx.dim_size = src.shape[0]
+        ctx.backward_csr = backward_csr
         if _use_hip(src):
             ext = get_extension(required=True)
             return ext.gather_fwd(src.contiguous(), index)
@@ -45,13 +47,21 @@ class _Gather(torch.autograd.Function):
     @staticmethod
     def backward(ctx, grad_out: torch.Tensor):
         (index,) = ctx.saved_tensors
-        grad_src = scatter(grad_out, index, ctx.dim_size, reduce="sum")
-        return grad_src, None, None
+        grad_src = scatter(grad_out, index, ctx.dim_size, reduce="sum",
+                           csr=ctx.backward_csr)
+        return grad_src, None, None, None
 
 
-def gather(src: torch.Tensor, index: torch.Tensor) -> torch.Tensor:
-    """out[e] = src[index[e]] with double-backward support."""
-    return _Gather.apply(src, index, src.shape[0])
+def gather(src: torch.Tensor, index: torch.Tensor,
+           backward_csr=None) -> torch.Tensor:
+    """out[e] = src[index[e]] with double-backward support.
+
+    backward_csr: optional (perm, rowptr) precomputed sort of `index`
+    (perm = argsort(index), rowptr over src rows) — the backward
+    scatter then runs the contention-free indexed CSR kernel instead
+    of atomics (the index is typically UNsorted here: gathers go by
+    edge source while batches are destination-sorted)."""
+    return _Gather.apply(src, index, src.shape[0], backward_csr)
 
 
 # ---------------------------------------------------------------------------
@@ -70,11 +80,18 @@ def _rowptr_from_sorted(index: torch.Tensor, dim_size: int):
 
 class _ScatterSum(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, src, index, dim_size, sorted_index=False):
+    def forward(ctx, src, index, dim_size, sorted_index=False, csr=None):
         ctx.save_for_backward(index)
         if _use_hip(src):
             import os
             ext = get_extension(required=True)
+            if csr is not None and \
+                    os.environ.get("HYDRAGNN_CSR_SCATTER", "1") == "1":
+                # precomputed index-sort: out[r] = sum src[perm[e]] —
+                # contention-free and deterministic for UNSORTED index
+                perm, rowptr = csr
+                return ext.segment_sum_csr(src.contiguous(), rowptr,
+                                           perm)
             if sorted_index and hasattr(ext, "segment_sum_csr") and \
                     os.environ.get("HYDRAGNN_CSR_SCATTER", "1") == "1":
                 rowptr = _rowptr_from_sorted(index, dim_size)
@@ -94,7 +111,7 @@ class _ScatterSum(torch.autograd.Function):
     @staticmethod
     def backward(ctx, grad_out):
         (index,) = ctx.saved_tensors
-        return gather(grad_out, index), None, None, None
+        return gather(grad_out, index), None, None, None, None
 
 
 class _ScatterMean(torch.autograd.Function):
@@ -184,11 +201,15 @@ def scatter(
     reduce: str = "sum",
     dim: int = 0,
     sorted_index: bool = False,
+    csr=None,
 ) -> torch.Tensor:
     """Segment reduction along dim 0.
 
     src   [E, ...]  values
     index [E]       destination ids in [0, dim_size)
+    csr             optional (perm, rowptr) precomputed sort of index
+                    (sum only): runs the indexed CSR kernel instead of
+                    atomics even for unsorted index
     """
     assert dim == 0, "hydragnn_amd.ops.scatter reduces along dim 0"
     if dim_size is None:
@@ -197,10 +218,10 @@ def scatter(
         index = index.long()
     if src.dim() == 1:
         out = scatter(src.unsqueeze(1), index, dim_size, reduce,
-                      sorted_index=sorted_index)
+                      sorted_index=sorted_index, csr=csr)
         return out.squeeze(1)
     if reduce in ("sum", "add"):
-        return _ScatterSum.apply(src, index, dim_size, sorted_index)
+        return _ScatterSum.apply(src, index, dim_size, sorted_index, csr)
     if reduce == "mean":
         return _ScatterMean.apply(src, index, dim_size)
     if reduce in ("max", "amax"):
