@@ -231,9 +231,10 @@ def test_gather_backward_csr_parity():
     src = torch.randn(N, F, device="cuda", requires_grad=True)
     w = torch.randn(E, F, device="cuda")
     g1 = torch.autograd.grad((gather(src, idx, backward_csr=csr)
-                              * w).sum(), src, create_graph=True)[0]
-    g2 = torch.autograd.grad((gather(src, idx) * w).sum(), src,
+                              .square() * w).sum(), src,
                              create_graph=True)[0]
+    g2 = torch.autograd.grad((gather(src, idx).square() * w).sum(),
+                             src, create_graph=True)[0]
     assert (g1 - g2).abs().max() < 1e-3
     gg1 = torch.autograd.grad(g1.square().sum(), src)[0]
     gg2 = torch.autograd.grad(g2.square().sum(), src)[0]
