@@ -102,6 +102,54 @@ __global__ void etp_general_kernel(
   }
 }
 
+// Occupancy variant of etp_general: A/B/C rows are PRIVATE per thread
+// and only ~1-3 cache lines each, so after first touch they are
+// L1-hot — staging them in LDS buys nothing but caps the block count
+// at 1/CU (zero latency hiding; PMC showed SQ busy ~10% of kernel
+// wall time).  Here only the runtime-indexed OUTPUT accumulator lives
+// in LDS; operand reads go straight to L1.  The host picks whichever
+// variant yields more blocks/CU.
+template <typename T>
+__global__ void etp_general_l1_kernel(
+    const T* __restrict__ A, const T* __restrict__ B,
+    const T* __restrict__ C, T* __restrict__ out,
+    const int4* __restrict__ entries,
+    const float* __restrict__ coefs, int n_ent,
+    long NC, int nch, int da, int db, int dg, int do_,
+    const long* __restrict__ ai, const long* __restrict__ bi,
+    const long* __restrict__ ci) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  const int stride = do_ | 1;
+  float* slices = reinterpret_cast<float*>(smem);
+  int4* ent_lds = reinterpret_cast<int4*>(
+      smem + (size_t)blockDim.x * stride * 4);
+  float* coef_lds = reinterpret_cast<float*>(ent_lds + n_ent);
+  for (int k = threadIdx.x; k < n_ent; k += blockDim.x) {
+    ent_lds[k] = entries[k];
+    coef_lds[k] = coefs[k];
+  }
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  float* mo = slices + (size_t)threadIdx.x * stride;
+  __syncthreads();
+  if (i >= NC) return;
+  long e = i / nch;
+  int c = (int)(i - e * nch);
+  long ea = ai ? ai[e] : e;
+  long eb = bi ? bi[e] : e;
+  long ec = ci ? ci[e] : e;
+  const T* ap = A + (ea * nch + c) * da;
+  const T* bp = B + eb * db;
+  const T* cp = C + (ec * nch + c) * dg;
+  for (int k = 0; k < do_; ++k) mo[k] = 0.f;
+  for (int k = 0; k < n_ent; ++k) {
+    int4 q = ent_lds[k];
+    mo[q.w] += coef_lds[k] * to_f32(ap[q.x]) * to_f32(bp[q.y]) *
+               to_f32(cp[q.z]);
+  }
+  T* op = out + i * do_;
+  for (int k = 0; k < do_; ++k) op[k] = from_f32<T>(mo[k]);
+}
+
 // out[e,b] = sum_c sum over entries(coef, a, b, g, o) of
 //            coef * A[e,c,a] * C[e,c,g] * D[e,c,o]
 // (the B-slot gradient: reduce over channels).  Same structure as
@@ -324,15 +372,29 @@ torch::Tensor etp_general(torch::Tensor A, torch::Tensor B, torch::Tensor C,
   int n_ent = entries.size(0);
   int block = etp_block_size();
   int stride = (da + db + dg + (int)do_) | 1;
-  size_t lds_bytes = (size_t)block * stride * 4 + n_ent * 20;
+  size_t lds_full = (size_t)block * stride * 4 + n_ent * 20;
+  int stride_l1 = (int)do_ | 1;
+  size_t lds_l1 = (size_t)block * stride_l1 * 4 + n_ent * 20;
+  // pick the variant with more resident blocks/CU (160 KB LDS): the
+  // staged kernel usually fits once (zero latency hiding), the
+  // L1-operand kernel 2-4x — see kernel comment
+  long occ_full = lds_full ? (160 * 1024) / lds_full : 0;
+  long occ_l1 = lds_l1 ? (160 * 1024) / lds_l1 : 0;
+  bool use_l1 = occ_l1 > occ_full;
+  const char* env = getenv("HYDRAGNN_ETP_L1");
+  if (env && env[0] == '0') use_l1 = false;
+  if (env && env[0] == '1') use_l1 = true;
+  size_t lds_bytes = use_l1 ? lds_l1 : lds_full;
   TORCH_CHECK(lds_bytes <= 150 * 1024, "etp LDS budget exceeded");
   long blocks = (NC + block - 1) / block;
   // fp64 is routed to the eager path in Python (float LDS staging here)
   AT_DISPATCH_FLOATING_TYPES_AND2(
       at::ScalarType::BFloat16, at::ScalarType::Half, A.scalar_type(),
       "etp_general", [&] {
+        auto kern = use_l1 ? etp_general_l1_kernel<scalar_t>
+                           : etp_general_kernel<scalar_t>;
         hipLaunchKernelGGL(
-            etp_general_kernel<scalar_t>, dim3(blocks), dim3(block),
+            kern, dim3(blocks), dim3(block),
             lds_bytes, etp_stream(),
             A.data_ptr<scalar_t>(), B.data_ptr<scalar_t>(),
             C.data_ptr<scalar_t>(), out.data_ptr<scalar_t>(),
