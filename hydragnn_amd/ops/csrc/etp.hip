@@ -375,12 +375,12 @@ torch::Tensor etp_general(torch::Tensor A, torch::Tensor B, torch::Tensor C,
   size_t lds_full = (size_t)block * stride * 4 + n_ent * 20;
   int stride_l1 = (int)do_ | 1;
   size_t lds_l1 = (size_t)block * stride_l1 * 4 + n_ent * 20;
-  // pick the variant with more resident blocks/CU (160 KB LDS): the
-  // staged kernel usually fits once (zero latency hiding), the
-  // L1-operand kernel 2-4x — see kernel comment
-  long occ_full = lds_full ? (160 * 1024) / lds_full : 0;
-  long occ_l1 = lds_l1 ? (160 * 1024) / lds_l1 : 0;
-  bool use_l1 = occ_l1 > occ_full;
+  // Measured A/B (default bench, b1024): staged 32.9k g/s vs
+  // L1-operand 30.9k — the higher occupancy does NOT pay for the 3
+  // L1 loads + converts per entry, so staged is the default; the L1
+  // variant stays for shapes that exceed the staged LDS budget and
+  // for HYDRAGNN_ETP_L1=1 experiments.
+  bool use_l1 = lds_full > 150 * 1024;
   const char* env = getenv("HYDRAGNN_ETP_L1");
   if (env && env[0] == '0') use_l1 = false;
   if (env && env[0] == '1') use_l1 = true;

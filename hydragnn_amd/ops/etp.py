@@ -91,7 +91,10 @@ def _dense_general(A, B, C, table: ETPTable):
 def _kernel_ok(table: ETPTable, *tensors) -> bool:
     da, db, dg, do = table.dims
     # real constraint is the per-block LDS budget (256 thread slices +
-    # staged entry table) — mirrors the check in csrc/etp.hip
+    # staged entry table) — mirrors the check in csrc/etp.hip.  Kept
+    # at the FULL-staging footprint even though etp_general can fall
+    # to its L1-operand variant: the gradient kernels (etp_reduce)
+    # still stage fully, and this predicate gates the whole family.
     lds_bytes = 256 * (da + db + dg + do + 1) * 4 + \
         table.entries.shape[0] * 20
     if lds_bytes > 150 * 1024 or max(da, db, dg, do) > 192:
