@@ -219,14 +219,14 @@ class RealAgnosticResidualInteractionBlock(nn.Module):
             m = scatter(mji.reshape(mji.shape[0], -1), dst, n, "sum",
                         sorted_index=edges_sorted).view(n, c, -1)
         m = m / self.avg_num_neighbors
-        out = self.linear(m)
-        # skip: per-l linear on the input, padded to lmax_out
+        # skip: per-l linear on the input, padded to lmax_out; fused
+        # into the main linear's epilogue on the kernel path
         sc = self.skip_linear(node_feats[:, :, :dim(min(self.lmax_node,
                                                         self.lmax_out))])
-        if sc.shape[-1] < out.shape[-1]:
-            sc = torch.nn.functional.pad(sc, (0, out.shape[-1]
-                                              - sc.shape[-1]))
-        return out + sc
+        want = dim(self.lmax_out)
+        if sc.shape[-1] < want:
+            sc = torch.nn.functional.pad(sc, (0, want - sc.shape[-1]))
+        return self.linear(m, add=sc)
 
 
 class EquivariantProductBasisBlock(nn.Module):
@@ -240,10 +240,9 @@ class EquivariantProductBasisBlock(nn.Module):
     def forward(self, node_feats: torch.Tensor, node_elem: torch.Tensor,
                 sc: Optional[torch.Tensor] = None) -> torch.Tensor:
         out = self.symmetric_contractions(node_feats, node_elem)
-        out = self.linear(out)
-        if sc is not None:
-            out = out + sc[:, :, :out.shape[-1]]
-        return out
+        want = self.linear.lmap.numel()
+        add = sc[:, :, :want] if sc is not None else None
+        return self.linear(out, add=add)
 
 
 class LinearReadoutBlock(nn.Module):

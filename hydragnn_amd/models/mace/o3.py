@@ -124,15 +124,19 @@ class IrrepsLinear(nn.Module):
                           for l in range(lmax + 1)])
         self.register_buffer("lmap", lmap)
 
-    def forward(self, x: torch.Tensor) -> torch.Tensor:
+    def forward(self, x: torch.Tensor,
+                add: "torch.Tensor | None" = None) -> torch.Tensor:
         from ...ops.irreps_linear import (irreps_linear,
                                           irreps_linear_eligible)
         if irreps_linear_eligible(x, self.weight):
             # single-pass MFMA kernel (csrc/irreps_linear.hip): x and
-            # the whole weight stack LDS-resident, no permute copies
+            # the whole weight stack LDS-resident, no permute copies;
+            # optional residual fused into the epilogue
             return irreps_linear(x.to(torch.bfloat16).contiguous(),
                                  self.weight.to(torch.bfloat16),
-                                 self.lmap, self.bias)
+                                 self.lmap, self.bias,
+                                 add=None if add is None
+                                 else add.to(torch.bfloat16))
         # fallback: one batched GEMM [D](N,C_in)@(C_in,C_out)
         W_m = self.weight.to(x.dtype)[self.lmap]  # [D, C_in, C_out]
         out = torch.bmm(x.permute(2, 0, 1), W_m).permute(1, 2, 0)
@@ -140,6 +144,8 @@ class IrrepsLinear(nn.Module):
             out = torch.cat([
                 out[:, :, :1] + self.bias.to(x.dtype).view(1, -1, 1),
                 out[:, :, 1:]], dim=-1)
+        if add is not None:
+            out = out + add.to(out.dtype)
         return out.contiguous()
 
 

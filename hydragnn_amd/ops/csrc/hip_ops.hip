@@ -702,7 +702,8 @@ torch::Tensor varlen_attention(torch::Tensor Q, torch::Tensor K,
 torch::Tensor irreps_linear(torch::Tensor X, torch::Tensor W,
                             torch::Tensor lmap,
                             c10::optional<torch::Tensor> bias,
-                            bool trans_w);
+                            bool trans_w,
+                            c10::optional<torch::Tensor> add);
 torch::Tensor irreps_linear_gw(torch::Tensor X, torch::Tensor G,
                                torch::Tensor lmap, long L,
                                long nblocks);
@@ -749,7 +750,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "per-l channel-mixing MFMA linear (HIP)",
         pybind11::arg("X"), pybind11::arg("W"), pybind11::arg("lmap"),
         pybind11::arg("bias") = pybind11::none(),
-        pybind11::arg("trans_w") = false);
+        pybind11::arg("trans_w") = false,
+        pybind11::arg("add") = pybind11::none());
   m.def("irreps_linear_gw", &irreps_linear_gw,
         "irreps-linear weight-grad partials (HIP)",
         pybind11::arg("X"), pybind11::arg("G"), pybind11::arg("lmap"),

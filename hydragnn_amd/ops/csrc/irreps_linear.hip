@@ -37,6 +37,7 @@ __global__ __launch_bounds__(256) void irreps_linear_kernel(
     const __hip_bfloat16* __restrict__ X,   // [N, Cin, D]
     const __hip_bfloat16* __restrict__ W,   // [L, Cin, Cout] ([L,Cout,Cin] if TRANS_W)
     const float* __restrict__ bias,         // [Cout] or nullptr (m=0 only)
+    const __hip_bfloat16* __restrict__ add, // [N, Cout, D] residual or nullptr
     __hip_bfloat16* __restrict__ out,       // [N, Cout, D]
     const long* __restrict__ lmap,          // [D] -> l index
     long N, int Cin, int Cout, int D, int L) {
@@ -123,15 +124,24 @@ __global__ __launch_bounds__(256) void irreps_linear_kernel(
             afrag, bfrag, acc[m], 0, 0, 0);
       }
     }
-    // epilogue: C layout col = lane&15, row = (lane>>4)*4 + reg
+    // epilogue: C layout col = lane&15, row = (lane>>4)*4 + reg;
+    // optional fused residual add (saves a full elementwise pass)
     int col = ct * 16 + (lane & 15);
     float b = bias != nullptr ? bias[col] : 0.f;
     for (int reg = 0; reg < 4; ++reg) {
       long row = n0 + rt * 16 + (lane >> 4) * 4 + reg;
       if (row < N) {
-        __hip_bfloat16* op = out + (row * Cout + col) * D;
-        for (int m = 0; m < D; ++m)
-          op[m] = __float2bfloat16(acc[m][reg] + (m == 0 ? b : 0.f));
+        long off = (row * Cout + col) * D;
+        __hip_bfloat16* op = out + off;
+        if (add != nullptr) {
+          const __hip_bfloat16* rp = add + off;
+          for (int m = 0; m < D; ++m)
+            op[m] = __float2bfloat16(acc[m][reg] + (m == 0 ? b : 0.f) +
+                                     __bfloat162float(rp[m]));
+        } else {
+          for (int m = 0; m < D; ++m)
+            op[m] = __float2bfloat16(acc[m][reg] + (m == 0 ? b : 0.f));
+        }
       }
     }
   }
@@ -246,7 +256,8 @@ __global__ __launch_bounds__(256) void irreps_linear_gw_kernel(
 torch::Tensor irreps_linear(torch::Tensor X, torch::Tensor W,
                             torch::Tensor lmap,
                             c10::optional<torch::Tensor> bias,
-                            bool trans_w) {
+                            bool trans_w,
+                            c10::optional<torch::Tensor> add) {
   TORCH_CHECK(X.is_cuda() && X.is_contiguous());
   TORCH_CHECK(W.is_cuda() && W.is_contiguous());
   TORCH_CHECK(X.scalar_type() == at::ScalarType::BFloat16 &&
@@ -272,6 +283,15 @@ torch::Tensor irreps_linear(torch::Tensor X, torch::Tensor W,
     bias_f = bias->to(torch::kFloat).contiguous();
     bias_ptr = bias_f.data_ptr<float>();
   }
+  const __hip_bfloat16* add_ptr = nullptr;
+  torch::Tensor add_c;
+  if (add.has_value()) {
+    add_c = add->contiguous();
+    TORCH_CHECK(add_c.sizes() == out.sizes() &&
+                add_c.scalar_type() == at::ScalarType::BFloat16,
+                "irreps_linear residual shape/dtype mismatch");
+    add_ptr = reinterpret_cast<const __hip_bfloat16*>(add_c.data_ptr());
+  }
   auto lmap_c = lmap.contiguous();
   dim3 grid((N + BM - 1) / BM);
   auto stream = at::hip::getCurrentHIPStream().stream();
@@ -279,7 +299,7 @@ torch::Tensor irreps_linear(torch::Tensor X, torch::Tensor W,
     hipLaunchKernelGGL(kern, grid, dim3(256), lds_bytes, stream,
                        reinterpret_cast<const __hip_bfloat16*>(X.data_ptr()),
                        reinterpret_cast<const __hip_bfloat16*>(W.data_ptr()),
-                       bias_ptr,
+                       bias_ptr, add_ptr,
                        reinterpret_cast<__hip_bfloat16*>(out.data_ptr()),
                        lmap_c.data_ptr<long>(), N, Cin, Cout, D, L);
   };

@@ -43,10 +43,10 @@ class _IrrepsLinearFn(torch.autograd.Function):
     Wk[i,o] = W[l,i,o] (trans_w=False) or W[l,o,i] (trans_w=True)."""
 
     @staticmethod
-    def forward(ctx, x, W, lmap, bias, trans_w):
+    def forward(ctx, x, W, lmap, bias, trans_w, add=None):
         ext = get_extension(required=True)
         out = ext.irreps_linear(x.contiguous(), W.contiguous(), lmap,
-                                bias, trans_w)
+                                bias, trans_w, add)
         ctx.save_for_backward(x, W, lmap, bias)
         ctx.trans_w = trans_w
         return out
@@ -57,6 +57,8 @@ class _IrrepsLinearFn(torch.autograd.Function):
         trans_w = ctx.trans_w
         g = g.contiguous()
         gx = gw = gb = None
+        ga = g if (len(ctx.needs_input_grad) > 5
+                   and ctx.needs_input_grad[5]) else None
         if ctx.needs_input_grad[0]:
             gx = _IrrepsLinearFn.apply(g, W, lmap, None, not trans_w)
         if ctx.needs_input_grad[1]:
@@ -81,14 +83,15 @@ class _IrrepsLinearFn(torch.autograd.Function):
             gw = gw.to(W.dtype)
         if bias is not None and ctx.needs_input_grad[3]:
             gb = g[:, :, 0].sum(0).to(bias.dtype)
-        return gx, gw, None, gb, None
+        return gx, gw, None, gb, None, ga
 
 
 def irreps_linear(x: torch.Tensor, W: torch.Tensor, lmap: torch.Tensor,
-                  bias=None) -> torch.Tensor:
+                  bias=None, add=None) -> torch.Tensor:
     """[N, Cin, D] x, [L, Cin, Cout] W -> [N, Cout, D] on the MFMA
-    kernel (bf16)."""
-    return _IrrepsLinearFn.apply(x, W, lmap, bias, False)
+    kernel (bf16); `add` is an optional residual fused into the
+    epilogue (its gradient is the identity)."""
+    return _IrrepsLinearFn.apply(x, W, lmap, bias, False, add)
 
 
 def irreps_linear_eligible(x: torch.Tensor, W: torch.Tensor) -> bool:

@@ -150,6 +150,12 @@ def test_irreps_linear_matches_bmm(N, Cin, Cout, lmax):
     ref = _irreps_ref(x.float(), W.float(), lmap, bias)
     rel = (out.float() - ref).abs().max() / ref.abs().max()
     assert rel < 2e-2, f"rel {rel:.3e}"
+    # fused residual epilogue
+    res = _asym((N, Cout, D), 7).contiguous()
+    out2 = irreps_linear(x, W, lmap, bias, add=res)
+    rel2 = ((out2.float() - (ref + res.float())).abs().max()
+            / ref.abs().max())
+    assert rel2 < 2e-2, f"residual rel {rel2:.3e}"
 
 
 def test_irreps_linear_grads_match_reference():
