@@ -34,6 +34,11 @@ __device__ inline float to_f32<__hip_bfloat16>(__hip_bfloat16 v) {
 }
 template <typename T>
 __device__ inline T from_f32(float v) { return (T)v; }
+
+// accumulator type: fp32 for bf16/half/float inputs, fp64 for double
+// (fp64 ETP support: slices and channel reductions keep full width)
+template <typename T> struct acc_of { using type = float; };
+template <> struct acc_of<double> { using type = double; };
 template <>
 __device__ inline __hip_bfloat16 from_f32<__hip_bfloat16>(float v) {
   return __float2bfloat16(v);
@@ -59,10 +64,11 @@ __global__ void etp_general_kernel(
     const long* __restrict__ ai, const long* __restrict__ bi,
     const long* __restrict__ ci) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
+  using ACC = typename acc_of<T>::type;
   const int stride = ((da + db + dg + do_) | 1);  // odd word stride
-  float* slices = reinterpret_cast<float*>(smem);
+  ACC* slices = reinterpret_cast<ACC*>(smem);
   int4* ent_lds = reinterpret_cast<int4*>(
-      smem + (size_t)blockDim.x * stride * 4);
+      smem + (size_t)blockDim.x * stride * sizeof(ACC));
   float* coef_lds = reinterpret_cast<float*>(ent_lds + n_ent);
 
   // stage the entry table once per block
@@ -72,11 +78,11 @@ __global__ void etp_general_kernel(
   }
 
   long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
-  float* my = slices + (size_t)threadIdx.x * stride;
-  float* ma = my;
-  float* mb = ma + da;
-  float* mc = mb + db;
-  float* mo = mc + dg;
+  ACC* my = slices + (size_t)threadIdx.x * stride;
+  ACC* ma = my;
+  ACC* mb = ma + da;
+  ACC* mc = mb + db;
+  ACC* mo = mc + dg;
   if (i < NC) {
     long e = i / nch;
     int c = (int)(i - e * nch);
@@ -86,9 +92,9 @@ __global__ void etp_general_kernel(
     const T* ap = A + (ea * nch + c) * da;
     const T* bp = B + eb * db;
     const T* cp = C + (ec * nch + c) * dg;
-    for (int k = 0; k < da; ++k) ma[k] = to_f32(ap[k]);
-    for (int k = 0; k < db; ++k) mb[k] = to_f32(bp[k]);
-    for (int k = 0; k < dg; ++k) mc[k] = to_f32(cp[k]);
+    for (int k = 0; k < da; ++k) ma[k] = (ACC)ap[k];
+    for (int k = 0; k < db; ++k) mb[k] = (ACC)bp[k];
+    for (int k = 0; k < dg; ++k) mc[k] = (ACC)cp[k];
     for (int k = 0; k < do_; ++k) mo[k] = 0.f;
   }
   __syncthreads();
@@ -98,7 +104,7 @@ __global__ void etp_general_kernel(
       mo[q.w] += coef_lds[k] * ma[q.x] * mb[q.y] * mc[q.z];
     }
     T* op = out + i * do_;
-    for (int k = 0; k < do_; ++k) op[k] = from_f32<T>(mo[k]);
+    for (int k = 0; k < do_; ++k) op[k] = (T)mo[k];
   }
 }
 
@@ -119,17 +125,18 @@ __global__ void etp_general_l1_kernel(
     const long* __restrict__ ai, const long* __restrict__ bi,
     const long* __restrict__ ci) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
+  using ACC = typename acc_of<T>::type;
   const int stride = do_ | 1;
-  float* slices = reinterpret_cast<float*>(smem);
+  ACC* slices = reinterpret_cast<ACC*>(smem);
   int4* ent_lds = reinterpret_cast<int4*>(
-      smem + (size_t)blockDim.x * stride * 4);
+      smem + (size_t)blockDim.x * stride * sizeof(ACC));
   float* coef_lds = reinterpret_cast<float*>(ent_lds + n_ent);
   for (int k = threadIdx.x; k < n_ent; k += blockDim.x) {
     ent_lds[k] = entries[k];
     coef_lds[k] = coefs[k];
   }
   long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
-  float* mo = slices + (size_t)threadIdx.x * stride;
+  ACC* mo = slices + (size_t)threadIdx.x * stride;
   __syncthreads();
   if (i >= NC) return;
   long e = i / nch;
@@ -143,11 +150,11 @@ __global__ void etp_general_l1_kernel(
   for (int k = 0; k < do_; ++k) mo[k] = 0.f;
   for (int k = 0; k < n_ent; ++k) {
     int4 q = ent_lds[k];
-    mo[q.w] += coef_lds[k] * to_f32(ap[q.x]) * to_f32(bp[q.y]) *
-               to_f32(cp[q.z]);
+    mo[q.w] += coef_lds[k] * (ACC)ap[q.x] * (ACC)bp[q.y] *
+               (ACC)cp[q.z];
   }
   T* op = out + i * do_;
-  for (int k = 0; k < do_; ++k) op[k] = from_f32<T>(mo[k]);
+  for (int k = 0; k < do_; ++k) op[k] = (T)mo[k];
 }
 
 // out[e,b] = sum_c sum over entries(coef, a, b, g, o) of
@@ -159,15 +166,17 @@ __global__ void etp_general_l1_kernel(
 template <typename T>
 __global__ void etp_reduce_kernel(
     const T* __restrict__ A, const T* __restrict__ C,
-    const T* __restrict__ D, float* __restrict__ out,
+    const T* __restrict__ D,
+    typename acc_of<T>::type* __restrict__ out,
     const int4* __restrict__ entries,
     const float* __restrict__ coefs, int n_ent,
     long E, int nch, int da, int db, int dg, int do_) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
+  using ACC = typename acc_of<T>::type;
   const int stride = ((da + dg + do_ + db) | 1);
-  float* slices = reinterpret_cast<float*>(smem);
+  ACC* slices = reinterpret_cast<ACC*>(smem);
   int4* ent_lds = reinterpret_cast<int4*>(
-      smem + (size_t)blockDim.x * stride * 4);
+      smem + (size_t)blockDim.x * stride * sizeof(ACC));
   float* coef_lds = reinterpret_cast<float*>(ent_lds + n_ent);
   for (int k = threadIdx.x; k < n_ent; k += blockDim.x) {
     ent_lds[k] = entries[k];
@@ -175,18 +184,18 @@ __global__ void etp_reduce_kernel(
   }
   long NC = E * nch;
   long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
-  float* my = slices + (size_t)threadIdx.x * stride;
-  float* ma = my;
-  float* mc = ma + da;
-  float* md = mc + dg;
-  float* mb = md + do_;
+  ACC* my = slices + (size_t)threadIdx.x * stride;
+  ACC* ma = my;
+  ACC* mc = ma + da;
+  ACC* md = mc + dg;
+  ACC* mb = md + do_;
   if (i < NC) {
     const T* ap = A + i * da;
     const T* cp = C + i * dg;
     const T* dp = D + i * do_;
-    for (int k = 0; k < da; ++k) ma[k] = to_f32(ap[k]);
-    for (int k = 0; k < dg; ++k) mc[k] = to_f32(cp[k]);
-    for (int k = 0; k < do_; ++k) md[k] = to_f32(dp[k]);
+    for (int k = 0; k < da; ++k) ma[k] = (ACC)ap[k];
+    for (int k = 0; k < dg; ++k) mc[k] = (ACC)cp[k];
+    for (int k = 0; k < do_; ++k) md[k] = (ACC)dp[k];
     for (int k = 0; k < db; ++k) mb[k] = 0.f;
   }
   __syncthreads();
@@ -201,7 +210,7 @@ __global__ void etp_reduce_kernel(
     // a wave spans exactly one edge's channels: shuffle-reduce each b
     // across the 64 lanes, then ONE atomic per b per wave
     for (int b = 0; b < db; ++b) {
-      float v = (i < NC) ? mb[b] : 0.f;
+      ACC v = (i < NC) ? mb[b] : (ACC)0;
       for (int off = 32; off >= 1; off >>= 1)
         v += __shfl_down(v, off, 64);
       if ((threadIdx.x % 64) == 0 && i < NC)
@@ -230,10 +239,11 @@ __global__ void etp_nodesum_kernel(
     const long* __restrict__ ai, const long* __restrict__ bi,
     const long* __restrict__ ci) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
+  using ACC = typename acc_of<T>::type;
   const int stride = ((da + db + dg + do_) | 1);
-  float* slices = reinterpret_cast<float*>(smem);
+  ACC* slices = reinterpret_cast<ACC*>(smem);
   int4* ent_lds = reinterpret_cast<int4*>(
-      smem + (size_t)blockDim.x * stride * 4);
+      smem + (size_t)blockDim.x * stride * sizeof(ACC));
   float* coef_lds = reinterpret_cast<float*>(ent_lds + n_ent);
   for (int k = threadIdx.x; k < n_ent; k += blockDim.x) {
     ent_lds[k] = entries[k];
@@ -241,11 +251,11 @@ __global__ void etp_nodesum_kernel(
   }
   long RC = R * nch;
   long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
-  float* my = slices + (size_t)threadIdx.x * stride;
-  float* ma = my;
-  float* mb = ma + da;
-  float* mc = mb + db;
-  float* mo = mc + dg;
+  ACC* my = slices + (size_t)threadIdx.x * stride;
+  ACC* ma = my;
+  ACC* mb = ma + da;
+  ACC* mc = mb + db;
+  ACC* mo = mc + dg;
   __syncthreads();
   if (i >= RC) return;
   long r = i / nch;
@@ -259,33 +269,35 @@ __global__ void etp_nodesum_kernel(
     const T* ap = A + (ea * nch + c) * da;
     const T* bp = B + eb * db;
     const T* cp = C + (ec * nch + c) * dg;
-    for (int k = 0; k < da; ++k) ma[k] = to_f32(ap[k]);
-    for (int k = 0; k < db; ++k) mb[k] = to_f32(bp[k]);
-    for (int k = 0; k < dg; ++k) mc[k] = to_f32(cp[k]);
+    for (int k = 0; k < da; ++k) ma[k] = (ACC)ap[k];
+    for (int k = 0; k < db; ++k) mb[k] = (ACC)bp[k];
+    for (int k = 0; k < dg; ++k) mc[k] = (ACC)cp[k];
     for (int k = 0; k < n_ent; ++k) {
       int4 q = ent_lds[k];
       mo[q.w] += coef_lds[k] * ma[q.x] * mb[q.y] * mc[q.z];
     }
   }
   T* op = out + i * do_;
-  for (int k = 0; k < do_; ++k) op[k] = from_f32<T>(mo[k]);
+  for (int k = 0; k < do_; ++k) op[k] = (T)mo[k];
 }
 
 // etp_reduce with per-slot row indices
 template <typename T>
 __global__ void etp_reduce_idx_kernel(
     const T* __restrict__ A, const T* __restrict__ C,
-    const T* __restrict__ D, float* __restrict__ out,
+    const T* __restrict__ D,
+    typename acc_of<T>::type* __restrict__ out,
     const int4* __restrict__ entries,
     const float* __restrict__ coefs, int n_ent,
     long E, int nch, int da, int db, int dg, int do_,
     const long* __restrict__ ai, const long* __restrict__ ci,
     const long* __restrict__ di) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
+  using ACC = typename acc_of<T>::type;
   const int stride = ((da + dg + do_ + db) | 1);
-  float* slices = reinterpret_cast<float*>(smem);
+  ACC* slices = reinterpret_cast<ACC*>(smem);
   int4* ent_lds = reinterpret_cast<int4*>(
-      smem + (size_t)blockDim.x * stride * 4);
+      smem + (size_t)blockDim.x * stride * sizeof(ACC));
   float* coef_lds = reinterpret_cast<float*>(ent_lds + n_ent);
   for (int k = threadIdx.x; k < n_ent; k += blockDim.x) {
     ent_lds[k] = entries[k];
@@ -293,11 +305,11 @@ __global__ void etp_reduce_idx_kernel(
   }
   long NC = E * nch;
   long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
-  float* my = slices + (size_t)threadIdx.x * stride;
-  float* ma = my;
-  float* mc = ma + da;
-  float* md = mc + dg;
-  float* mb = md + do_;
+  ACC* my = slices + (size_t)threadIdx.x * stride;
+  ACC* ma = my;
+  ACC* mc = ma + da;
+  ACC* md = mc + dg;
+  ACC* mb = md + do_;
   if (i < NC) {
     long e = i / nch;
     int c = (int)(i - e * nch);
@@ -307,9 +319,9 @@ __global__ void etp_reduce_idx_kernel(
     const T* ap = A + (ea * nch + c) * da;
     const T* cp = C + (ec * nch + c) * dg;
     const T* dp = D + (ed * nch + c) * do_;
-    for (int k = 0; k < da; ++k) ma[k] = to_f32(ap[k]);
-    for (int k = 0; k < dg; ++k) mc[k] = to_f32(cp[k]);
-    for (int k = 0; k < do_; ++k) md[k] = to_f32(dp[k]);
+    for (int k = 0; k < da; ++k) ma[k] = (ACC)ap[k];
+    for (int k = 0; k < dg; ++k) mc[k] = (ACC)cp[k];
+    for (int k = 0; k < do_; ++k) md[k] = (ACC)dp[k];
     for (int k = 0; k < db; ++k) mb[k] = 0.f;
   }
   __syncthreads();
@@ -322,7 +334,7 @@ __global__ void etp_reduce_idx_kernel(
   long e = i / nch;
   if (nch % 64 == 0) {
     for (int b = 0; b < db; ++b) {
-      float v = (i < NC) ? mb[b] : 0.f;
+      ACC v = (i < NC) ? mb[b] : (ACC)0;
       for (int off = 32; off >= 1; off >>= 1)
         v += __shfl_down(v, off, 64);
       if ((threadIdx.x % 64) == 0 && i < NC)
@@ -371,10 +383,11 @@ torch::Tensor etp_general(torch::Tensor A, torch::Tensor B, torch::Tensor C,
   if (NC == 0) return out;
   int n_ent = entries.size(0);
   int block = etp_block_size();
+  size_t accs = A.scalar_type() == at::ScalarType::Double ? 8 : 4;
   int stride = (da + db + dg + (int)do_) | 1;
-  size_t lds_full = (size_t)block * stride * 4 + n_ent * 20;
+  size_t lds_full = (size_t)block * stride * accs + n_ent * 20;
   int stride_l1 = (int)do_ | 1;
-  size_t lds_l1 = (size_t)block * stride_l1 * 4 + n_ent * 20;
+  size_t lds_l1 = (size_t)block * stride_l1 * accs + n_ent * 20;
   // Measured A/B (default bench, b1024): staged 32.9k g/s vs
   // L1-operand 30.9k — the higher occupancy does NOT pay for the 3
   // L1 loads + converts per entry, so staged is the default; the L1
@@ -387,7 +400,7 @@ torch::Tensor etp_general(torch::Tensor A, torch::Tensor B, torch::Tensor C,
   size_t lds_bytes = use_l1 ? lds_l1 : lds_full;
   TORCH_CHECK(lds_bytes <= 150 * 1024, "etp LDS budget exceeded");
   long blocks = (NC + block - 1) / block;
-  // fp64 is routed to the eager path in Python (float LDS staging here)
+  // fp64 runs with double LDS slices (acc_of<double>); others stage fp32
   AT_DISPATCH_FLOATING_TYPES_AND2(
       at::ScalarType::BFloat16, at::ScalarType::Half, A.scalar_type(),
       "etp_general", [&] {
@@ -424,7 +437,8 @@ torch::Tensor etp_nodesum(torch::Tensor A, torch::Tensor B,
   int n_ent = entries.size(0);
   int block = etp_block_size();
   int stride = (da + db + dg + (int)do_) | 1;
-  size_t lds_bytes = (size_t)block * stride * 4 + n_ent * 20;
+  size_t accs = A.scalar_type() == at::ScalarType::Double ? 8 : 4;
+  size_t lds_bytes = (size_t)block * stride * accs + n_ent * 20;
   TORCH_CHECK(lds_bytes <= 150 * 1024, "etp LDS budget exceeded");
   long blocks = (RC + block - 1) / block;
   AT_DISPATCH_FLOATING_TYPES_AND2(
@@ -455,12 +469,16 @@ torch::Tensor etp_reduce(torch::Tensor A, torch::Tensor C, torch::Tensor D,
   int nch = A.size(1);
   int da = A.size(2), dg = C.size(2), do_ = D.size(2);
   TORCH_CHECK(db <= 12, "etp_reduce db limit");
-  auto out = torch::zeros({E, db}, A.options().dtype(torch::kFloat));
+  auto out = torch::zeros(
+      {E, db}, A.options().dtype(
+          A.scalar_type() == at::ScalarType::Double ? torch::kDouble
+                                                    : torch::kFloat));
   if (E == 0) return out.to(A.scalar_type());
   int n_ent = entries.size(0);
   int block = etp_block_size();
   int stride = (da + dg + do_ + (int)db) | 1;
-  size_t lds_bytes = (size_t)block * stride * 4 + n_ent * 20;
+  size_t accs = A.scalar_type() == at::ScalarType::Double ? 8 : 4;
+  size_t lds_bytes = (size_t)block * stride * accs + n_ent * 20;
   TORCH_CHECK(lds_bytes <= 150 * 1024, "etp_reduce LDS budget exceeded");
   long blocks = (E * nch + block - 1) / block;
   AT_DISPATCH_FLOATING_TYPES_AND2(
@@ -470,7 +488,7 @@ torch::Tensor etp_reduce(torch::Tensor A, torch::Tensor C, torch::Tensor D,
             etp_reduce_idx_kernel<scalar_t>, dim3(blocks), dim3(block),
             lds_bytes, etp_stream(), A.data_ptr<scalar_t>(),
             C.data_ptr<scalar_t>(),
-            D.data_ptr<scalar_t>(), out.data_ptr<float>(),
+            D.data_ptr<scalar_t>(), out.data_ptr<typename acc_of<scalar_t>::type>(),
             reinterpret_cast<const int4*>(entries.data_ptr<int>()),
             coefs.data_ptr<float>(), n_ent, E, nch, da, (int)db, dg, do_,
             idx_ptr(ai), idx_ptr(ci), idx_ptr(di));

@@ -21,7 +21,8 @@ import torch
 from ._extension import get_extension, use_eager
 from .scatter import _rowptr_from_sorted
 
-_KERNEL_DTYPES = (torch.float32, torch.bfloat16, torch.float16)
+_KERNEL_DTYPES = (torch.float32, torch.bfloat16, torch.float16,
+                  torch.float64)
 
 
 class ETPTable:
@@ -95,11 +96,13 @@ def _kernel_ok(table: ETPTable, *tensors) -> bool:
     # at the FULL-staging footprint even though etp_general can fall
     # to its L1-operand variant: the gradient kernels (etp_reduce)
     # still stage fully, and this predicate gates the whole family.
-    lds_bytes = 256 * (da + db + dg + do + 1) * 4 + \
+    # fp64 slices are 8 bytes (full-precision accumulation in LDS).
+    t = tensors[0]
+    acc = 8 if t.dtype == torch.float64 else 4
+    lds_bytes = 256 * (da + db + dg + do + 1) * acc + \
         table.entries.shape[0] * 20
     if lds_bytes > 150 * 1024 or max(da, db, dg, do) > 192:
         return False
-    t = tensors[0]
     return (t.is_cuda and t.dtype in _KERNEL_DTYPES and not use_eager())
 
 

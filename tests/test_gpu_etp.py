@@ -128,3 +128,32 @@ def test_mace_gpu_matches_cpu():
         (e_cpu - e_gpu).abs().max())
     assert torch.allclose(f_cpu, f_gpu, atol=1e-3, rtol=1e-3), (
         (f_cpu - f_gpu).abs().max())
+
+
+def test_etp_fp64_full_precision():
+    """fp64 ETP runs on the kernels with double LDS accumulation: match
+    the fp64 dense reference to ~1e-12 (not fp32-level error)."""
+    torch.manual_seed(0)
+    tab = _table()
+    E, C = 300, 32
+    A = torch.randn(E, C, 4, device="cuda", dtype=torch.float64,
+                    requires_grad=True)
+    B = torch.randn(E, 9, device="cuda", dtype=torch.float64,
+                    requires_grad=True)
+    Cw = torch.randn(E, C, 6, device="cuda", dtype=torch.float64)
+    out = etp_general(A, B, Cw, tab)
+    ref = _dense_general(A.detach(), B.detach(), Cw, tab)
+    assert out.dtype == torch.float64
+    assert (out - ref).abs().max() < 1e-12
+    ga, gb = torch.autograd.grad(out.square().sum(), (A, B),
+                                 create_graph=True)
+    Ar = A.detach().requires_grad_(True)
+    Br = B.detach().requires_grad_(True)
+    gar, gbr = torch.autograd.grad(
+        _dense_general(Ar, Br, Cw, tab).square().sum(), (Ar, Br),
+        create_graph=True)
+    assert (ga - gar).abs().max() < 1e-10
+    assert (gb - gbr).abs().max() < 1e-10
+    gga = torch.autograd.grad(ga.square().sum(), A)[0]
+    ggar = torch.autograd.grad(gar.square().sum(), Ar)[0]
+    assert (gga - ggar).abs().max() < 1e-9
