@@ -207,3 +207,31 @@ def test_radial_bases():
     assert (ch.abs() <= 1.0 + 1e-6).all()
     g = gaussian_basis(r, torch.linspace(0, 5, 10).view(1, -1), -0.5)
     assert g.shape == (50, 10) and (g <= 1.0).all()
+
+
+def test_formation_enthalpy_two_components():
+    """Reference test_enthalpy pattern: linear-mixing energies of a
+    2-component system give ~zero formation enthalpy against the pure
+    references."""
+    from hydragnn_amd.utils.lsms import get_formation_enthalpy
+    e_pure = {13: -3.5, 29: -4.2}
+    # perfectly linear mixture: E_total = sum n_i * e_pure_i
+    for n_al, n_cu in [(4, 0), (0, 4), (3, 1), (2, 2)]:
+        comp = {13: n_al, 29: n_cu}
+        e_tot = n_al * e_pure[13] + n_cu * e_pure[29]
+        h = get_formation_enthalpy(e_tot, comp, e_pure)
+        assert abs(h) < 1e-10
+    # non-linear mixing shows up as the total excess energy
+    h = get_formation_enthalpy(2 * e_pure[13] + 2 * e_pure[29] - 0.4,
+                               {13: 2, 29: 2}, e_pure)
+    assert h == pytest.approx(-0.4)
+
+
+def test_parse_deepspeed_config():
+    from hydragnn_amd.utils.config import parse_deepspeed_config
+    cfg = {"NeuralNetwork": {"Training": {
+        "batch_size": 16,
+        "Optimizer": {"type": "AdamW", "learning_rate": 5e-4}}}}
+    ds = parse_deepspeed_config(cfg)
+    assert ds["train_batch_size"] == 16
+    assert ds["optimizer"]["params"]["lr"] == 5e-4
