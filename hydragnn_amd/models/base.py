@@ -478,6 +478,14 @@ class Base(Module):
     def encoder_forward(self, data):
         """Embedding + conv stack + pooling (the shared encoder half
         used by MultiTaskModelMP)."""
+        # propagate the collation-time dst-sortedness flag to every
+        # conv so edge aggregations take the deterministic CSR kernel
+        # instead of atomics (host-side attr; capture-safe)
+        es = bool(data.get("edges_sorted_", False))
+        if getattr(self, "_edges_sorted_flag", None) != es:
+            for m in self.modules():
+                m._edges_sorted = es
+            self._edges_sorted_flag = es
         inv_node_feat, equiv_node_feat, conv_args = self._embedding(data)
         batch = data.get("batch")
 

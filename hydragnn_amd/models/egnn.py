@@ -51,7 +51,8 @@ class E_GCL(nn.Module):
         if self.equivariant:
             trans = vec.to(h.dtype) * self.coord_mlp(m)
             pos = pos + scatter(trans, dst, h.shape[0], "mean").to(pos.dtype)
-        agg = scatter(m, dst, h.shape[0], "sum")
+        agg = scatter(m, dst, h.shape[0], "sum",
+                      sorted_index=getattr(self, "_edges_sorted", False))
         h = self.node_mlp(torch.cat([h, agg], dim=-1))
         return h, pos
 

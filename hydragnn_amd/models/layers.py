@@ -52,7 +52,8 @@ class GINConv(nn.Module):
 
     def forward(self, x, edge_index):
         src, dst = edge_index[0], edge_index[1]
-        agg = scatter(gather(x, src), dst, x.shape[0], "sum")
+        agg = scatter(gather(x, src), dst, x.shape[0], "sum",
+                      sorted_index=getattr(self, "_edges_sorted", False))
         return self.nn((1.0 + self.eps) * x + agg)
 
 
@@ -86,7 +87,8 @@ class MFConv(nn.Module):
 
     def forward(self, x, edge_index):
         src, dst = edge_index[0], edge_index[1]
-        agg = scatter(gather(x, src), dst, x.shape[0], "sum")
+        agg = scatter(gather(x, src), dst, x.shape[0], "sum",
+                      sorted_index=getattr(self, "_edges_sorted", False))
         deg = degree(dst, x.shape[0], torch.long).clamp(max=self.max_degree)
         out = x.new_zeros(x.shape[0], self.lins_root[0].out_features)
         for d in deg.unique().tolist():
@@ -132,7 +134,8 @@ class GATv2Conv(nn.Module):
             alpha = torch.nn.functional.dropout(alpha, p=self.dropout)
         msg = gather(xl.reshape(-1, H * C), src).view(-1, H, C) * \
             alpha.unsqueeze(-1)
-        out = scatter(msg.reshape(-1, H * C), dst, x.shape[0], "sum")
+        out = scatter(msg.reshape(-1, H * C), dst, x.shape[0], "sum",
+                      sorted_index=getattr(self, "_edges_sorted", False))
         out = out.view(-1, H, C)
         out = out.reshape(-1, H * C) if self.concat else out.mean(dim=1)
         return out + self.bias
@@ -159,7 +162,8 @@ class CGConv(nn.Module):
                       dim=-1)
         msg = torch.sigmoid(self.lin_f(z)) * torch.nn.functional.softplus(
             self.lin_s(z))
-        agg = scatter(msg, dst, x.shape[0], self.aggr)
+        agg = scatter(msg, dst, x.shape[0], self.aggr,
+                      sorted_index=getattr(self, "_edges_sorted", False))
         return x + agg
 
 
@@ -183,7 +187,10 @@ class DegreeScalerAggregation(nn.Module):
             (((bins + 1).log() * deg).sum() / max(num, 1)).clamp(min=1e-6))
 
     def forward(self, msg, index, dim_size):
-        outs = [scatter(msg, index, dim_size, a) for a in self.aggregators]
+        outs = [scatter(msg, index, dim_size, a,
+                        sorted_index=getattr(self, "_edges_sorted",
+                                             False))
+                for a in self.aggregators]
         out = torch.cat(outs, dim=-1)
         d = degree(index, dim_size, out.dtype).clamp(min=1).view(-1, 1)
         scaled = []

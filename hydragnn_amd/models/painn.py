@@ -56,10 +56,12 @@ class PainnMessage(nn.Module):
             gate_edge_vector.unsqueeze(1) * unit.unsqueeze(-1)
 
         n = node_scalar.shape[0]
-        residual_scalar = scatter(message_scalar, dst, n, "sum")
+        residual_scalar = scatter(message_scalar, dst, n, "sum",
+                                  sorted_index=getattr(self, "_edges_sorted", False))
         residual_vector = scatter(
             message_vector.reshape(-1, 3 * self.node_size), dst, n,
-            "sum").view(-1, 3, self.node_size)
+            "sum", sorted_index=getattr(self, "_edges_sorted", False)
+        ).view(-1, 3, self.node_size)
         return node_scalar + residual_scalar, node_vector + residual_vector
 
 

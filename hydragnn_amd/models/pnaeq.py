@@ -60,7 +60,10 @@ class PNAEqMessage(nn.Module):
         n = node_scalar.shape[0]
         res_s = self.scalar_proj(self.aggr(msg_s, dst, n))
         res_v = scatter(msg_v.reshape(-1, 3 * self.node_size), dst, n,
-                        "sum").view(-1, 3, self.node_size)
+                        "sum",
+                        sorted_index=getattr(self, "_edges_sorted",
+                                             False)
+                        ).view(-1, 3, self.node_size)
         return node_scalar + res_s, node_vector + res_v
 
 
