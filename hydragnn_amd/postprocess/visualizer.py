@@ -98,3 +98,124 @@ class Visualizer:
         fig.tight_layout()
         fig.savefig(os.path.join(self.outdir, "error_hist.png"), dpi=120)
         plt.close(fig)
+
+    @staticmethod
+    def add_identity(axes, *line_args, **line_kwargs):
+        """y = x guide that tracks axis-limit changes (reference
+        visualizer.py:612)."""
+        (identity,) = axes.plot([], [], *line_args, **line_kwargs)
+
+        def callback(ax):
+            lo = max(ax.get_xlim()[0], ax.get_ylim()[0])
+            hi = min(ax.get_xlim()[1], ax.get_ylim()[1])
+            identity.set_data([lo, hi], [lo, hi])
+
+        callback(axes)
+        axes.callbacks.connect("xlim_changed", callback)
+        axes.callbacks.connect("ylim_changed", callback)
+        return axes
+
+    def create_plot_global_analysis(self, true_values, predicted_values,
+                                    output_names=None, iepoch=None):
+        """2D-density parity + conditional mean-|error| per head
+        (reference create_plot_global_analysis pattern)."""
+        if not _HAS_MPL:
+            return
+        import numpy as np
+        n = len(true_values)
+        fig, axes = plt.subplots(2, max(n, 1),
+                                 figsize=(5 * max(n, 1), 8),
+                                 squeeze=False)
+        for ihead in range(n):
+            t = true_values[ihead].detach().cpu().flatten().numpy()
+            p = predicted_values[ihead].detach().cpu().flatten().numpy()
+            if t.size == 0:
+                continue
+            ax = axes[0][ihead]
+            h, xe, ye = np.histogram2d(t, p, bins=40)
+            xc, yc = 0.5 * (xe[:-1] + xe[1:]), 0.5 * (ye[:-1] + ye[1:])
+            if h.max() > 0:
+                ax.contourf(xc, yc, h.T + 1e-12, levels=12,
+                            cmap="viridis")
+            self.add_identity(ax, "w--", lw=1)
+            ax.set_xlabel("true")
+            ax.set_ylabel("predicted")
+            # conditional mean |error| over true-value bins
+            ax2 = axes[1][ihead]
+            bins = np.linspace(t.min(), t.max() + 1e-12, 21)
+            which = np.digitize(t, bins) - 1
+            err = np.abs(p - t)
+            cm = [err[which == b].mean() if (which == b).any() else 0.0
+                  for b in range(20)]
+            ax2.plot(0.5 * (bins[:-1] + bins[1:]), cm, "o-")
+            ax2.set_xlabel("true")
+            ax2.set_ylabel("mean |error|")
+        fig.tight_layout()
+        suffix = f"_epoch{iepoch}" if iepoch is not None else ""
+        fig.savefig(os.path.join(self.outdir,
+                                 f"global_analysis{suffix}.png"), dpi=120)
+        plt.close(fig)
+
+    def create_error_histogram_per_node(self, true_values,
+                                        predicted_values, node_counts,
+                                        output_names=None):
+        """Error distribution grouped by graph size (reference
+        create_error_histogram_per_node pattern)."""
+        if not _HAS_MPL:
+            return
+        import numpy as np
+        t = true_values[0].detach().cpu().flatten().numpy()
+        p = predicted_values[0].detach().cpu().flatten().numpy()
+        nc = torch.as_tensor(node_counts).cpu().flatten().numpy()
+        if t.size == 0 or nc.size != t.size:
+            return
+        fig, ax = plt.subplots(figsize=(6, 4))
+        sizes = np.unique(nc)
+        for s in sizes[:8]:
+            sel = nc == s
+            ax.hist((p - t)[sel], bins=30, histtype="step",
+                    label=f"N={int(s)}")
+        ax.set_xlabel("error")
+        ax.legend(fontsize=7)
+        fig.tight_layout()
+        fig.savefig(os.path.join(self.outdir, "error_hist_per_size.png"),
+                    dpi=120)
+        plt.close(fig)
+
+    def create_parity_plot_vector(self, true_values, predicted_values,
+                                  components=("x", "y", "z"),
+                                  name="forces"):
+        """Per-component parity for vector heads (reference
+        create_parity_plot_vector pattern)."""
+        if not _HAS_MPL:
+            return
+        t = true_values.detach().cpu().reshape(-1, len(components))
+        p = predicted_values.detach().cpu().reshape(-1, len(components))
+        fig, axes = plt.subplots(1, len(components),
+                                 figsize=(5 * len(components), 4))
+        for i, c in enumerate(components):
+            ax = axes[i] if len(components) > 1 else axes
+            ax.scatter(t[:, i].numpy(), p[:, i].numpy(), s=3, alpha=0.4)
+            self.add_identity(ax, "k--", lw=1)
+            rmse = float(torch.sqrt(((t[:, i] - p[:, i]) ** 2).mean()))
+            ax.set_title(f"{name}.{c} (RMSE {rmse:.4f})")
+            ax.set_xlabel("true")
+            ax.set_ylabel("predicted")
+        fig.tight_layout()
+        fig.savefig(os.path.join(self.outdir, f"parity_{name}.png"),
+                    dpi=120)
+        plt.close(fig)
+
+    def num_nodes_plot(self, node_counts):
+        """Graph-size histogram of the dataset (reference
+        num_nodes_plot)."""
+        if not _HAS_MPL:
+            return
+        nc = torch.as_tensor(node_counts).cpu().flatten().numpy()
+        fig, ax = plt.subplots(figsize=(5, 4))
+        ax.hist(nc, bins=min(40, max(len(set(nc.tolist())), 2)))
+        ax.set_xlabel("nodes per graph")
+        ax.set_ylabel("count")
+        fig.tight_layout()
+        fig.savefig(os.path.join(self.outdir, "num_nodes.png"), dpi=120)
+        plt.close(fig)

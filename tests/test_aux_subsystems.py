@@ -235,3 +235,30 @@ def test_parse_deepspeed_config():
     ds = parse_deepspeed_config(cfg)
     assert ds["train_batch_size"] == 16
     assert ds["optimizer"]["params"]["lr"] == 5e-4
+
+
+def test_visualizer_plot_suite(tmp_path):
+    """All visualizer plot types render to files (reference
+    visualizer coverage pattern)."""
+    pytest.importorskip("matplotlib")
+    from hydragnn_amd.postprocess.visualizer import Visualizer
+    torch.manual_seed(0)
+    viz = Visualizer("viz_test", num_heads=2, path=str(tmp_path))
+    t = [torch.randn(50, 1), torch.randn(50, 1)]
+    p = [x + 0.1 * torch.randn_like(x) for x in t]
+    viz.add_history(1.0, 1.1, 1.2)
+    viz.add_history(0.5, 0.6, 0.7)
+    viz.plot_history()
+    viz.create_scatter_plots(t, p)
+    viz.create_error_histograms(t, p)
+    viz.create_plot_global_analysis(t, p)
+    nc = torch.randint(4, 8, (50,))
+    viz.create_error_histogram_per_node(t, p, nc)
+    viz.create_parity_plot_vector(torch.randn(30, 3),
+                                  torch.randn(30, 3))
+    viz.num_nodes_plot(nc)
+    out = tmp_path / "viz_test"
+    for f in ["history.png", "scatter.png", "error_hist.png",
+              "global_analysis.png", "error_hist_per_size.png",
+              "parity_forces.png", "num_nodes.png"]:
+        assert (out / f).exists(), f
