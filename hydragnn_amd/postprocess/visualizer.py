@@ -1,7 +1,8 @@
 """Matplotlib visualization suite (reference:
 hydragnn/postprocess/visualizer.py, 740 LoC): predicted-vs-true scatter
-per head, training history curves, error histograms. Thin but
-API-compatible; degrades to no-op without matplotlib."""
+per head, training history curves, error histograms, 2D-density parity
+analysis, per-size error breakdowns, vector-component parity and
+graph-size statistics; degrades to no-op without matplotlib."""
 
 from __future__ import annotations
 
