@@ -68,3 +68,10 @@ def test_open_catalyst_example():
     r = _run("examples/open_catalyst/open_catalyst.py",
              "--num_epoch", "2", "--num_samples", "12")
     assert r.returncode == 0, r.stderr[-2000:]
+
+
+def test_qm9_hpo_example():
+    r = _run("examples/qm9_hpo/qm9_hpo.py", "--trials", "2",
+             "--epochs", "1", "--samples", "48")
+    assert r.returncode == 0, r.stderr[-2000:]
+    assert "best params" in r.stdout
