@@ -98,3 +98,8 @@ class EGCLStack(Base):
 
     def __str__(self):
         return "EGCLStack"
+
+
+def unsorted_segment_sum(data, segment_ids, num_segments):
+    """Reference EGCLStack helper name; HIP scatter underneath."""
+    return scatter(data, segment_ids, num_segments, "sum")

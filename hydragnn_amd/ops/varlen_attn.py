@@ -6,11 +6,11 @@ wastes both HBM traffic and FLOPs.  The HIP kernel
 (csrc/varlen_attn.hip) runs one (graph, head) per workgroup with K/V
 staged in LDS and an online softmax over exactly the graph's nodes.
 
-Backward recomputes through a pure-torch reference (block-diagonal
-attention composed from dense SDPA), so first and second order
-gradients are exact without a hand-written backward kernel; the
-recompute only triggers on the GPS training path, which is not the
-headline bench.
+Backward recomputes through a pure-torch reference (dense-batch
+explicit-softmax attention with a finite mask bias — double
+differentiable on every backend), so first and second order gradients
+are exact without a hand-written backward kernel; the recompute only
+triggers on the GPS training path, which is not the headline bench.
 """
 
 from __future__ import annotations

@@ -12,6 +12,7 @@ same number of times.
 from __future__ import annotations
 
 import random
+from dataclasses import dataclass
 from typing import Iterator, List
 
 import torch.distributed as dist
@@ -127,3 +128,33 @@ class DistributedCostAwareBatchSampler(CostAwareBatchSampler):
 
     def __len__(self) -> int:
         return len(self._steps())
+
+
+@dataclass
+class BatchStatistics:
+    """Summary of one sampler epoch's batch plan (reference
+    batch_sampler.py:25)."""
+
+    num_batches: int
+    num_samples: int
+    skipped_samples: int
+    min_cost: int
+    max_cost: int
+    mean_cost: float
+
+
+def compute_batch_statistics(sampler) -> BatchStatistics:
+    """Statistics of a CostAwareBatchSampler's current plan."""
+    plan = sampler._plan()
+    costs = getattr(sampler, "costs", None) or graph_node_costs(
+        sampler.dataset)
+    batch_costs = [sum(costs[i] for i in b) for b in plan]
+    n_used = sum(len(b) for b in plan)
+    return BatchStatistics(
+        num_batches=len(plan),
+        num_samples=n_used,
+        skipped_samples=len(costs) - n_used,
+        min_cost=min(batch_costs) if batch_costs else 0,
+        max_cost=max(batch_costs) if batch_costs else 0,
+        mean_cost=(sum(batch_costs) / len(batch_costs)
+                   if batch_costs else 0.0))

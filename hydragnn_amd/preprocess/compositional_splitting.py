@@ -63,3 +63,79 @@ def compositional_stratified_split(dataset, perc_train: float,
         test_idx = idx[n_train + n_val:]
     take = lambda ids: [dataset[int(i)] for i in ids]
     return take(train_idx), take(val_idx), take(test_idx)
+
+
+def get_max_graph_size(dataset) -> int:
+    return max(int(d.num_nodes) for d in dataset)
+
+
+def get_elements_list(dataset):
+    """Sorted unique element identities (column 0 of x) across the
+    dataset."""
+    import torch
+    vals = torch.cat([torch.unique(d.x[:, 0]) for d in dataset])
+    return torch.unique(vals)
+
+
+def create_dictionary_from_elements_list(elements_list):
+    return {float(e): i for i, e in enumerate(elements_list)}
+
+
+def create_dataset_categories(dataset):
+    """Category id per sample = its composition signature (reference
+    compositional_data_splitting.py pattern)."""
+    import torch
+    elements = get_elements_list(dataset)
+    lookup = create_dictionary_from_elements_list(elements)
+    categories = []
+    for d in dataset:
+        counts = [0] * len(lookup)
+        vals, n = torch.unique(d.x[:, 0], return_counts=True)
+        for v, c in zip(vals.tolist(), n.tolist()):
+            counts[lookup[float(v)]] = c
+        categories.append(tuple(counts))
+    uniq = {c: i for i, c in enumerate(sorted(set(categories)))}
+    return [uniq[c] for c in categories]
+
+
+def duplicate_unique_data_samples(dataset, categories):
+    """Ensure every category has >= 2 samples so each split can see
+    it (reference behavior: duplicate singletons)."""
+    from collections import Counter
+    counts = Counter(categories)
+    out, out_cat = list(dataset), list(categories)
+    for i, c in enumerate(categories):
+        if counts[c] == 1:
+            out.append(dataset[i])
+            out_cat.append(c)
+    return out, out_cat
+
+
+def generate_partition(categories, perc_train: float, seed: int = 0):
+    """Per-category proportional train/val/test index partition."""
+    import random
+    from collections import defaultdict
+    rng = random.Random(seed)
+    by_cat = defaultdict(list)
+    for i, c in enumerate(categories):
+        by_cat[c].append(i)
+    train, val, test = [], [], []
+    for idxs in by_cat.values():
+        rng.shuffle(idxs)
+        n = len(idxs)
+        n_train = max(int(round(n * perc_train)), 1)
+        n_val = max((n - n_train) // 2, 0)
+        train += idxs[:n_train]
+        val += idxs[n_train:n_train + n_val]
+        test += idxs[n_train + n_val:]
+    return train, val, test
+
+
+def compositional_stratified_splitting(dataset, perc_train: float):
+    """Reference-named entry point: category-stratified split so every
+    composition appears in every split where possible."""
+    ds, cats = duplicate_unique_data_samples(
+        list(dataset), create_dataset_categories(dataset))
+    tr, va, te = generate_partition(cats, perc_train)
+    pick = lambda idxs: [ds[i] for i in idxs]  # noqa: E731
+    return pick(tr), pick(va), pick(te)

@@ -66,3 +66,17 @@ def shift_energies(dataset: Iterable, e_per_element: np.ndarray) -> None:
             d.energy = d.energy - baseline
         if d.get("y") is not None and d.y.numel() == 1:
             d.y = d.y - baseline
+
+
+def solve_least_squares_svd(A, b):
+    """Pseudo-inverse least squares via SVD (reference
+    energy_linear_regression.py:28); numpy in, numpy out."""
+    import numpy as np
+    U, S, Vt = np.linalg.svd(np.asarray(A, dtype=float),
+                             full_matrices=False)
+    keep = S > 1e-12 * (S[0] if S.size else 1.0)
+    S_inv = np.zeros_like(S)
+    S_inv[keep] = 1.0 / S[keep]
+    tmp = U.T @ np.asarray(b, dtype=float)
+    tmp = tmp * S_inv if tmp.ndim == 1 else S_inv[:, None] * tmp
+    return Vt.T @ tmp

@@ -161,3 +161,65 @@ def transform_raw_data_to_serialized(config):
                      minmax_node_feature=ds.minmax_node_feature,
                      minmax_graph_feature=ds.minmax_graph_feature)
     return ds
+
+
+class SimpleDataLoader(TorchDataLoader):
+    """Plain single-process loader over graph batches — the reference's
+    naive custom-loader baseline (load_data.py:70); HydraDataLoader is
+    the prefetching variant."""
+
+    def __init__(self, dataset, batch_size=1, shuffle=False, **kwargs):
+        kwargs.pop("collate_fn", None)
+        super().__init__(dataset, batch_size=batch_size,
+                         shuffle=shuffle, collate_fn=_collate, **kwargs)
+
+
+def total_to_train_val_test_pkls(config, isdist=False):
+    """Split a 'total' serialized pickle into train/validate/test
+    pickles and update config paths (reference load_data.py:475)."""
+    import pickle
+
+    from ..utils.datasets.serializeddataset import (SerializedDataset,
+                                                    SerializedWriter)
+    paths = config["Dataset"]["path"]
+    base = os.environ.get("SERIALIZED_DATA_PATH", os.getcwd())
+    sdir = f"{base}/serialized_dataset"
+    name = config["Dataset"]["name"]
+    if str(list(paths.values())[0]).endswith(".pkl"):
+        file_dir = str(paths["total"])
+        sdir = os.path.dirname(file_dir)
+        name = os.path.splitext(os.path.basename(file_dir))[0]
+    ds = SerializedDataset(sdir, name)
+    perc_train = config["NeuralNetwork"]["Training"]["perc_train"]
+    trainset, valset, testset = split_dataset(
+        list(ds), perc_train,
+        config["Dataset"].get("compositional_stratified_splitting",
+                              False))
+    out = {}
+    for label, subset in (("train", trainset), ("validate", valset),
+                          ("test", testset)):
+        SerializedWriter(subset, sdir, f"{name}_{label}",
+                         minmax_node_feature=getattr(
+                             ds, "minmax_node_feature", None),
+                         minmax_graph_feature=getattr(
+                             ds, "minmax_graph_feature", None))
+        out[label] = os.path.join(sdir, f"{name}_{label}.pkl")
+    config["Dataset"]["path"] = out
+    return config
+
+
+def load_train_val_test_sets(config, isdist=False):
+    """Load the per-split serialized pickles named by the config
+    (reference load_data.py:430)."""
+    from ..utils.datasets.serializeddataset import SerializedDataset
+    base = os.environ.get("SERIALIZED_DATA_PATH", os.getcwd())
+    sets = {}
+    for split, raw_path in config["Dataset"]["path"].items():
+        if str(raw_path).endswith(".pkl"):
+            sdir = os.path.dirname(str(raw_path))
+            label = os.path.splitext(os.path.basename(str(raw_path)))[0]
+        else:
+            sdir = f"{base}/serialized_dataset"
+            label = f"{config['Dataset']['name']}_{split}"
+        sets[split] = SerializedDataset(sdir, label)
+    return (sets["train"], sets["validate"], sets["test"])

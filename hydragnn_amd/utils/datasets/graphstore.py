@@ -242,3 +242,27 @@ class DistDataset(AbstractBaseDataset):
 
     def get(self, idx):
         return self.store[idx]
+
+
+def allgatherv_numpy(array, comm=None):
+    """Variable-length all-gather of 1-D numpy arrays over
+    torch.distributed (reference distdataset.allgatherv pattern);
+    returns the concatenation across ranks."""
+    import numpy as np
+    import torch
+    import torch.distributed as dist
+    if not (dist.is_available() and dist.is_initialized()):
+        return np.asarray(array)
+    t = torch.from_numpy(np.ascontiguousarray(array))
+    n = torch.tensor([t.numel()], dtype=torch.long)
+    sizes = [torch.zeros(1, dtype=torch.long)
+             for _ in range(dist.get_world_size())]
+    dist.all_gather(sizes, n)
+    maxn = int(max(s.item() for s in sizes))
+    pad = torch.zeros(maxn, dtype=t.dtype)
+    pad[:t.numel()] = t
+    outs = [torch.zeros(maxn, dtype=t.dtype)
+            for _ in range(dist.get_world_size())]
+    dist.all_gather(outs, pad)
+    return np.concatenate([o[:int(s.item())].numpy()
+                           for o, s in zip(outs, sizes)])

@@ -285,3 +285,53 @@ def check_remaining_time(start_time, epoch_time, broadcast: bool = True
     if broadcast and dist.is_initialized():
         dist.broadcast(should_stop, src=0)
     return bool(should_stop.item())
+
+
+def find_ifname(myaddr: str):
+    """Network interface name for an IP address (gloo ifname pinning;
+    reference distributed.py:60)."""
+    import socket
+    ipaddr = socket.gethostbyname(myaddr)
+    try:
+        import psutil
+        for nic, addrs in psutil.net_if_addrs().items():
+            for addr in addrs:
+                if addr.address == ipaddr:
+                    return nic
+    except ImportError:  # pragma: no cover
+        pass
+    return None
+
+
+def get_device_list():
+    """Visible GPU ordinals ([] on CPU-only hosts)."""
+    if torch.cuda.is_available():
+        return list(range(torch.cuda.device_count()))
+    return []
+
+
+def get_device_from_name(name: str) -> torch.device:
+    if name.startswith("cuda"):
+        return torch.device(name)
+    return torch.device("cpu")
+
+
+def is_model_distributed(model) -> bool:
+    return isinstance(
+        model, torch.nn.parallel.distributed.DistributedDataParallel)
+
+
+def timedelta_parse(value: str):
+    """Parse '[[DD-]HH:]MM:SS'-style SLURM remaining-time strings into
+    a datetime.timedelta."""
+    import datetime
+    value = value.strip()
+    days = 0
+    if "-" in value:
+        d, value = value.split("-", 1)
+        days = int(d)
+    parts = [int(p) for p in value.split(":")]
+    while len(parts) < 3:
+        parts.insert(0, 0)
+    h, m, s = parts[-3:]
+    return datetime.timedelta(days=days, hours=h, minutes=m, seconds=s)

@@ -87,3 +87,16 @@ def read_node_list() -> List[str]:
             return sorted(set(v.split()))
         return parse_slurm_nodelist(v)
     return ["127.0.0.1"]
+
+
+def read_job_node_list(job_id, job_nodes=None, nodes_per_job: int = 4):
+    """Slice the allocation's node list into a per-trial sub-list and
+    pick its master host (reference deephyper.py:88)."""
+    if job_nodes is None:
+        nodelist = os.environ.get("SLURM_JOB_NODELIST", "")
+        nodes = parse_slurm_nodelist(nodelist) if nodelist else []
+        if not nodes:
+            return "127.0.0.1", ""
+        start = (int(job_id) * nodes_per_job) % len(nodes)
+        job_nodes = nodes[start:start + nodes_per_job]
+    return job_nodes[0], ",".join(job_nodes)

@@ -60,3 +60,26 @@ def compositional_histogram_cutoff(dataset: Sequence, element: int,
             bins[b] = bins.get(b, 0) + 1
             kept.append(d)
     return kept
+
+
+# reference-named aliases (convert_total_energy_to_formation_gibbs.py)
+compute_formation_enthalpy = get_formation_enthalpy
+
+
+def read_file(path: str):
+    """Read one raw LSMS text sample -> (free_energy, atom_rows)
+    (reference convert_total_energy_to_formation_gibbs.read_file)."""
+    with open(path) as f:
+        lines = [ln.split() for ln in f.read().splitlines() if ln.strip()]
+    free_energy = float(lines[0][0])
+    atoms = [[float(v) for v in row] for row in lines[1:]]
+    return free_energy, atoms
+
+
+def find_bin(value: float, edges) -> int:
+    """Histogram bin index for the compositional cutoff (reference
+    compositional_histogram_cutoff.find_bin)."""
+    for i in range(len(edges) - 1):
+        if edges[i] <= value < edges[i + 1]:
+            return i
+    return len(edges) - 2

@@ -71,3 +71,7 @@ def validate_atomistic_sample(data, require_forces: bool = False,
         raise ValueError("'cell' must be 3x3")
     if not torch.isfinite(pos).all():
         raise ValueError("'pos' contains non-finite values")
+
+
+# reference-named alias
+validate_materials_sample = validate_atomistic_sample

@@ -176,3 +176,74 @@ class Checkpoint:
             self.best = val_loss
             return True
         return False
+
+
+def print_model(model):
+    """Layer-by-layer parameter sizes (reference model.py:453)."""
+    from ..print.print_utils import print_master
+    num_params = 0
+    num_bytes = 0
+    for k, v in _unwrap(model).state_dict().items():
+        print_master("%50s\t%20s\t%10d" % (k, list(v.shape), v.numel()))
+        num_params += v.numel()
+        num_bytes += v.numel() * v.element_size()
+    print_master("-" * 50)
+    print_master("%50s\t%20s\t%10d" % ("Total", "", num_params))
+    print_master("All (total, MB): %d %g"
+                 % (num_params, num_bytes / 1024 / 1024))
+    return num_params
+
+
+def print_optimizer(optimizer):
+    """Optimizer state summary (reference model.py pattern)."""
+    from ..print.print_utils import print_master
+    for i, group in enumerate(optimizer.param_groups):
+        n = sum(p.numel() for p in group["params"])
+        print_master(f"param_group {i}: {len(group['params'])} tensors,"
+                     f" {n} params, lr={group.get('lr')}")
+
+
+def tensor_divide(x1, x2):
+    """Elementwise division with 0 where the denominator is 0."""
+    out = torch.zeros_like(x1)
+    mask = x2 != 0
+    out[mask] = x1[mask] / x2[mask]
+    return out
+
+
+def unsorted_segment_mean(data, segment_ids, num_segments):
+    """Segment mean by scatter (reference model.py:443); the HIP path
+    is ops.scatter(..., 'mean')."""
+    from ...ops import scatter
+    return scatter(data, segment_ids, num_segments, "mean")
+
+
+def calculate_PNA_degree(dataset, max_neighbours=None):
+    """In-degree histogram for PNA scalers (dist-reduced when a
+    process group is initialized)."""
+    from ..config.config_utils import _gather_deg
+    return _gather_deg(dataset)
+
+
+def calculate_avg_deg(dataset):
+    from ..config.config_utils import _calculate_avg_deg
+    return _calculate_avg_deg(dataset)
+
+
+calculate_PNA_degree_dist = calculate_PNA_degree
+calculate_PNA_degree_mpi = calculate_PNA_degree
+calculate_avg_deg_dist = calculate_avg_deg
+calculate_avg_deg_mpi = calculate_avg_deg
+
+
+def multitask_optim_state_dict(dual_optimizer):
+    """Consolidated state dicts for the branch model-parallel
+    DualOptimizer (reference model.py multitask pattern)."""
+    out = {}
+    for name in ("optimizer1", "optimizer2"):
+        opt = getattr(dual_optimizer, name, None)
+        if opt is not None and hasattr(opt, "state_dict"):
+            out[name] = opt.state_dict()
+    if not out and hasattr(dual_optimizer, "state_dict"):
+        out = {"optimizer": dual_optimizer.state_dict()}
+    return out

@@ -35,3 +35,18 @@ def select_optimizer(model, config):
         return ZeroRedundancyOptimizer(
             model.parameters(), optimizer_class=cls, **kwargs)
     return cls(model.parameters(), **kwargs)
+
+
+def select_standard_optimizer(model, config):
+    """Non-sharded optimizer selection (reference optimizer.py
+    split)."""
+    cfg = dict(config)
+    cfg["use_zero_redundancy"] = False
+    return select_optimizer(model, cfg)
+
+
+def select_zero_redundancy_optimizer(model, config):
+    """ZeRO-1 sharded optimizer selection."""
+    cfg = dict(config)
+    cfg["use_zero_redundancy"] = True
+    return select_optimizer(model, cfg)

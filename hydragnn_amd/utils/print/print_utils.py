@@ -67,3 +67,14 @@ def log(*args):
 def log0(*args):
     if _rank() == 0:
         log(*args)
+
+
+def print_nothing(*args, **kwargs):
+    """Verbosity sink (reference print_utils)."""
+
+
+def print_all_processes(*args, **kwargs):
+    """Print from every rank, rank-prefixed."""
+    import torch.distributed as dist
+    rank = dist.get_rank() if dist.is_initialized() else 0
+    print(f"[{rank}]", *args, **kwargs)

@@ -13,6 +13,7 @@ from __future__ import annotations
 import os
 import time
 from collections import defaultdict
+from contextlib import contextmanager
 from functools import wraps
 from typing import Dict, List
 
@@ -176,3 +177,14 @@ def reset():
     global _backends, _enabled
     _backends = []
     _enabled = False
+
+
+@contextmanager
+def timer(name: str):
+    """Context manager sugar over start/stop (reference
+    tracer.py:485)."""
+    start(name)
+    try:
+        yield
+    finally:
+        stop(name)
