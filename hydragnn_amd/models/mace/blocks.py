@@ -266,3 +266,123 @@ class NonLinearReadoutBlock(nn.Module):
 
     def forward(self, node_feats: torch.Tensor) -> torch.Tensor:
         return self.linear_2(self.act(self.linear_1(node_feats[:, :, 0])))
+
+
+# ---------------------------------------------------------------------------
+# Reference-named radial/embedding module wrappers
+# (reference utils/model/mace_utils/modules/radial.py and blocks.py) —
+# thin Modules over this framework's differentiable basis ops so a
+# migrating user finds the classes they imported.
+# ---------------------------------------------------------------------------
+class PolynomialCutoff(nn.Module):
+    def __init__(self, r_max: float, p: int = 6):
+        super().__init__()
+        self.r_max = float(r_max)
+        self.p = float(p)
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        return polynomial_cutoff(x, self.r_max, self.p)
+
+
+class GaussianBasis(nn.Module):
+    def __init__(self, r_max: float, num_basis: int = 128,
+                 trainable: bool = False):
+        super().__init__()
+        offsets = torch.linspace(0.0, r_max, num_basis)
+        if trainable:
+            self.offsets = nn.Parameter(offsets)
+        else:
+            self.register_buffer("offsets", offsets)
+        self.coeff = -0.5 / float(offsets[1] - offsets[0]) ** 2
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        return gaussian_basis(x, self.offsets.view(1, -1).to(x.dtype),
+                              self.coeff)
+
+
+class ChebychevBasis(nn.Module):
+    def __init__(self, r_max: float, num_basis: int = 8):
+        super().__init__()
+        self.r_max = float(r_max)
+        self.num_basis = num_basis
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        return chebyshev_basis(x, self.r_max, self.num_basis)
+
+
+class AgnesiTransform(nn.Module):
+    """Agnesi distance transform (ACEpotentials.jl / JCP 2023)."""
+
+    def __init__(self, q: float = 0.9183, p: float = 4.5791,
+                 a: float = 1.0805, trainable: bool = False):
+        super().__init__()
+        t = torch.tensor
+        if trainable:
+            self.q, self.p, self.a = (nn.Parameter(t(q)),
+                                      nn.Parameter(t(p)),
+                                      nn.Parameter(t(a)))
+        else:
+            self.register_buffer("q", t(q))
+            self.register_buffer("p", t(p))
+            self.register_buffer("a", t(a))
+
+    def forward(self, x, node_attrs=None, edge_index=None,
+                atomic_numbers=None):
+        r0 = 1.0  # covalent-radius scale (no ase tables in this image)
+        xs = x / r0
+        return 1.0 / (1.0 + self.a * xs.pow(self.q)
+                      / (1.0 + xs.pow(self.q - self.p)))
+
+
+class SoftTransform(nn.Module):
+    def __init__(self, alpha: float = 4.0, trainable: bool = False):
+        super().__init__()
+        t = torch.tensor(alpha)
+        if trainable:
+            self.alpha = nn.Parameter(t)
+        else:
+            self.register_buffer("alpha", t)
+
+    def forward(self, x, node_attrs=None, edge_index=None,
+                atomic_numbers=None):
+        return x * torch.sigmoid(self.alpha * (x - 1.0)) + (
+            1.0 - torch.sigmoid(self.alpha * (x - 1.0)))
+
+
+class LinearNodeEmbeddingBlock(nn.Module):
+    """One-hot element -> channel embedding (reference blocks.py)."""
+
+    def __init__(self, num_elements: int, num_channels: int):
+        super().__init__()
+        self.linear = nn.Linear(num_elements, num_channels, bias=False)
+
+    def forward(self, node_attrs: torch.Tensor) -> torch.Tensor:
+        return self.linear(node_attrs)
+
+
+class AtomicEnergiesBlock(nn.Module):
+    """Per-element reference energies: E0 contribution of each atom."""
+
+    def __init__(self, atomic_energies):
+        super().__init__()
+        self.register_buffer(
+            "atomic_energies",
+            torch.as_tensor(atomic_energies, dtype=torch.get_default_dtype()))
+
+    def forward(self, one_hot: torch.Tensor) -> torch.Tensor:
+        return one_hot @ self.atomic_energies.view(-1, 1)
+
+
+class ScaleShiftBlock(nn.Module):
+    def __init__(self, scale: float, shift: float):
+        super().__init__()
+        self.register_buffer("scale", torch.tensor(float(scale)))
+        self.register_buffer("shift", torch.tensor(float(shift)))
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        return self.scale * x + self.shift
+
+
+# reference-named readout aliases
+LinearMLPNode = LinearReadoutBlock
+NonLinearMLPNode = NonLinearReadoutBlock

@@ -159,3 +159,65 @@ def tp_paths(l_in_max: int, l_edge_max: int, l_out_max: int
             for l3 in allowed_l3(l1, l2, l_out_max):
                 paths.append((l1, l2, l3))
     return paths
+
+
+# ---------------------------------------------------------------------------
+# Reference-named irreps tools (reference utils/model/irreps_tools.py
+# and mace_utils/tools/cg.py) for this framework's dense
+# uniform-multiplicity layout [N, C, (lmax+1)^2].
+# ---------------------------------------------------------------------------
+def create_irreps_string(num_channels: int, lmax: int) -> str:
+    """e.g. 64x0e+64x1o+64x2e (alternating parity convention)."""
+    return "+".join(f"{num_channels}x{l}{'e' if l % 2 == 0 else 'o'}"
+                    for l in range(lmax + 1))
+
+
+def tp_out_irreps_with_instructions(l_in_max: int, l_edge_max: int,
+                                    l_out_max: int):
+    """uvu path table plus per-path instruction tuples
+    (l1_block, l2_block, path_index, 'uvu', True) — the reference's
+    instruction shape over this framework's path list."""
+    paths = tp_paths(l_in_max, l_edge_max, l_out_max)
+    instructions = [(l1, l2, i, "uvu", True)
+                    for i, (l1, l2, l3) in enumerate(paths)]
+    return paths, instructions
+
+
+def linear_out_irreps(lmax: int, lmax_target: int) -> int:
+    """Output tower lmax after a linear map restricted to the
+    target's l-content."""
+    return min(lmax, lmax_target)
+
+
+def reshape_irreps(x: torch.Tensor, num_channels: int,
+                   lmax: int) -> torch.Tensor:
+    """Flat e3nn-style [N, sum_l C*(2l+1)] -> dense [N, C, (lmax+1)^2]
+    (the layout every block here consumes)."""
+    n = x.shape[0]
+    out = x.new_zeros(n, num_channels, (lmax + 1) ** 2)
+    off = 0
+    for l in range(lmax + 1):
+        w = 2 * l + 1
+        blk = x[:, off:off + num_channels * w].reshape(n, num_channels,
+                                                       w)
+        out[:, :, l * l:(l + 1) ** 2] = blk
+        off += num_channels * w
+    return out
+
+
+def extract_invariant(x: torch.Tensor, num_layers: int = 1) -> torch.Tensor:
+    """l=0 channel of a dense tower [N, C, D] -> [N, C] (reference
+    extract_invariant on concatenated layer outputs)."""
+    return x[:, :, 0]
+
+
+def U_matrix_real(irreps_in, irreps_out, correlation: int,
+                  **unused):
+    """Reference cg.py entry-point name; the generalized
+    Clebsch-Gordan contraction matrices for the symmetric
+    contraction.  Accepts lmax ints for in/out on this framework's
+    uniform-multiplicity towers."""
+    from .symmetric_contraction import u_matrix
+    lmax_in = int(irreps_in)
+    lmax_out = int(irreps_out)
+    return u_matrix(lmax_in, lmax_out, correlation)

@@ -205,3 +205,32 @@ class MACEStack(Base):
 
     def __str__(self):
         return "MACEStack"
+
+
+def process_node_attributes(node_attributes, num_elements: int):
+    """Validate-and-one-hot raw atomic numbers (reference
+    MACEStack.process_node_attributes): squeeze to 1-D, warn on
+    non-integer or out-of-range values, clamp, one-hot encode."""
+    import warnings
+    z = node_attributes.squeeze()
+    assert z.dim() == 1, (
+        "MACE only supports raw atomic numbers as node_attributes")
+    if not torch.all(z == z.round()):
+        warnings.warn("MACE node_attributes contain non-integer values;"
+                      " expected atomic numbers")
+    zi = z.round().long()
+    if not torch.all((zi >= 1) & (zi <= num_elements)):
+        warnings.warn("atomic numbers outside [1, num_elements];"
+                      " clamping")
+        zi = zi.clamp(1, num_elements)
+    return torch.nn.functional.one_hot(zi - 1, num_elements).float()
+
+
+def get_multihead_decoder(num_channels: int, hidden: int, out_dim: int,
+                          nonlinear: bool = True):
+    """Reference-named readout factory over this framework's readout
+    blocks."""
+    from .blocks import LinearReadoutBlock, NonLinearReadoutBlock
+    if nonlinear:
+        return NonLinearReadoutBlock(num_channels, hidden, out_dim)
+    return LinearReadoutBlock(num_channels, out_dim)

@@ -169,3 +169,9 @@ class PAINNStack(Base):
 
     def __str__(self):
         return "PAINNStack"
+
+
+def sinc_expansion(edge_dist, edge_size: int, cutoff: float):
+    """Reference PAINNStack helper name: sinc radial basis
+    sin(n pi d / rc) / d for n = 1..edge_size."""
+    return sinc_basis(edge_dist.view(-1, 1), cutoff, edge_size)

@@ -124,3 +124,21 @@ class PNAEqStack(Base):
 
     def __str__(self):
         return "PNAEqStack"
+
+
+class rbf_BasisLayer(torch.nn.Module):
+    """Reference PNAEqStack helper name: Bessel RBF x polynomial
+    cutoff module."""
+
+    def __init__(self, num_rbf: int, cutoff: float):
+        super().__init__()
+        import math
+        self.cutoff = float(cutoff)
+        self.weights = torch.nn.Parameter(
+            torch.arange(1, num_rbf + 1).float() * math.pi / cutoff)
+
+    def forward(self, dist):
+        from ..ops import bessel_basis, polynomial_cutoff
+        r = dist.view(-1, 1)
+        return bessel_basis(r, self.cutoff, self.weights) * \
+            polynomial_cutoff(r, self.cutoff)

@@ -26,3 +26,14 @@ def unscale_features_by_num_nodes(data):
         if v is not None:
             data[key] = v * n
     return data
+
+
+def unscale_features_by_num_nodes_config(data, config):
+    """Config-gated per-node unscaling (reference postprocess.py
+    variant): only denormalizes when the config asked for
+    total-energy-per-atom style scaling."""
+    vo = config["NeuralNetwork"]["Variables_of_interest"]
+    if vo.get("denormalize_output", False) or vo.get(
+            "scale_features_by_num_nodes", False):
+        return unscale_features_by_num_nodes(data)
+    return data

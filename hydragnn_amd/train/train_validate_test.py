@@ -425,3 +425,18 @@ def train_validate_test(
                      **{f"pred_{i}": p_.cpu().numpy()
                         for i, p_ in enumerate(pv)})
     tr.save(f"logs/{log_name}")
+
+
+# reference-named split entry points (train_validate_test.py:523-617)
+def get_head_indices_graph(model, data):
+    """Head index slices when every head is graph-level."""
+    return get_head_indices(model, data)
+
+
+def get_head_indices_node_or_mixed(model, data):
+    """Head index slices with node-level or mixed heads."""
+    return get_head_indices(model, data)
+
+
+reduce_values_ranks_dist = reduce_values_ranks
+reduce_values_ranks_mpi = reduce_values_ranks

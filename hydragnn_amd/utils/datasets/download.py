@@ -87,3 +87,7 @@ def main(argv=None):
 
 if __name__ == "__main__":
     raise SystemExit(main())
+
+
+# reference-named alias
+download_file = download

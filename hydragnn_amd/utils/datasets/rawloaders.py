@@ -184,3 +184,10 @@ class CFGDataset(AbstractRawDataset):
             data.supercell_size = torch.tensor(cell_rows,
                                                dtype=torch.float)
         return data
+
+
+# reference-named aliases (preprocess/*_raw_dataset_loader.py)
+LSMS_RawDataLoader = LSMSDataset
+CFG_RawDataLoader = CFGDataset
+XYZ_RawDataLoader = XYZDataset
+AbstractRawDataLoader = AbstractRawDataset

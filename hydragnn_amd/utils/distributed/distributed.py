@@ -335,3 +335,15 @@ def timedelta_parse(value: str):
         parts.insert(0, 0)
     h, m, s = parts[-3:]
     return datetime.timedelta(days=days, hours=h, minutes=m, seconds=s)
+
+
+def get_deepspeed_init_args():
+    """Rendezvous kwargs a deepspeed.initialize-style entry point
+    would need (reference distributed.py); RCCL fills the NCCL role
+    on MI355X."""
+    return {
+        "rank": int(os.environ.get("RANK", 0)),
+        "world_size": int(os.environ.get("WORLD_SIZE", 1)),
+        "distributed_port": int(os.environ.get("MASTER_PORT", 29500)),
+        "dist_backend": "nccl" if torch.cuda.is_available() else "gloo",
+    }

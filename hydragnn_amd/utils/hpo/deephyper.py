@@ -100,3 +100,18 @@ def read_job_node_list(job_id, job_nodes=None, nodes_per_job: int = 4):
         start = (int(job_id) * nodes_per_job) % len(nodes)
         job_nodes = nodes[start:start + nodes_per_job]
     return job_nodes[0], ",".join(job_nodes)
+
+
+def create_ds_config(params, job_id, DEEPHYPER_LOG_DIR="."):
+    """Write a per-trial DeepSpeed-style JSON config (reference
+    deephyper.py create_ds_config); returns the path."""
+    import json
+    path = os.path.join(DEEPHYPER_LOG_DIR, f"ds_config_{job_id}.json")
+    cfg = {
+        "train_batch_size": params.get("batch_size", 32),
+        "optimizer": {"type": params.get("optimizer", "AdamW"),
+                      "params": {"lr": params.get("lr", 1e-3)}},
+    }
+    with open(path, "w") as f:
+        json.dump(cfg, f, indent=2)
+    return path

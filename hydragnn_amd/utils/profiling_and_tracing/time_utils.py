@@ -50,3 +50,7 @@ def print_timers(verbosity=0):
         if rank == 0:
             print(f"Timer {name}: min {tmin:.4f}s max {tmax:.4f}s "
                   f"avg {tavg:.4f}s")
+
+
+class TimerError(Exception):
+    """Raised on Timer misuse (start twice / stop before start)."""

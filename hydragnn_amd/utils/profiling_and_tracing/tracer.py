@@ -188,3 +188,24 @@ def timer(name: str):
         yield
     finally:
         stop(name)
+
+
+# gptl4py-compat shims (reference gptl4py_dummy.py): the region tracer
+# here fills the role of GPTL; these names keep drop-in scripts alive.
+# (initialize() above already covers gptl's initialize.)
+def finalize():
+    disable()
+
+
+def pr_file(path: str):
+    save(os.path.dirname(path) or ".")
+
+
+def pr_summary_file(path: str):
+    save(os.path.dirname(path) or ".")
+
+
+@contextmanager
+def nvtx_timer(name: str):
+    with timer(name):
+        yield
