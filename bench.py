@@ -283,6 +283,7 @@ def main():
                 "num_conv_layers": MODEL_CONFIG["num_conv_layers"],
                 "max_ell": MODEL_CONFIG["max_ell"],
                 "correlation": MODEL_CONFIG["correlation"],
+                "interaction": "att",
             },
         }))
     if dist.is_initialized():

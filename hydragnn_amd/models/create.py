@@ -74,6 +74,7 @@ def create_model(
     max_ell=None,
     node_max_ell=None,
     avg_num_neighbors=None,
+    interaction_type=None,
     conv_checkpointing: bool = False,
     enable_interatomic_potential: bool = False,
     energy_weight: float = 0.0,
@@ -175,6 +176,7 @@ def create_model(
             num_bessel=num_radial, edge_dim=edge_dim, max_ell=max_ell,
             node_max_ell=node_max_ell, avg_num_neighbors=avg_num_neighbors,
             envelope_exponent=envelope_exponent, correlation=correlation,
+            interaction_type=interaction_type or "att",
             equivariance=True, **common)
     else:
         raise ValueError(f"Unknown mpnn_type: {mpnn_type}")
@@ -236,6 +238,7 @@ def create_model_config(config: dict, verbosity: int = 0,
         max_ell=arch.get("max_ell"),
         node_max_ell=arch.get("node_max_ell"),
         avg_num_neighbors=arch.get("avg_num_neighbors"),
+        interaction_type=arch.get("interaction_type"),
         conv_checkpointing=training.get("conv_checkpointing", False),
         enable_interatomic_potential=arch.get(
             "enable_interatomic_potential", False),

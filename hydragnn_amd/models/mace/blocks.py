@@ -166,6 +166,10 @@ class RealAgnosticResidualInteractionBlock(nn.Module):
     """linear_up -> gather -> edge TP (radial-MLP weights) ->
     scatter-sum / avg_num_neighbors -> linear -> (+ per-l skip)."""
 
+    # extra radial-MLP input features beyond the radial embedding
+    # (the Att subclass appends 2C down-projected endpoint scalars)
+    _radial_extra = 0
+
     def __init__(self, num_channels: int, lmax_node: int, lmax_edge: int,
                  lmax_out: int, radial_dim: int,
                  avg_num_neighbors: float,
@@ -175,9 +179,10 @@ class RealAgnosticResidualInteractionBlock(nn.Module):
         self.linear_up = IrrepsLinear(num_channels, num_channels, lmax_node)
         self.conv_tp = EdgeTensorProduct(lmax_node, lmax_edge, lmax_out)
         from ...ops.mfma_linear import MFMALinear as SplitKLinear
+        self._setup_extra(num_channels)
         hidden = radial_mlp or [64, 64, 64]
         mods = []
-        prev = radial_dim
+        prev = radial_dim + self._radial_extra
         for h in hidden:
             mods += [SplitKLinear(prev, h), nn.SiLU()]
             prev = h
@@ -191,6 +196,12 @@ class RealAgnosticResidualInteractionBlock(nn.Module):
         self.lmax_out = lmax_out
         self.lmax_node = lmax_node
 
+    def _setup_extra(self, num_channels: int):
+        pass
+
+    def _edge_weights(self, node_feats, src, dst, edge_radial, etp_meta):
+        return self.radial_mlp(edge_radial)
+
     def forward(self, node_feats: torch.Tensor, edge_index: torch.Tensor,
                 edge_sh: torch.Tensor, edge_radial: torch.Tensor,
                 edges_sorted: bool = False,
@@ -199,7 +210,8 @@ class RealAgnosticResidualInteractionBlock(nn.Module):
         src, dst = edge_index[0], edge_index[1]
         n, c, _ = node_feats.shape
         x = self.linear_up(node_feats)
-        w = self.radial_mlp(edge_radial).view(
+        w = self._edge_weights(node_feats, src, dst, edge_radial,
+                               etp_meta).view(
             -1, c, self.conv_tp.num_paths)
         import os
         if etp_meta is not None and _kernel_ok(
@@ -227,6 +239,33 @@ class RealAgnosticResidualInteractionBlock(nn.Module):
         if sc.shape[-1] < want:
             sc = torch.nn.functional.pad(sc, (0, want - sc.shape[-1]))
         return self.linear(m, add=sc)
+
+
+class RealAgnosticAttResidualInteractionBlock(
+        RealAgnosticResidualInteractionBlock):
+    """The reference MACEStack's default interaction (reference
+    blocks.py:311): the per-edge tensor-product weights attend to BOTH
+    endpoints — the radial-MLP input is [radial embedding,
+    down-projected sender scalars, down-projected receiver scalars].
+    Everything downstream (TP, aggregation, skip) matches the residual
+    block."""
+
+    def _setup_extra(self, num_channels: int):
+        self._radial_extra = 2 * num_channels
+        self.linear_down = nn.Linear(num_channels, num_channels,
+                                     bias=False)
+
+    def _edge_weights(self, node_feats, src, dst, edge_radial, etp_meta):
+        down = self.linear_down(node_feats[:, :, 0])  # [N, C] scalars
+        src_csr = getattr(etp_meta, "src_csr", None) \
+            if etp_meta is not None else None
+        dst_csr = getattr(etp_meta, "dst_csr", None) \
+            if etp_meta is not None else None
+        augmented = torch.cat(
+            [edge_radial,
+             gather(down, src, backward_csr=src_csr),
+             gather(down, dst, backward_csr=dst_csr)], dim=-1)
+        return self.radial_mlp(augmented)
 
 
 class EquivariantProductBasisBlock(nn.Module):

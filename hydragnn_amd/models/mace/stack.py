@@ -26,6 +26,7 @@ from .blocks import (
     LinearReadoutBlock,
     NonLinearReadoutBlock,
     RadialEmbeddingBlock,
+    RealAgnosticAttResidualInteractionBlock,
     RealAgnosticResidualInteractionBlock,
 )
 from .o3 import dim
@@ -46,6 +47,7 @@ class MACEStack(Base):
         envelope_exponent: Optional[int] = 5,
         correlation: Optional[int] = 2,
         edge_dim: Optional[int] = None,
+        interaction_type: str = "att",
         **kwargs,
     ):
         self.r_max = r_max or 5.0
@@ -58,6 +60,10 @@ class MACEStack(Base):
         self.envelope_exponent = envelope_exponent or 5
         self.correlation = correlation if isinstance(correlation, int) \
             else (correlation[0] if correlation else 2)
+        # "att" = reference MACEStack default (blocks.py:121: radial
+        # weights attend to endpoint scalars); "residual" = lighter
+        # distance-only radial weights
+        self.interaction_type = interaction_type or "att"
         self.is_edge_model = True
         super().__init__(edge_dim=edge_dim, **kwargs)
 
@@ -77,7 +83,10 @@ class MACEStack(Base):
             # interaction always emits the full hidden tower; the
             # product basis contracts to the layer's target (scalars
             # only on the last layer — MACE convention)
-            self.interactions.append(RealAgnosticResidualInteractionBlock(
+            inter_cls = (RealAgnosticAttResidualInteractionBlock
+                         if self.interaction_type == "att"
+                         else RealAgnosticResidualInteractionBlock)
+            self.interactions.append(inter_cls(
                 C, lmax_node, self.max_ell, self.node_max_ell,
                 self.radial_embedding.out_dim, self.avg_num_neighbors))
             self.products.append(EquivariantProductBasisBlock(
@@ -158,6 +167,7 @@ class MACEStack(Base):
         # atomics
         perm_s = torch.argsort(src, stable=True)
         meta.src_csr = (perm_s, _rowptr_from_sorted(src[perm_s], n))
+        meta.dst_csr = (eid_d, rowptr_dst)
         data[key] = meta
         return meta
 

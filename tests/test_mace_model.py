@@ -104,3 +104,28 @@ def test_mace_force_training_decreases():
         err, _ = train_fn(loaders[0], model, opt, 0)
         errs.append(float(err))
     assert errs[-1] < errs[0], f"MACE force loss not decreasing: {errs}"
+
+
+def test_interaction_type_switch():
+    """Default interaction matches the reference (att: radial weights
+    attend to endpoint scalars); 'residual' stays selectable."""
+    from hydragnn_amd.models.mace.blocks import (
+        RealAgnosticAttResidualInteractionBlock,
+        RealAgnosticResidualInteractionBlock)
+    from hydragnn_amd.utils.datasets.synthetic import md17_shape_dataset
+    ds = md17_shape_dataset(num_samples=4)
+    m_att, _, _ = _build(_mace_config(), ds)
+    core = m_att.model if hasattr(m_att, "model") else m_att
+    assert isinstance(core.interactions[0],
+                      RealAgnosticAttResidualInteractionBlock)
+    cfg = _mace_config()
+    cfg["NeuralNetwork"]["Architecture"]["interaction_type"] = "residual"
+    m_res, _, _ = _build(cfg, ds)
+    core = m_res.model if hasattr(m_res, "model") else m_res
+    assert type(core.interactions[0]) is \
+        RealAgnosticResidualInteractionBlock
+    # att radial MLP consumes the augmented input
+    core_att = m_att.model if hasattr(m_att, "model") else m_att
+    att0 = core_att.interactions[0]
+    expected = core_att.radial_embedding.out_dim + 2 * core_att.hidden_dim
+    assert att0.radial_mlp[0].in_features == expected
