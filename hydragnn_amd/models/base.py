@@ -40,7 +40,7 @@ class BatchNormNode(nn.BatchNorm1d):
 
 
 def global_mean_pool(x, batch, size=None):
-    return scatter(x, batch, size, "mean")
+    return scatter(x, batch, size, "mean", sorted_index=True)
 
 
 def global_add_pool(x, batch, size=None):

@@ -50,7 +50,10 @@ class E_GCL(nn.Module):
         m = self.edge_mlp(torch.cat(parts, dim=-1))
         if self.equivariant:
             trans = vec.to(h.dtype) * self.coord_mlp(m)
-            pos = pos + scatter(trans, dst, h.shape[0], "mean").to(pos.dtype)
+            pos = pos + scatter(
+                trans, dst, h.shape[0], "mean",
+                sorted_index=getattr(self, "_edges_sorted", False)
+            ).to(pos.dtype)
         agg = scatter(m, dst, h.shape[0], "sum",
                       sorted_index=getattr(self, "_edges_sorted", False))
         h = self.node_mlp(torch.cat([h, agg], dim=-1))

@@ -67,7 +67,8 @@ class SAGEConv(nn.Module):
 
     def forward(self, x, edge_index):
         src, dst = edge_index[0], edge_index[1]
-        agg = scatter(gather(x, src), dst, x.shape[0], "mean")
+        agg = scatter(gather(x, src), dst, x.shape[0], "mean",
+                      sorted_index=getattr(self, "_edges_sorted", False))
         return self.lin_r(x) + self.lin_l(agg)
 
 

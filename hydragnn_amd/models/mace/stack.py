@@ -133,7 +133,7 @@ class MACEStack(Base):
         n_graphs = data.get("num_graphs_")
         n_graphs = int(n_graphs) if n_graphs is not None else \
             int(batch.max()) + 1
-        mean_pos = scatter(pos, batch, n_graphs, "mean")
+        mean_pos = scatter(pos, batch, n_graphs, "mean", sorted_index=True)
         pos_c = pos - mean_pos[batch]
         vec, lengths = get_edge_vectors_and_lengths(
             pos_c, data.edge_index, data.get("edge_shifts"))

@@ -78,7 +78,10 @@ class CFConv(nn.Module):
         msg = gather(self.lin1(x), src) * W
         if self.equivariant_coords:
             trans = vec.to(x.dtype) * self.coord_nn(msg)
-            pos = pos + scatter(trans, dst, x.shape[0], "mean").to(pos.dtype)
+            pos = pos + scatter(
+                trans, dst, x.shape[0], "mean",
+                sorted_index=getattr(self, "_edges_sorted", False)
+            ).to(pos.dtype)
         out = scatter(msg, dst, x.shape[0], "sum",
                       sorted_index=getattr(self, "_edges_sorted", False))
         return self.lin2(out), pos

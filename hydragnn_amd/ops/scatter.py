@@ -116,9 +116,20 @@ class _ScatterSum(torch.autograd.Function):
 
 class _ScatterMean(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, src, index, dim_size):
+    def forward(ctx, src, index, dim_size, sorted_index=False):
         if _use_hip(src):
+            import os
             ext = get_extension(required=True)
+            if sorted_index and \
+                    os.environ.get("HYDRAGNN_CSR_SCATTER", "1") == "1":
+                # contention-free: CSR sum + rowptr-derived counts
+                rowptr = _rowptr_from_sorted(index, dim_size)
+                count = (rowptr[1:] - rowptr[:-1]).to(src.dtype)
+                s = ext.segment_sum_csr(src.contiguous(), rowptr)
+                out = s / count.clamp(min=1).view(
+                    -1, *([1] * (src.dim() - 1)))
+                ctx.save_for_backward(index, count)
+                return out
             out, count = ext.scatter_mean_fwd(src.contiguous(), index, dim_size)
         else:
             out = src.new_zeros((dim_size,) + src.shape[1:])
@@ -133,7 +144,7 @@ class _ScatterMean(torch.autograd.Function):
         index, count = ctx.saved_tensors
         inv = 1.0 / count.clamp(min=1)
         g = grad_out * inv.view(-1, *([1] * (grad_out.dim() - 1))).to(grad_out.dtype)
-        return gather(g, index), None, None
+        return gather(g, index), None, None, None
 
 
 class _ScatterMax(torch.autograd.Function):
@@ -223,7 +234,7 @@ def scatter(
     if reduce in ("sum", "add"):
         return _ScatterSum.apply(src, index, dim_size, sorted_index, csr)
     if reduce == "mean":
-        return _ScatterMean.apply(src, index, dim_size)
+        return _ScatterMean.apply(src, index, dim_size, sorted_index)
     if reduce in ("max", "amax"):
         return _ScatterMax.apply(src, index, dim_size, True)
     if reduce in ("min", "amin"):
