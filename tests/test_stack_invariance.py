@@ -51,3 +51,41 @@ def test_scalar_output_rotation_invariant(mpnn_type):
     # geometric features change under rotation
     assert torch.allclose(o1, o2, atol=1e-8), (
         f"{mpnn_type}: {(o1 - o2).abs().max():.2e}")
+
+
+@pytest.mark.parametrize("mpnn_type", ["MACE", "EGNN", "SchNet"])
+def test_forces_rotate_equivariantly(mpnn_type):
+    """F(R x) = R F(x): predicted forces co-rotate with the molecule
+    (reference test_forces_equivariant pattern)."""
+    torch.manual_seed(0)
+    from test_mace_model import _build, _mace_config
+    from hydragnn_amd.data import Batch
+    from hydragnn_amd.utils.datasets.synthetic import md17_shape_dataset
+    ds = md17_shape_dataset(num_samples=2)
+    cfg = _mace_config()
+    cfg["NeuralNetwork"]["Architecture"]["mpnn_type"] = mpnn_type
+    model, config, _ = _build(cfg, ds)
+    model.eval()
+
+    def forces(data_list):
+        batch = Batch.from_data_list([d.clone() for d in data_list])
+        batch.pos.requires_grad_(True)
+        pred = model(batch)
+        e = pred[0].sum()
+        return torch.autograd.grad(e, batch.pos)[0].neg()
+
+    # a random rotation
+    q, _ = torch.linalg.qr(torch.randn(3, 3, dtype=torch.float64))
+    if torch.det(q) < 0:
+        q[:, 0] = -q[:, 0]
+    R = q.float()
+    f0 = forces(ds)
+    rot = []
+    for d in ds:
+        dr = d.clone()
+        dr.pos = d.pos @ R.t()
+        rot.append(dr)
+    f1 = forces(rot)
+    assert f0.abs().max() > 1e-6, "forces are trivially zero"
+    err = (f1 - f0 @ R.t()).abs().max() / (f0.abs().max() + 1e-9)
+    assert err < 5e-3, f"{mpnn_type}: force equivariance err {err:.2e}"
