@@ -270,7 +270,7 @@ def main():
             "warmup": args.warmup,
             "ms_per_step": ms_per_step,
             "higher_is_better": True,
-            "scaling": "weak",
+            "scaling": os.environ.get("HYDRAGNN_BENCH_SCALING", "weak"),
             "vs_baseline": None,
             "dtype": PRECISION,
             "data": "synthetic",
