@@ -192,3 +192,55 @@ def test_bench_two_rank_gloo():
     d = json.loads(line)
     assert d["n_gpus"] == 2
     assert d["config"]["global_batch"] == 8
+
+
+def _mace_force_ddp_worker(rank, world_size, port, q):
+    try:
+        _init(rank, world_size, port)
+        import sys
+        sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+        torch.manual_seed(11)
+        from hydragnn_amd.data import Batch
+        from hydragnn_amd.utils.datasets.synthetic import md17_shape_dataset
+        from test_mace_model import _build, _mace_config
+        ds = md17_shape_dataset(num_samples=8)
+        model, config, _ = _build(_mace_config(), ds)
+        ddp = torch.nn.parallel.DistributedDataParallel(model)
+        opt = torch.optim.AdamW(ddp.parameters(), lr=1e-3)
+        shard = ds[rank::world_size]
+        for _ in range(2):
+            batch = Batch.from_data_list([d.clone() for d in shard])
+            batch.pos.requires_grad_(True)
+            opt.zero_grad()
+            pred = ddp(batch)
+            loss, _ = model.energy_force_loss(pred, batch,
+                                              create_graph=True)
+            loss.backward()
+            opt.step()
+        p = torch.cat([x.flatten() for x in ddp.parameters()])
+        plist = [torch.zeros_like(p) for _ in range(world_size)]
+        dist.all_gather(plist, p)
+        same = all(torch.allclose(plist[0], pi, atol=1e-6)
+                   for pi in plist)
+        q.put((rank, bool(same), float(loss)))
+        dist.destroy_process_group()
+    except Exception as e:  # pragma: no cover
+        q.put((rank, False, repr(e)))
+
+
+def test_mace_force_training_ddp_gloo():
+    """MACE att-interaction force training (double backward) under DDP
+    on 2 gloo ranks: ranks end bitwise-synchronized."""
+    import multiprocessing as mp
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    port = 29531
+    ps = [ctx.Process(target=_mace_force_ddp_worker, args=(r, 2, port, q))
+          for r in range(2)]
+    for p in ps:
+        p.start()
+    results = [q.get(timeout=300) for _ in range(2)]
+    for p in ps:
+        p.join(timeout=60)
+    for rank, same, info in results:
+        assert same, f"rank {rank}: {info}"
