@@ -140,11 +140,9 @@ class MACEStack(Base):
         edge_sh = spherical_harmonics(vec, self.max_ell, normalize=True)
         edge_radial = self.radial_embedding(lengths)
         elem = self._node_elements(data)
-        # one_hot(elem) @ W == row lookup of W^T: an embedding gather
-        # instead of materializing [N, 118] + a GEMM (same parameters,
-        # identical numerics, same state dict)
-        h0 = torch.nn.functional.embedding(
-            elem, self.node_embedding.weight.t())  # [N, C]
+        one_hot = torch.nn.functional.one_hot(
+            elem, NUM_ELEMENTS).to(self.node_embedding.weight.dtype)
+        h0 = self.node_embedding(one_hot)  # [N, C]
         return h0, elem, edge_sh.to(h0.dtype), edge_radial.to(h0.dtype)
 
     def _edge_struct(self, data):
