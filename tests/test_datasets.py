@@ -246,3 +246,45 @@ def test_scatter_max_cpu_argmax_backward():
     assert out.flatten().tolist() == [5.0, 3.0]
     out.sum().backward()
     assert src.grad.flatten().tolist() == [0.0, 1.0, 1.0, 0.0]
+
+
+def test_train_from_serialized_pkl_flow(tmp_path, monkeypatch):
+    """End-to-end: total serialized pickle -> per-split pkls ->
+    load_train_val_test_sets -> a short training run (reference
+    test_datasetclass_inheritance pattern)."""
+    import os
+    import sys
+    sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+    from deterministic_graph_data import (base_config,
+                                          make_deterministic_dataset)
+    from hydragnn_amd.models import create_model_config
+    from hydragnn_amd.preprocess import (create_dataloaders,
+                                         load_train_val_test_sets,
+                                         total_to_train_val_test_pkls)
+    from hydragnn_amd.train import train
+    from hydragnn_amd.utils.config import update_config
+    from hydragnn_amd.utils.datasets.serializeddataset import (
+        SerializedWriter)
+    from hydragnn_amd.utils.optimizer import select_optimizer
+
+    monkeypatch.setenv("SERIALIZED_DATA_PATH", str(tmp_path))
+    ds = make_deterministic_dataset(num_samples=40, num_heads_node=0)
+    SerializedWriter(ds, f"{tmp_path}/serialized_dataset", "unit_test")
+    config = base_config("GIN", heads=("graph",), num_epoch=2)
+    config["Dataset"] = {"name": "unit_test", "path": {"total": "x"},
+                         "format": "pickle"}
+    config["NeuralNetwork"]["Training"]["perc_train"] = 0.7
+    total_to_train_val_test_pkls(config)
+    tr, va, te = load_train_val_test_sets(config)
+    assert len(tr) == 28
+    loaders = create_dataloaders(list(tr), list(va), list(te), 8,
+                                 config=config)
+    config = update_config(config, *loaders)
+    model = create_model_config(config["NeuralNetwork"], use_gpu=False)
+    opt = select_optimizer(
+        model, config["NeuralNetwork"]["Training"]["Optimizer"])
+    first = None
+    for _ in range(3):
+        err, _ = train(loaders[0], model, opt, 0)
+        first = first if first is not None else float(err)
+    assert float(err) < first
