@@ -106,3 +106,30 @@ def test_pbc_lj_forces_match_open_in_big_box():
     e2, f2 = _lj_energy_forces(pos, ei_p, sh, 0.01, 1.0)
     assert torch.allclose(e1, e2, atol=1e-10)
     assert torch.allclose(f1, f2, atol=1e-10)
+
+
+def test_rotated_sample_equivalence_after_normalization():
+    """Reference test_rotational_invariance pattern: rotate a
+    structure, rebuild edges, NormalizeRotation both — the samples are
+    equivalent (edge-order-insensitive) to 1e-14-grade tolerance."""
+    from hydragnn_amd.preprocess import (check_data_samples_equivalence,
+                                         get_radius_graph)
+    torch.manual_seed(0)
+    pos = torch.rand(24, 3, dtype=torch.float64)
+    q, _ = torch.linalg.qr(torch.randn(3, 3, dtype=torch.float64))
+    if torch.det(q) < 0:
+        q[:, 0] = -q[:, 0]
+
+    def build(p):
+        d = Data(pos=p.clone(), x=torch.ones(24, 1, dtype=torch.float64),
+                 y=torch.zeros(1, dtype=torch.float64))
+        normalize_rotation(d)
+        get_radius_graph(0.6, 32)(d)
+        src, dst = d.edge_index[0], d.edge_index[1]
+        d.edge_attr = (d.pos[src] - d.pos[dst]).norm(dim=-1,
+                                                     keepdim=True)
+        return d
+
+    d1 = build(pos)
+    d2 = build(pos @ q.t())
+    assert check_data_samples_equivalence(d1, d2, 1e-12)
