@@ -13,11 +13,10 @@ from typing import List
 import torch
 
 from ..data import Data
-from ..ops.geometry import (get_edge_vectors_and_lengths, radius_graph,
-                            radius_graph_pbc)
+from ..ops.geometry import radius_graph, radius_graph_pbc
 from ..utils.config.config_utils import (_calculate_avg_deg, _gather_deg,
                                          check_if_graph_size_variable)
-from .transforms import pbc_as_tensor, pbc_distance, pbc_local_cartesian
+from .transforms import pbc_distance, pbc_local_cartesian
 
 
 class RadiusGraph:
