@@ -22,6 +22,10 @@ from .basis import (
     cosine_cutoff,
 )
 from ._extension import has_extension, get_extension
+from .irreps_linear import irreps_linear
+from .varlen_attn import varlen_attention
+from .mfma_linear import MFMALinear
+from .splitk_linear import SplitKLinear
 
 __all__ = [
     "scatter", "gather", "segment_softmax", "degree",
@@ -30,4 +34,5 @@ __all__ = [
     "bessel_basis", "gaussian_basis", "chebyshev_basis", "sinc_basis",
     "polynomial_cutoff", "cosine_cutoff",
     "has_extension", "get_extension",
+    "irreps_linear", "varlen_attention", "MFMALinear", "SplitKLinear",
 ]
