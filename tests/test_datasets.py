@@ -288,3 +288,53 @@ def test_train_from_serialized_pkl_flow(tmp_path, monkeypatch):
         err, _ = train(loaders[0], model, opt, 0)
         first = first if first is not None else float(err)
     assert float(err) < first
+
+
+def test_graph_dataset_prepare_pipeline(tmp_path):
+    """load_pickled_graphs + prepare_graph_dataset: the historical
+    three-object pickle -> edges rebuilt per config -> y/y_loc
+    assembled (reference test_graph_dataset pattern)."""
+    import pickle
+
+    from hydragnn_amd.data import Data
+    from hydragnn_amd.preprocess.graph_dataset import (
+        load_and_prepare_graph_dataset)
+    torch.manual_seed(0)
+    samples = []
+    for _ in range(6):
+        pos = torch.rand(10, 3)
+        x = torch.cat([torch.randint(1, 4, (10, 1)).float(),
+                       torch.rand(10, 2)], dim=1)
+        samples.append(Data(x=x, pos=pos,
+                            y=torch.rand(3)))
+    p = tmp_path / "total.pkl"
+    with open(p, "wb") as f:
+        pickle.dump(None, f)
+        pickle.dump(None, f)
+        pickle.dump(samples, f)
+    config = {
+        "NeuralNetwork": {
+            "Architecture": {"radius": 0.8, "max_neighbours": 8,
+                             "periodic_boundary_conditions": False},
+            "Variables_of_interest": {
+                "type": ["graph", "node"],
+                "output_index": [0, 1],
+                "input_node_features": [0],
+            },
+        },
+        "Dataset": {
+            "graph_features": {"dim": [1, 2]},
+            "node_features": {"name": ["Z", "a", "b"],
+                              "dim": [1, 1, 1],
+                              "column_index": [0, 1, 2]},
+        },
+        "Verbosity": {"level": 0},
+    }
+    ds = load_and_prepare_graph_dataset(str(p), config)
+    d = ds[0]
+    assert d.edge_index.shape[0] == 2 and d.edge_index.shape[1] > 0
+    # y = 1 graph value + 10 node values; y_loc = [0, 1, 11]
+    assert d.y.shape[0] == 11
+    assert d.y_loc.tolist() == [[0, 1, 11]]
+    # input features restricted to column 0
+    assert d.x.shape[1] == 1
