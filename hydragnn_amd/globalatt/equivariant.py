@@ -313,7 +313,12 @@ class EquivariantLocalGlobalConv(nn.Module):
             **({"vec_state": vec_state} if vec_state is not None
                else {}), **kwargs)
         vec = vec_state.get("v") if vec_state is not None else None
-        feats = self.adapter.encode(inv_node_feat, vec)
+        # the adapter expects hidden-width scalars; a first layer whose
+        # input is still raw node features (stacks that embed inside
+        # conv 0, e.g. EGNN) feeds the conv OUTPUT to the global branch
+        enc_in = inv_node_feat if \
+            inv_node_feat.shape[-1] == inv.shape[-1] else inv
+        feats = self.adapter.encode(enc_in, vec)
         feats = self.layer(feats, equiv_node_feat, batch)
         g_inv, g_vec = self.adapter.decode(
             feats, torch.zeros_like(inv_node_feat), vec)

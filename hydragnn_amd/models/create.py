@@ -253,6 +253,13 @@ def create_model_config(config: dict, verbosity: int = 0,
         graph_attr_dim=arch.get("graph_attr_dim", 0),
         verbosity=verbosity,
         use_gpu=use_gpu,
+        **{k: arch[k] for k in (
+            "equivariant_attn_lmax", "equivariant_attn_num_radial",
+            "equivariant_attn_feedforward_multiplier",
+            "equivariant_attn_allow_scalar_only",
+            "equivariant_attn_require_tensor_coupling",
+            "equivariant_attn_chunk_size",
+            "equivariant_attn_coupling_mode") if k in arch},
     )
     _, param_dtype, _ = resolve_precision(training.get("precision", "fp32"))
     torch.set_default_dtype(param_dtype)

@@ -116,3 +116,31 @@ def test_painn_with_equivariant_transformer_trains():
         overrides=overrides)
     err, rmses = evaluate_error(model, loaders[2], config)
     assert rmses[0] < 0.6, f"PAINN+ET RMSE {rmses[0]:.3f}"
+
+
+@pytest.mark.parametrize("mpnn_type", ["PNAEq", "MACE", "EGNN"])
+def test_equivariant_transformer_integrations(mpnn_type):
+    """Reference test_equivariant_{pnaeq,mace,scalar_mpnn}_integration
+    pattern: each family trains with the equivariant global attention
+    engine and reaches its accuracy gate."""
+    overrides = {"NeuralNetwork": {"Architecture": {
+        "global_attn_engine": "EquivariantTransformer",
+        "global_attn_heads": 2,
+        "equivariant_attn_lmax": 1,
+        "equivariant_attn_num_radial": 8,
+        "equivariant_attn_chunk_size": 512,
+    }}}
+    if mpnn_type == "MACE":
+        overrides["NeuralNetwork"]["Architecture"].update(
+            {"max_ell": 2, "node_max_ell": 1, "correlation": 2,
+             "num_radial": 8})
+    if mpnn_type == "EGNN":
+        # scalar-only coupling (reference scalar_mpnn integration)
+        overrides["NeuralNetwork"]["Architecture"][
+            "equivariant_attn_allow_scalar_only"] = True
+    epochs = 25 if mpnn_type == "PNAEq" else 10
+    model, config, loaders = run_training(
+        mpnn_type, heads=("graph",), num_samples=48, num_epoch=epochs,
+        overrides=overrides)
+    err, rmses = evaluate_error(model, loaders[2], config)
+    assert rmses[0] < 0.8, f"{mpnn_type}+ET RMSE {rmses[0]:.3f}"
