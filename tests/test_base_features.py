@@ -170,3 +170,29 @@ def test_valtest_and_trace_level_flags(monkeypatch, tmp_path):
     tr.disable()
     tr.reset()
     assert (tmp_path / "gp_timing.p0").exists()
+
+
+def test_output_denormalize_and_unscale():
+    """postprocess: min-max denormalization per head and per-num-nodes
+    unscaling (reference postprocess.py patterns)."""
+    import torch
+    from hydragnn_amd.data import Data
+    from hydragnn_amd.postprocess.postprocess import (
+        output_denormalize, unscale_features_by_num_nodes,
+        unscale_features_by_num_nodes_config)
+    t = [torch.tensor([0.0, 0.5, 1.0])]
+    p = [torch.tensor([0.25, 0.5, 0.75])]
+    tt, pp = output_denormalize([[10.0, 30.0]], t, p)
+    assert torch.allclose(tt[0], torch.tensor([10.0, 20.0, 30.0]))
+    assert torch.allclose(pp[0], torch.tensor([15.0, 20.0, 25.0]))
+    d = Data(x=torch.ones(4, 1), y=torch.tensor([2.0]),
+             pos=torch.zeros(4, 3))
+    d.num_nodes = 4
+    unscale_features_by_num_nodes(d)
+    assert float(d.y) == 8.0
+    d2 = Data(x=torch.ones(4, 1), y=torch.tensor([2.0]),
+              pos=torch.zeros(4, 3))
+    d2.num_nodes = 4
+    cfg = {"NeuralNetwork": {"Variables_of_interest": {}}}
+    unscale_features_by_num_nodes_config(d2, cfg)  # gate off -> no-op
+    assert float(d2.y) == 2.0
