@@ -178,3 +178,38 @@ def test_evalonly_and_max_batch(monkeypatch, tmp_path):
     from hydragnn_amd.train import get_nbatch
     assert get_nbatch(loaders[0]) == 1
     monkeypatch.delenv("HYDRAGNN_MAX_NUM_BATCH")
+
+
+def test_all_example_configs_parse():
+    """Every committed example JSON passes update_config with a tiny
+    synthetic dataset (config-schema regression)."""
+    import glob
+    import json as _json
+    import os
+    from hydragnn_amd.preprocess import create_dataloaders
+    from hydragnn_amd.utils.config import update_config
+    from hydragnn_amd.utils.datasets.synthetic import lj_dataset
+    import torch
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    configs = sorted(glob.glob(os.path.join(repo, "examples", "*",
+                                            "*.json")))
+    assert configs, "no example configs found"
+    for path in configs:
+        with open(path) as f:
+            config = _json.load(f)
+        if "NeuralNetwork" not in config:
+            continue
+        ds = lj_dataset(num_samples=8, num_atoms=8, pbc=False)
+        # synthesize y/y_loc matching the config's head structure
+        types = config["NeuralNetwork"]["Variables_of_interest"].get(
+            "type", ["graph"])
+        for d in ds:
+            spans = [1 if t == "graph" else d.num_nodes for t in types]
+            locs = [0]
+            for sp in spans:
+                locs.append(locs[-1] + sp)
+            d.y = torch.zeros(locs[-1], 1)
+            d.y_loc = torch.tensor([locs], dtype=torch.long)
+        loaders = create_dataloaders(ds, ds, ds, 4, config=config)
+        out = update_config(_json.loads(_json.dumps(config)), *loaders)
+        assert "output_dim" in out["NeuralNetwork"]["Architecture"], path
