@@ -16,7 +16,7 @@ from deterministic_graph_data import base_config, make_deterministic_dataset
 
 def run_training(mpnn_type, heads=("graph",), num_samples=64,
                  num_epoch=30, overrides=None, dataset_kwargs=None,
-                 use_gpu=False):
+                 use_gpu=False, train_kwargs=None):
     torch.manual_seed(7)
     config = base_config(mpnn_type, heads=heads, num_epoch=num_epoch)
     if overrides:
@@ -39,7 +39,8 @@ def run_training(mpnn_type, heads=("graph",), num_samples=64,
     train_validate_test(
         model, optimizer, train_loader, val_loader, test_loader,
         writer=None, scheduler=None, config=config["NeuralNetwork"],
-        log_name=f"test_{mpnn_type}", verbosity=0)
+        log_name=f"test_{mpnn_type}", verbosity=0,
+        **(train_kwargs or {}))
     return model, config, (train_loader, val_loader, test_loader)
 
 

@@ -196,3 +196,21 @@ def test_output_denormalize_and_unscale():
     cfg = {"NeuralNetwork": {"Variables_of_interest": {}}}
     unscale_features_by_num_nodes_config(d2, cfg)  # gate off -> no-op
     assert float(d2.y) == 2.0
+
+
+def test_training_with_plot_creation(tmp_path, monkeypatch):
+    """create_plots=True renders the end-of-training scatter and
+    error-histogram files (reference final-plot branch)."""
+    pytest.importorskip("matplotlib")
+    import os
+    import sys
+    sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+    monkeypatch.chdir(tmp_path)
+    from _training_workflow import run_training
+    run_training("GIN", heads=("graph",), num_samples=16, num_epoch=2,
+                 train_kwargs={"create_plots": True})
+    import glob
+    pngs = glob.glob(str(tmp_path / "logs" / "**" / "*.png"),
+                     recursive=True)
+    assert any("scatter" in p for p in pngs), pngs
+    assert any("error_hist" in p for p in pngs), pngs
