@@ -213,3 +213,17 @@ def test_all_example_configs_parse():
         loaders = create_dataloaders(ds, ds, ds, 4, config=config)
         out = update_config(_json.loads(_json.dumps(config)), *loaders)
         assert "output_dim" in out["NeuralNetwork"]["Architecture"], path
+
+
+def test_checkpoint_best_metric_gating():
+    """Checkpoint saves only on improvement after warmup (reference
+    model.py:533-573)."""
+    from hydragnn_amd.utils.model.model import Checkpoint, EarlyStopping
+    ck = Checkpoint("t", warmup=2)
+    assert not ck(0, 1.0) and not ck(1, 0.5)   # warmup epochs
+    assert ck(2, 0.7)                          # first post-warmup
+    assert not ck(3, 0.8)                      # worse -> no save
+    assert ck(4, 0.6)                          # better -> save
+    es = EarlyStopping(patience=2, min_delta=0.0)
+    assert not es(1.0) and not es(1.1)
+    assert es(1.2)                             # patience exceeded
