@@ -266,3 +266,121 @@ def allgatherv_numpy(array, comm=None):
     dist.all_gather(outs, pad)
     return np.concatenate([o[:int(s.item())].numpy()
                            for o, s in zip(outs, sizes)])
+
+
+class ShardedDistDataset(AbstractBaseDataset):
+    """Distributed in-memory store with cross-rank sample fetches —
+    the DDStore data plane (reference distdataset.py DDStore method
+    'mpi'): each rank holds a shard; get(idx) for a remote sample
+    fetches it from the owner inside an epoch_begin/epoch_end window.
+
+    Transport is torch.distributed send/recv over gloo (multi-node
+    capable, GPU-independent): every rank runs a service thread during
+    the epoch window answering (idx) requests with the pickled sample.
+    Thread-safety: requests TO rank k travel on a per-rank process
+    group, so within each process the service thread and the main
+    thread operate on DISJOINT groups.  Construct collectively on all
+    ranks; use num_workers=0 loaders with this dataset."""
+
+    _REQ_TAG = 7771
+    _LEN_TAG = 7772
+    _PAYLOAD_TAG = 7773
+
+    def __init__(self, local_samples):
+        super().__init__()
+        import pickle as _pickle
+
+        import torch.distributed as dist
+        self._dist = dist
+        self.rank = dist.get_rank() if dist.is_initialized() else 0
+        self.world = dist.get_world_size() if dist.is_initialized() \
+            else 1
+        self._local = [_pickle.dumps(s) for s in local_samples]
+        counts = [len(self._local)]
+        if self.world > 1:
+            t = torch.tensor(counts, dtype=torch.long)
+            sizes = [torch.zeros(1, dtype=torch.long)
+                     for _ in range(self.world)]
+            dist.all_gather(sizes, t)
+            counts = [int(s) for s in sizes]
+            # per-owner request/reply groups: the rank-k service thread
+            # only touches groups[k]; a requesting main thread only
+            # touches groups[owner != k] -> disjoint per thread
+            self._req_g = [dist.new_group(backend="gloo")
+                           for _ in range(self.world)]
+            self._rep_g = [dist.new_group(backend="gloo")
+                           for _ in range(self.world)]
+        self.offsets = [0]
+        for c in counts:
+            self.offsets.append(self.offsets[-1] + c)
+        self.total = self.offsets[-1]
+        self._server = None
+        self._in_epoch = False
+
+    # -- ownership ----------------------------------------------------
+    def _owner(self, idx: int) -> int:
+        import bisect
+        return bisect.bisect_right(self.offsets, idx) - 1
+
+    def _serve(self):
+        dist = self._dist
+        g_req = self._req_g[self.rank]
+        g_rep = self._rep_g[self.rank]
+        req = torch.zeros(1, dtype=torch.long)
+        while True:
+            src = dist.recv(req, group=g_req, tag=self._REQ_TAG)
+            gidx = int(req[0])
+            if gidx < 0:          # stop sentinel
+                return
+            payload = self._local[gidx - self.offsets[self.rank]]
+            buf = torch.frombuffer(bytearray(payload), dtype=torch.uint8)
+            n = torch.tensor([buf.numel()], dtype=torch.long)
+            dist.send(n, dst=src, group=g_rep, tag=self._LEN_TAG)
+            dist.send(buf, dst=src, group=g_rep,
+                      tag=self._PAYLOAD_TAG)
+
+    def epoch_begin(self):
+        if self.world > 1 and self._server is None:
+            import threading
+            self._server = threading.Thread(target=self._serve,
+                                            daemon=True)
+            self._server.start()
+        self._in_epoch = True
+
+    def epoch_end(self):
+        self._in_epoch = False
+        if self.world > 1 and self._server is not None:
+            dist = self._dist
+            dist.barrier()
+            # ring stop: each rank stops its right neighbor's server
+            nbr = (self.rank + 1) % self.world
+            stop = torch.tensor([-1], dtype=torch.long)
+            dist.send(stop, dst=nbr, group=self._req_g[nbr],
+                      tag=self._REQ_TAG)
+            self._server.join(timeout=60)
+            self._server = None
+            dist.barrier()
+
+    # -- dataset API --------------------------------------------------
+    def len(self):
+        return self.total
+
+    def get(self, idx):
+        import pickle as _pickle
+        owner = self._owner(idx)
+        if owner == self.rank:
+            return _pickle.loads(self._local[idx - self.offsets[owner]])
+        if not self._in_epoch:
+            raise RuntimeError(
+                "remote sample fetch outside epoch_begin/epoch_end")
+        dist = self._dist
+        req = torch.tensor([idx], dtype=torch.long)
+        dist.send(req, dst=owner, group=self._req_g[owner],
+                  tag=self._REQ_TAG)
+        n = torch.zeros(1, dtype=torch.long)
+        dist.recv(n, src=owner, group=self._rep_g[owner],
+                  tag=self._LEN_TAG)
+        buf = torch.zeros(int(n[0]), dtype=torch.uint8)
+        dist.recv(buf, src=owner, group=self._rep_g[owner],
+                  tag=self._PAYLOAD_TAG)
+        return _pickle.loads(buf.numpy().tobytes())
