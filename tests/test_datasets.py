@@ -397,3 +397,64 @@ def test_sharded_dist_dataset_cross_rank_fetch():
         p.join(timeout=60)
     for rank, ok, info in results:
         assert ok, f"rank {rank}: {info}"
+
+
+def _sharded_train_worker(rank, world_size, port, q):
+    try:
+        import os
+        import sys
+
+        import torch.distributed as dist
+        os.environ.update({"MASTER_ADDR": "127.0.0.1",
+                           "MASTER_PORT": str(port)})
+        dist.init_process_group("gloo", rank=rank,
+                                world_size=world_size)
+        sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+        from deterministic_graph_data import (base_config,
+                                              make_deterministic_dataset)
+        from hydragnn_amd.models import create_model_config
+        from hydragnn_amd.preprocess import create_dataloaders
+        from hydragnn_amd.train import train as train_fn
+        from hydragnn_amd.utils.config import update_config
+        from hydragnn_amd.utils.datasets.graphstore import (
+            ShardedDistDataset)
+        from hydragnn_amd.utils.optimizer import select_optimizer
+        torch.manual_seed(5)
+        # each rank contributes a different shard to the global store
+        full = make_deterministic_dataset(num_samples=16,
+                                          num_heads_node=0)
+        ds = ShardedDistDataset(full[rank::world_size])
+        config = base_config("GIN", heads=("graph",), num_epoch=1)
+        # config derivation iterates samples -> needs the fetch window
+        ds.epoch_begin()
+        loaders = create_dataloaders(ds, ds, ds, 4, config=config)
+        config = update_config(config, *loaders)
+        ds.epoch_end()
+        model = create_model_config(config["NeuralNetwork"],
+                                    use_gpu=False)
+        opt = select_optimizer(
+            model, config["NeuralNetwork"]["Training"]["Optimizer"])
+        # the train loop opens/closes the fetch window itself
+        err, _ = train_fn(loaders[0], model, opt, 0)
+        q.put((rank, bool(err == err), float(err)))
+        dist.destroy_process_group()
+    except Exception as e:  # pragma: no cover
+        q.put((rank, False, repr(e)))
+
+
+def test_sharded_dataset_inside_train_loop():
+    """The train loop drives the sharded store's epoch windows and
+    every rank trains over the GLOBAL dataset (cross-rank fetches)."""
+    import multiprocessing as mp
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    port = 29541
+    ps = [ctx.Process(target=_sharded_train_worker,
+                      args=(r, 2, port, q)) for r in range(2)]
+    for p in ps:
+        p.start()
+    results = [q.get(timeout=240) for _ in range(2)]
+    for p in ps:
+        p.join(timeout=60)
+    for rank, ok, info in results:
+        assert ok, f"rank {rank}: {info}"
