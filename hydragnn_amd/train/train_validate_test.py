@@ -235,6 +235,11 @@ def train(loader, model, opt, verbosity, precision="fp32", profiler=None):
 def _eval_pass(loader, model, verbosity, precision, return_samples=False):
     m = _unwrap(model)
     use_ia = _use_interatomic(m)
+    # DDStore-style fetch window for sharded/remote datasets
+    _ds = getattr(loader, "dataset", None)
+    _window = _ds is not None and hasattr(_ds, "epoch_begin")
+    if _window:
+        _ds.epoch_begin()
     num_tasks = len(m.loss_weights) if not use_ia else 3
     device = get_device()
     total_error = torch.zeros(1, device=device)
@@ -277,6 +282,8 @@ def _eval_pass(loader, model, verbosity, precision, return_samples=False):
             tasks_error[it] += tl.detach() * n
         num_samples_local += n
 
+    if _window:
+        _ds.epoch_end()
     denom = max(num_samples_local, 1)
     err = reduce_values_ranks(total_error / denom)
     tasks_err = reduce_values_ranks(tasks_error / denom)
