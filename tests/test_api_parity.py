@@ -112,3 +112,23 @@ def test_train_and_misc_aliases():
     assert np.allclose(x, np.ones(2))
     assert not is_model_distributed(torch.nn.Linear(2, 2))
     assert get_device_from_name("cpu").type == "cpu"
+
+
+def test_top_level_package_surfaces():
+    """Reference import paths: package-level names a migrating script
+    uses directly."""
+    import hydragnn_amd
+    import hydragnn_amd.models as m
+    import hydragnn_amd.utils as u
+    assert callable(hydragnn_amd.run_training)
+    assert callable(hydragnn_amd.run_prediction)
+    for n in ("MACEStack", "PAINNStack", "EGCLStack", "SCFStack",
+              "DIMEStack", "PNAEqStack", "PNAPlusStack", "CGCNNStack",
+              "GATStack", "GINStack", "MFCStack", "PNAStack",
+              "SAGEStack", "MultiTaskModelMP", "DualOptimizer",
+              "create_model_config"):
+        assert hasattr(m, n), n
+    for n in ("setup_ddp", "get_comm_size_and_rank", "update_config",
+              "save_model", "load_existing_model", "setup_log",
+              "get_device", "distributed_model_wrapper"):
+        assert hasattr(u, n), n
