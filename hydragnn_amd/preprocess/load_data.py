@@ -76,7 +76,10 @@ def create_dataloaders(trainset, valset, testset, batch_size: int,
                 cls = (DistributedCostAwareBatchSampler if use_dist
                        else CostAwareBatchSampler)
                 bs = cls(ds, max_nodes=batching["max_nodes"],
-                         shuffle=shuffle, seed=batching.get("seed", 0))
+                         shuffle=shuffle, seed=batching.get("seed", 0),
+                         oversized_policy=batching.get(
+                             "oversized_policy", "error"),
+                         drop_last=batching.get("drop_last", False))
             return HydraDataLoader(
                 ds, batch_size=batch_size, shuffle=shuffle,
                 batch_sampler=bs,
