@@ -22,27 +22,10 @@
 
 namespace {
 
-constexpr int kGroup = 32;            // threads per (i, c) pair
-constexpr int kGroupsPerBlock = 8;    // 256-thread blocks
-constexpr int kMaxDim = 40;           // max of da/db/dg/do
-
-template <typename T>
-__device__ inline float to_f32(T v) { return (float)v; }
-template <>
-__device__ inline float to_f32<__hip_bfloat16>(__hip_bfloat16 v) {
-  return __bfloat162float(v);
-}
-template <typename T>
-__device__ inline T from_f32(float v) { return (T)v; }
-
 // accumulator type: fp32 for bf16/half/float inputs, fp64 for double
 // (fp64 ETP support: slices and channel reductions keep full width)
 template <typename T> struct acc_of { using type = float; };
 template <> struct acc_of<double> { using type = double; };
-template <>
-__device__ inline __hip_bfloat16 from_f32<__hip_bfloat16>(float v) {
-  return __float2bfloat16(v);
-}
 
 // out[i,c,o] = sum over entries(coef, a, b, g, o) of
 //              coef * A[i,c,a] * B[i,b] * C[i,c,g]

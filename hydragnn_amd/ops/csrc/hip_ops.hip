@@ -605,7 +605,6 @@ std::vector<torch::Tensor> scatter_minmax_fwd(torch::Tensor src,
   {
     float fill = is_max ? -INFINITY : INFINITY;
     unsigned int u;
-    float flipped;
     // host-side flip of the fill value
     unsigned int raw;
     memcpy(&raw, &fill, 4);

@@ -190,7 +190,6 @@ __global__ __launch_bounds__(256) void irreps_linear_gw_kernel(
         bool isx = idx < cx;
         int id2 = isx ? idx : idx - cx;
         int re = isx ? rex : reg_;
-        int C = isx ? Cin : Cout;
         int P = isx ? CP : GP;
         __hip_bfloat16* dst = isx ? lA : lG;
         const __hip_bfloat16* src = isx ? X : G;
