@@ -13,13 +13,22 @@ from hydragnn_amd.ops import spherical_harmonics
 
 def _complex_sh(l, v):
     """scipy complex SH evaluated at cartesian unit vectors, m=-l..l."""
-    from scipy.special import sph_harm
+    try:
+        from scipy.special import sph_harm_y
+
+        def _sh(m, l_, phi, theta):
+            return sph_harm_y(l_, m, theta, phi)
+    except ImportError:  # scipy < 1.15
+        from scipy.special import sph_harm
+
+        def _sh(m, l_, phi, theta):
+            return sph_harm(m, l_, phi, theta)
     x, y, z = v[:, 0], v[:, 1], v[:, 2]
     theta = np.arccos(np.clip(z, -1, 1))       # polar
     phi = np.arctan2(y, x)                     # azimuth
     out = np.zeros((v.shape[0], 2 * l + 1), dtype=np.complex128)
     for m in range(-l, l + 1):
-        out[:, m + l] = sph_harm(m, l, phi, theta)
+        out[:, m + l] = _sh(m, l, phi, theta)
     return out
 
 
