@@ -129,3 +129,25 @@ def test_interaction_type_switch():
     att0 = core_att.interactions[0]
     expected = core_att.radial_embedding.out_dim + 2 * core_att.hidden_dim
     assert att0.radial_mlp[0].in_features == expected
+
+
+def test_mace_checkpointing_with_forces():
+    """conv_checkpointing + att interaction + force double backward —
+    the checkpoint/create_graph combination must keep every gradient
+    finite (reference Training.conv_checkpointing path)."""
+    import torch
+    from hydragnn_amd.data import Batch
+    from hydragnn_amd.utils.datasets.synthetic import md17_shape_dataset
+    ds = md17_shape_dataset(num_samples=4)
+    cfg = _mace_config()
+    cfg["NeuralNetwork"]["Training"]["conv_checkpointing"] = True
+    model, config, _ = _build(cfg, ds)
+    core = model.model if hasattr(model, "model") else model
+    assert core.conv_checkpointing
+    batch = Batch.from_data_list(ds)
+    batch.pos.requires_grad_(True)
+    pred = model(batch)
+    loss, _ = model.energy_force_loss(pred, batch, create_graph=True)
+    loss.backward()
+    grads = [p.grad for p in model.parameters() if p.grad is not None]
+    assert grads and all(torch.isfinite(g).all() for g in grads)
