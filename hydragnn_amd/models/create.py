@@ -17,6 +17,7 @@ from .stacks import CGCNNStack, GATStack, GINStack, MFCStack, PNAStack, SAGEStac
 
 PRECISION_MAP = {
     "bf16": {"param_dtype": torch.float32, "autocast_dtype": torch.bfloat16},
+    "fp16": {"param_dtype": torch.float32, "autocast_dtype": torch.float16},
     "fp32": {"param_dtype": torch.float32, "autocast_dtype": None},
     "fp64": {"param_dtype": torch.float64, "autocast_dtype": None},
 }
@@ -26,7 +27,8 @@ def resolve_precision(precision):
     if precision is None:
         precision = "fp32"
     prec = str(precision).lower()
-    aliases = {"bfloat16": "bf16", "float32": "fp32", "float": "fp32",
+    aliases = {"bfloat16": "bf16", "float16": "fp16", "half": "fp16",
+               "float32": "fp32", "float": "fp32",
                "float64": "fp64", "double": "fp64"}
     prec = aliases.get(prec, prec)
     if prec not in PRECISION_MAP:

@@ -34,6 +34,10 @@ from .symmetric_contraction import SymmetricContraction
 
 
 class RadialEmbeddingBlock(nn.Module):
+    """Radial basis x polynomial cutoff (reference blocks.py:151-187):
+    the cutoff is evaluated on the RAW edge length; the optional
+    Agnesi/Soft distance transform is applied only to the basis input."""
+
     def __init__(self, r_max: float, num_bessel: int = 8,
                  num_polynomial_cutoff: int = 5,
                  radial_type: str = "bessel",
@@ -43,7 +47,12 @@ class RadialEmbeddingBlock(nn.Module):
         self.num_bessel = num_bessel
         self.p = float(num_polynomial_cutoff)
         self.radial_type = (radial_type or "bessel").lower()
-        self.distance_transform = distance_transform
+        if distance_transform == "Agnesi":
+            self.distance_transform = AgnesiTransform()
+        elif distance_transform == "Soft":
+            self.distance_transform = SoftTransform()
+        else:
+            self.distance_transform = None
         if self.radial_type == "bessel":
             self.bessel_weights = nn.Parameter(
                 torch.arange(1, num_bessel + 1, dtype=torch.float)
@@ -59,18 +68,14 @@ class RadialEmbeddingBlock(nn.Module):
         else:
             raise ValueError(f"unknown radial_type {radial_type}")
 
-    def _transform(self, r: torch.Tensor) -> torch.Tensor:
-        if self.distance_transform == "Agnesi":
-            a, q, p = 0.9, 0.9, 4.0
-            x = r / self.r_max
-            return r * (1.0 + a * x.pow(q) / (1.0 + x.pow(q - p)))
-        if self.distance_transform == "Soft":
-            return self.r_max * (
-                1.0 - torch.exp(-r / self.r_max)).clamp(max=0.999)
-        return r
-
-    def forward(self, lengths: torch.Tensor) -> torch.Tensor:
-        r = self._transform(lengths)  # [E, 1]
+    def forward(self, lengths: torch.Tensor,
+                z: Optional[torch.Tensor] = None,
+                edge_index: Optional[torch.Tensor] = None) -> torch.Tensor:
+        cutoff = polynomial_cutoff(lengths, self.r_max, self.p)
+        r = lengths
+        if self.distance_transform is not None:
+            r = self.distance_transform(lengths, z=z,
+                                        edge_index=edge_index)
         if self.radial_type == "bessel":
             rb = bessel_basis(r, self.r_max, self.bessel_weights.to(r.dtype))
         elif self.radial_type == "gaussian":
@@ -78,7 +83,6 @@ class RadialEmbeddingBlock(nn.Module):
                                 self.coeff)
         else:
             rb = chebyshev_basis(r, self.r_max, self.num_bessel)
-        cutoff = polynomial_cutoff(r, self.r_max, self.p)
         return rb * cutoff
 
 
@@ -231,14 +235,16 @@ class RealAgnosticResidualInteractionBlock(nn.Module):
             m = scatter(mji.reshape(mji.shape[0], -1), dst, n, "sum",
                         sorted_index=edges_sorted).view(n, c, -1)
         m = m / self.avg_num_neighbors
-        # skip: per-l linear on the input, padded to lmax_out; fused
-        # into the main linear's epilogue on the kernel path
+        # skip: per-l linear on the input, padded to lmax_out.  Returned
+        # SEPARATELY (reference blocks.py:379-412): the product basis
+        # contracts the message alone and adds sc as a residual in its
+        # own linear's epilogue.
         sc = self.skip_linear(node_feats[:, :, :dim(min(self.lmax_node,
                                                         self.lmax_out))])
         want = dim(self.lmax_out)
         if sc.shape[-1] < want:
             sc = torch.nn.functional.pad(sc, (0, want - sc.shape[-1]))
-        return self.linear(m, add=sc)
+        return self.linear(m), sc
 
 
 class RealAgnosticAttResidualInteractionBlock(
@@ -360,8 +366,35 @@ class ChebychevBasis(nn.Module):
         return chebyshev_basis(x, self.r_max, self.num_basis)
 
 
+# Cordero et al. 2008 single-bond covalent radii in Angstrom, indexed by
+# atomic number (index 0 = placeholder); 0.2 where no published value.
+COVALENT_RADII = [
+    0.2,
+    0.31, 0.28, 1.28, 0.96, 0.84, 0.76, 0.71, 0.66, 0.57, 0.58,
+    1.66, 1.41, 1.21, 1.11, 1.07, 1.05, 1.02, 1.06,
+    2.03, 1.76, 1.70, 1.60, 1.53, 1.39, 1.61, 1.52, 1.50, 1.24, 1.32,
+    1.22, 1.22, 1.20, 1.19, 1.20, 1.20, 1.16,
+    2.20, 1.95, 1.90, 1.75, 1.64, 1.54, 1.47, 1.46, 1.42, 1.39, 1.45,
+    1.44, 1.42, 1.39, 1.39, 1.38, 1.39, 1.40,
+    2.44, 2.15, 2.07, 2.04, 2.03, 2.01, 1.99, 1.98, 1.98, 1.96, 1.94,
+    1.92, 1.92, 1.89, 1.90, 1.87, 1.87, 1.75, 1.70, 1.62, 1.51, 1.44,
+    1.41, 1.36, 1.36, 1.32, 1.45, 1.46, 1.48, 1.40, 1.50, 1.50,
+    2.60, 2.21, 2.15, 2.06, 2.00, 1.96, 1.90, 1.87, 1.80, 1.69,
+] + [0.2] * 22  # Z=97..118 unknown
+
+
+def _edge_r0(z, edge_index, covalent_radii, scale):
+    """Per-edge covalent length scale: scale * (r_cov[Z_u] + r_cov[Z_v]).
+    z holds 1-based atomic numbers per node."""
+    zc = z.clamp(0, covalent_radii.numel() - 1)
+    rc = covalent_radii.index_select(0, zc)
+    return scale * (rc[edge_index[0]] + rc[edge_index[1]]).unsqueeze(-1)
+
+
 class AgnesiTransform(nn.Module):
-    """Agnesi distance transform (ACEpotentials.jl / JCP 2023)."""
+    """Agnesi distance transform (ACEpotentials.jl / JCP 2023; reference
+    radial.py:161-208): r -> (1 + a (r/r0)^q / (1 + (r/r0)^(q-p)))^-1
+    with r0 = mean covalent radius of the edge endpoints."""
 
     def __init__(self, q: float = 0.9183, p: float = 4.5791,
                  a: float = 1.0805, trainable: bool = False):
@@ -375,28 +408,53 @@ class AgnesiTransform(nn.Module):
             self.register_buffer("q", t(q))
             self.register_buffer("p", t(p))
             self.register_buffer("a", t(a))
+        self.register_buffer("covalent_radii",
+                             torch.tensor(COVALENT_RADII))
 
     def forward(self, x, node_attrs=None, edge_index=None,
-                atomic_numbers=None):
-        r0 = 1.0  # covalent-radius scale (no ase tables in this image)
+                atomic_numbers=None, z=None):
+        if z is None and node_attrs is not None and \
+                atomic_numbers is not None:
+            z = atomic_numbers[torch.argmax(node_attrs, dim=1)]
+        if z is not None and edge_index is not None:
+            r0 = _edge_r0(z, edge_index,
+                          self.covalent_radii.to(x.dtype), 0.5)
+        else:
+            r0 = 1.0
         xs = x / r0
         return 1.0 / (1.0 + self.a * xs.pow(self.q)
                       / (1.0 + xs.pow(self.q - self.p)))
 
 
 class SoftTransform(nn.Module):
-    def __init__(self, alpha: float = 4.0, trainable: bool = False):
+    """Soft distance transform (reference radial.py:214-258):
+    y = x + tanh(-(x/r0) - a (x/r0)^b)/2 + 1/2,
+    r0 = (r_cov[Z_u] + r_cov[Z_v]) / 4."""
+
+    def __init__(self, a: float = 0.2, b: float = 3.0,
+                 trainable: bool = False):
         super().__init__()
-        t = torch.tensor(alpha)
+        t = torch.tensor
         if trainable:
-            self.alpha = nn.Parameter(t)
+            self.a, self.b = nn.Parameter(t(a)), nn.Parameter(t(b))
         else:
-            self.register_buffer("alpha", t)
+            self.register_buffer("a", t(a))
+            self.register_buffer("b", t(b))
+        self.register_buffer("covalent_radii",
+                             torch.tensor(COVALENT_RADII))
 
     def forward(self, x, node_attrs=None, edge_index=None,
-                atomic_numbers=None):
-        return x * torch.sigmoid(self.alpha * (x - 1.0)) + (
-            1.0 - torch.sigmoid(self.alpha * (x - 1.0)))
+                atomic_numbers=None, z=None):
+        if z is None and node_attrs is not None and \
+                atomic_numbers is not None:
+            z = atomic_numbers[torch.argmax(node_attrs, dim=1)]
+        if z is not None and edge_index is not None:
+            r0 = _edge_r0(z, edge_index,
+                          self.covalent_radii.to(x.dtype), 0.25)
+        else:
+            r0 = 1.0
+        xs = x / r0
+        return x + 0.5 * torch.tanh(-xs - self.a * xs.pow(self.b)) + 0.5
 
 
 class LinearNodeEmbeddingBlock(nn.Module):

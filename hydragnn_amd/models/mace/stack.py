@@ -116,10 +116,13 @@ class MACEStack(Base):
 
     # ------------------------------------------------------------------
     def _node_elements(self, data) -> torch.Tensor:
+        """0-based element index from 1-based atomic number Z
+        (reference convention Z-1; keeps Z=118 from colliding with 117
+        and uses index 0 for hydrogen)."""
         z = data.get("z")
         if z is None:
             z = data.x[:, 0].long()
-        return z.clamp(min=0, max=NUM_ELEMENTS - 1).long()
+        return (z.long() - 1).clamp(min=0, max=NUM_ELEMENTS - 1)
 
     def _embedding(self, data):
         pos = data.pos
@@ -138,8 +141,9 @@ class MACEStack(Base):
         vec, lengths = get_edge_vectors_and_lengths(
             pos_c, data.edge_index, data.get("edge_shifts"))
         edge_sh = spherical_harmonics(vec, self.max_ell, normalize=True)
-        edge_radial = self.radial_embedding(lengths)
         elem = self._node_elements(data)
+        edge_radial = self.radial_embedding(lengths, z=elem + 1,
+                                            edge_index=data.edge_index)
         one_hot = torch.nn.functional.one_hot(
             elem, NUM_ELEMENTS).to(self.node_embedding.weight.dtype)
         h0 = self.node_embedding(one_hot)  # [N, C]
@@ -190,11 +194,12 @@ class MACEStack(Base):
             if node_feats.shape[-1] < want:
                 node_feats = torch.nn.functional.pad(
                     node_feats, (0, want - node_feats.shape[-1]))
-            m = inter(node_feats, data.edge_index, edge_sh, edge_radial,
-                      edges_sorted=bool(data.get("edges_sorted_",
-                                                 False)),
-                      etp_meta=etp_meta)
-            node_feats = prod(m, elem, sc=m)
+            m, sc = inter(node_feats, data.edge_index, edge_sh,
+                          edge_radial,
+                          edges_sorted=bool(data.get("edges_sorted_",
+                                                     False)),
+                          etp_meta=etp_meta)
+            node_feats = prod(m, elem, sc=sc)
             for ihead in range(self.num_heads):
                 r = self.readouts[ihead][ilayer](node_feats)
                 head_outputs[ihead] = r if head_outputs[ihead] is None \

@@ -137,25 +137,47 @@ def main():
     from .utils.datasets.synthetic import lj_dataset
     from .utils.model import load_existing_model
 
+    import os
+
     parser = argparse.ArgumentParser()
-    parser.add_argument("--config", required=True)
+    parser.add_argument("--config", default=None,
+                        help="raw config JSON (only needed when the "
+                             "checkpoint dir has no saved config.json)")
     parser.add_argument("--checkpoint", default=None,
                         help="log name to load the .pk checkpoint from")
+    parser.add_argument("--logdir", default="./logs/")
     parser.add_argument("--host", default="127.0.0.1")
     parser.add_argument("--port", type=int, default=8000)
     args = parser.parse_args()
 
-    with open(args.config) as f:
-        config = json.load(f)
-    # derive output dims the same way training does (tiny synthetic set)
-    ds = lj_dataset(num_samples=8, num_atoms=8, pbc=False)
-    loaders = create_dataloaders(
-        ds, ds, ds, config["NeuralNetwork"]["Training"]["batch_size"],
-        config=config)
-    config = update_config(config, *loaders)
+    # Prefer the config.json that training saved next to the checkpoint
+    # (save_config): it already carries the dataset-derived fields
+    # (output dims from y_loc, pna_deg, ...), so the architecture built
+    # here matches the checkpointed weights exactly.
+    saved_cfg = (os.path.join(args.logdir, args.checkpoint, "config.json")
+                 if args.checkpoint else None)
+    if saved_cfg and os.path.exists(saved_cfg):
+        with open(saved_cfg) as f:
+            config = json.load(f)
+    elif args.config:
+        with open(args.config) as f:
+            config = json.load(f)
+        if "output_dim" not in config["NeuralNetwork"]["Architecture"]:
+            # raw config: derive output dims the way training does,
+            # over a tiny synthetic set (shape-only; weights come from
+            # the checkpoint)
+            ds = lj_dataset(num_samples=8, num_atoms=8, pbc=False)
+            loaders = create_dataloaders(
+                ds, ds, ds,
+                config["NeuralNetwork"]["Training"]["batch_size"],
+                config=config)
+            config = update_config(config, *loaders)
+    else:
+        raise SystemExit("need --config or a checkpoint dir with a "
+                         "saved config.json")
     model = create_model_config(config["NeuralNetwork"])
     if args.checkpoint:
-        load_existing_model(model, args.checkpoint)
+        load_existing_model(model, args.checkpoint, path=args.logdir)
     app = create_app(model, config)
     uvicorn.run(app, host=args.host, port=args.port)
 

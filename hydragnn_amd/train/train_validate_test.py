@@ -42,14 +42,24 @@ def move_batch_to_device(data, param_dtype):
 
 
 def get_autocast_and_scaler(precision):
+    """Autocast context + loss scaler per precision (reference
+    train_validate_test.py:87-110 + GradScaler machinery :783-801):
+    bf16 -> autocast, no scaler; fp16 -> autocast + GradScaler (loss
+    scaling against underflow; GPU only); fp32/fp64 -> neither."""
     precision, _, autocast_dtype = resolve_precision(precision)
+    device = get_device()
     if precision == "bf16":
-        device = get_device()
         use_bf16 = device.type == "cuda" or bool(
             getattr(torch.backends.cpu, "has_bf16", False))
         if use_bf16:
             return torch.autocast(device_type=device.type,
                                   dtype=autocast_dtype), None
+    if precision == "fp16":
+        if device.type == "cuda":
+            scaler = torch.amp.GradScaler("cuda")
+            return torch.autocast(device_type="cuda",
+                                  dtype=autocast_dtype), scaler
+        log("Requested fp16 on CPU; falling back to full precision.")
     return nullcontext(), None
 
 
@@ -115,22 +125,42 @@ def _head_indices_node_or_mixed(m, data):
 # metric reductions
 # ---------------------------------------------------------------------------
 @torch.no_grad()
-def reduce_values_ranks(local_tensor):
+def reduce_values_ranks(local_tensor, count=None):
     """Cross-rank mean; HYDRAGNN_AGGR_BACKEND=mpi routes the tiny
     metric reductions through MPI instead of RCCL (the reference's
-    at-scale tuning) when mpi4py is available."""
+    at-scale tuning) when mpi4py is available.
+
+    With `count` (local sample count), `local_tensor` is a local SUM and
+    the global mean is sum-of-sums / sum-of-counts — unbiased when ranks
+    hold unequal sample counts (node-budget batching, uneven shards).
+    Without it, falls back to averaging per-rank values equally."""
     if not (dist.is_initialized() and dist.get_world_size() > 1):
+        if count is not None:
+            return local_tensor / max(float(count), 1.0)
         return local_tensor
+    if count is not None:
+        packed = torch.cat([
+            local_tensor.reshape(-1).double(),
+            torch.tensor([float(count)], dtype=torch.float64,
+                         device=local_tensor.device)])
+        packed = _allreduce_sum(packed)
+        total = packed[:-1] / packed[-1].clamp(min=1.0)
+        return total.to(local_tensor.dtype).view_as(local_tensor)
+    return _allreduce_sum(local_tensor) / dist.get_world_size()
+
+
+@torch.no_grad()
+def _allreduce_sum(t):
     if os.getenv("HYDRAGNN_AGGR_BACKEND") == "mpi":
         try:
             from mpi4py import MPI
             out = MPI.COMM_WORLD.allreduce(
-                local_tensor.detach().cpu().numpy(), op=MPI.SUM)
-            return torch.as_tensor(out) / dist.get_world_size()
+                t.detach().cpu().numpy(), op=MPI.SUM)
+            return torch.as_tensor(out, dtype=t.dtype, device=t.device)
         except ImportError:
             pass
-    dist.all_reduce(local_tensor, op=dist.ReduceOp.SUM)
-    return local_tensor / dist.get_world_size()
+    dist.all_reduce(t, op=dist.ReduceOp.SUM)
+    return t
 
 
 @torch.no_grad()
@@ -208,10 +238,17 @@ def train(loader, model, opt, verbosity, precision="fp32", profiler=None):
                                                 create_graph=True)
         tr.stop("forward")
         tr.start("backward")
-        loss.backward()
+        if scaler is not None:
+            scaler.scale(loss).backward()
+        else:
+            loss.backward()
         tr.stop("backward")
         tr.start("opt_step")
-        opt.step()
+        if scaler is not None:
+            scaler.step(opt)
+            scaler.update()
+        else:
+            opt.step()
         tr.stop("opt_step")
         if use_ia and is_fsdp2_enabled():
             set_reshard_after_backward(model, True)
@@ -225,9 +262,9 @@ def train(loader, model, opt, verbosity, precision="fp32", profiler=None):
 
     if dataset is not None and hasattr(dataset, "epoch_end"):
         dataset.epoch_end()
-    denom = max(num_samples_local, 1)
-    train_error = reduce_values_ranks(total_error / denom)
-    train_tasks_error = reduce_values_ranks(tasks_error / denom)
+    train_error = reduce_values_ranks(total_error, num_samples_local)
+    train_tasks_error = reduce_values_ranks(tasks_error,
+                                            num_samples_local)
     return train_error, train_tasks_error
 
 
@@ -284,9 +321,8 @@ def _eval_pass(loader, model, verbosity, precision, return_samples=False):
 
     if _window:
         _ds.epoch_end()
-    denom = max(num_samples_local, 1)
-    err = reduce_values_ranks(total_error / denom)
-    tasks_err = reduce_values_ranks(tasks_error / denom)
+    err = reduce_values_ranks(total_error, num_samples_local)
+    tasks_err = reduce_values_ranks(tasks_error, num_samples_local)
     if return_samples:
         tv = [gather_tensor_ranks(torch.cat(v, 0)) if v else torch.zeros(0, 1)
               for v in true_values]
