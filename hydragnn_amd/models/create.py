@@ -268,6 +268,32 @@ def create_model_config(config: dict, verbosity: int = 0,
     return model.to(dtype=param_dtype)
 
 
+def _make_weighted_loss(loss_type: str, w: torch.Tensor):
+    """Per-sample weighted mean loss: ``sum(w*err)/sum(w*1)`` with
+    ``w`` broadcast from the leading dim.  Equals the unweighted
+    nn.MSELoss/L1Loss mean when all weights are 1; zero-weight samples
+    contribute neither value nor gradient (used by the static-shape
+    pad graph, preprocess/static_batch.py)."""
+    lt = (loss_type or "mse").lower()
+
+    def fn(pred, true):
+        err = pred.float() - true.float()
+        if lt == "mse":
+            err = err * err
+        elif lt in ("mae", "l1"):
+            err = err.abs()
+        else:
+            raise NotImplementedError(
+                f"loss_weight_g not supported for loss '{loss_type}'")
+        ww = w.to(err.device)
+        while ww.dim() < err.dim():
+            ww = ww.unsqueeze(-1)
+        ww = ww.expand_as(err)
+        return (err * ww).sum() / ww.sum().clamp_min(1e-12)
+
+    return fn
+
+
 class EnhancedModelWrapper(torch.nn.Module):
     """MLIP composition wrapper: energy + energy/atom + forces
     (forces = -dE/dpos via autograd, create_graph=True during training
@@ -320,7 +346,15 @@ class EnhancedModelWrapper(torch.nn.Module):
             raise ValueError("Force training needs node or graph energy head")
 
         graph_energy_true = data.energy.squeeze().float()
-        loss_fn = self.loss_function
+        # Per-graph loss weights (e.g. the static-shape pad graph at 0,
+        # see preprocess/static_batch.py): weighted mean == the
+        # unweighted loss over the real graphs, exactly.
+        w_g = data.get("loss_weight_g")
+        if w_g is not None:
+            w_g = w_g.reshape(-1).float()
+            loss_fn = _make_weighted_loss(self.loss_function_type, w_g)
+        else:
+            loss_fn = self.loss_function
         tasks_loss = [loss_fn(graph_energy_pred, graph_energy_true)]
 
         if (self.energy_weight <= 0 and self.energy_peratom_weight <= 0
@@ -352,7 +386,12 @@ class EnhancedModelWrapper(torch.nn.Module):
         )[0]
         assert forces_pred is not None
         forces_pred = -forces_pred.float()
-        f_loss = loss_fn(forces_pred, forces_true)
+        if w_g is not None:
+            force_loss_fn = _make_weighted_loss(
+                self.loss_function_type, w_g[data.batch])
+        else:
+            force_loss_fn = loss_fn
+        f_loss = force_loss_fn(forces_pred, forces_true)
         tasks_loss.append(f_loss)
         if self.force_weight > 0:
             tot_loss = tot_loss + f_loss * self.force_weight

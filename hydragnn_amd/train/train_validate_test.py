@@ -222,39 +222,61 @@ def train(loader, model, opt, verbosity, precision="fp32", profiler=None):
         # FSDP2 double-backward workaround (see set_reshard_after_backward)
         set_reshard_after_backward(model, False)
 
+    from .captured import capture_enabled, get_or_build_stepper
+    stepper = None
+
     for ibatch, data in enumerate(iterate_tqdm(loader, verbosity)):
         if ibatch >= nbatch:
             break
         from ..globalatt.gps import redraw_performer_projections
         redraw_performer_projections(
             model, int(os.environ.get("HYDRAGNN_ATTN_REDRAW", "1000")))
-        tr.start("h2d")
-        data = move_batch_to_device(data, param_dtype)
-        tr.stop("h2d")
-        opt.zero_grad(set_to_none=True)
-        tr.start("forward")
-        with autocast:
-            loss, tasks_loss, _ = _compute_loss(model, data, use_ia,
-                                                create_graph=True)
-        tr.stop("forward")
-        tr.start("backward")
-        if scaler is not None:
-            scaler.scale(loss).backward()
+
+        # hipGraph-captured fast path (static-shape batches on GPU):
+        # H2D copy-in + one graph replay per batch (captured.py).
+        if (scaler is None and stepper is None
+                and capture_enabled(data)):
+            dev_data = move_batch_to_device(data, param_dtype)
+            stepper = get_or_build_stepper(
+                model, opt, dev_data, autocast,
+                lambda b: _compute_loss(model, b, use_ia,
+                                        create_graph=True),
+                param_dtype)
+            if stepper is not None:
+                data = dev_data
+        if stepper is not None and stepper.matches(data):
+            tr.start("captured_step")
+            loss, tasks_loss = stepper.step(data)
+            tr.stop("captured_step")
         else:
-            loss.backward()
-        tr.stop("backward")
-        tr.start("opt_step")
-        if scaler is not None:
-            scaler.step(opt)
-            scaler.update()
-        else:
-            opt.step()
-        tr.stop("opt_step")
+            tr.start("h2d")
+            data = move_batch_to_device(data, param_dtype)
+            tr.stop("h2d")
+            opt.zero_grad(set_to_none=True)
+            tr.start("forward")
+            with autocast:
+                loss, tasks_loss, _ = _compute_loss(model, data, use_ia,
+                                                    create_graph=True)
+            tr.stop("forward")
+            tr.start("backward")
+            if scaler is not None:
+                scaler.scale(loss).backward()
+            else:
+                loss.backward()
+            tr.stop("backward")
+            tr.start("opt_step")
+            if scaler is not None:
+                scaler.step(opt)
+                scaler.update()
+            else:
+                opt.step()
+            tr.stop("opt_step")
         if use_ia and is_fsdp2_enabled():
             set_reshard_after_backward(model, True)
             set_reshard_after_backward(model, False)
         profiler.step()
-        n = data.num_graphs
+        n = int(data.get("num_real_graphs_", 0) or data.num_graphs) \
+            if hasattr(data, "get") else data.num_graphs
         total_error += loss.detach() * n
         for it, tl in enumerate(tasks_loss):
             tasks_error[it] += tl.detach() * n

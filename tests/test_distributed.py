@@ -228,19 +228,39 @@ def _mace_force_ddp_worker(rank, world_size, port, q):
         q.put((rank, False, repr(e)))
 
 
-def test_mace_force_training_ddp_gloo():
+@pytest.mark.parametrize("world_size,port", [(2, 29531), (4, 29551),
+                                             (8, 29561)])
+def test_mace_force_training_ddp_gloo(world_size, port):
     """MACE att-interaction force training (double backward) under DDP
-    on 2 gloo ranks: ranks end bitwise-synchronized."""
+    on 2/4/8 gloo ranks (8 = the driver's full-node scaling shape):
+    ranks end bitwise-synchronized."""
     import multiprocessing as mp
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
-    port = 29531
-    ps = [ctx.Process(target=_mace_force_ddp_worker, args=(r, 2, port, q))
-          for r in range(2)]
+    ps = [ctx.Process(target=_mace_force_ddp_worker,
+                      args=(r, world_size, port, q))
+          for r in range(world_size)]
     for p in ps:
         p.start()
-    results = [q.get(timeout=300) for _ in range(2)]
+    results = [q.get(timeout=600) for _ in range(world_size)]
     for p in ps:
         p.join(timeout=60)
     for rank, same, info in results:
         assert same, f"rank {rank}: {info}"
+
+
+@pytest.mark.parametrize("world_size,port", [(4, 29571)])
+def test_fsdp2_force_grad_regression_multirank(world_size, port):
+    """FSDP2 double-backward force path on 4 gloo ranks."""
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_fsdp2_force_worker,
+                         args=(r, world_size, port, q))
+             for r in range(world_size)]
+    for p in procs:
+        p.start()
+    results = [q.get(timeout=600) for _ in range(world_size)]
+    for p in procs:
+        p.join(timeout=60)
+    for rank, ok, info in results:
+        assert ok, f"rank {rank}: {info}"

@@ -1,0 +1,188 @@
+"""Static-shape padding, weighted loss, fp32 grad sync, and the
+train-loop bench path (CPU; the hipGraph capture itself is covered by
+tests/test_gpu_capture.py)."""
+
+import json
+import os
+import subprocess
+import sys
+
+import pytest
+import torch
+import torch.distributed as dist
+
+from hydragnn_amd.data import Batch
+from hydragnn_amd.preprocess.static_batch import (
+    StaticShapeCollater, compute_static_caps, pad_batch_static)
+from hydragnn_amd.utils.datasets.synthetic import (
+    md17_shape_dataset, md17_shape_dataset_fast)
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+
+def _small_mace(dataset):
+    from test_mace_model import _build, _mace_config
+    model, config, _ = _build(_mace_config(), dataset)
+    return model
+
+
+def _loss_and_grads(model, batch):
+    model.zero_grad()
+    batch.pos.requires_grad_(True)
+    pred = model(batch)
+    loss, tasks = model.energy_force_loss(pred, batch, create_graph=True)
+    loss.backward()
+    g = torch.cat([p.grad.flatten() for p in model.parameters()
+                   if p.grad is not None])
+    return loss, tasks, g
+
+
+def test_pad_batch_loss_grad_equivalence():
+    """Padded batch (sentinel pad graph, loss_weight_g=0) gives the
+    same loss and parameter gradients as the unpadded batch."""
+    torch.manual_seed(3)
+    ds = md17_shape_dataset(num_samples=4)
+    model = _small_mace(ds)
+    b1 = Batch.from_data_list([d.clone() for d in ds])
+    b2 = Batch.from_data_list([d.clone() for d in ds])
+    nc, ec = compute_static_caps(ds, 4)
+    b2 = pad_batch_static(b2, nc + 5, ec + 17, pad_spacing=30.0)
+
+    l1, t1, g1 = _loss_and_grads(model, b1)
+    l2, t2, g2 = _loss_and_grads(model, b2)
+    assert torch.isfinite(l2)
+    assert abs(float(l1) - float(l2)) < 1e-5 * max(1.0, abs(float(l1)))
+    for a, b in zip(t1, t2):
+        assert abs(float(a) - float(b)) < 1e-4 * max(1.0, abs(float(a)))
+    assert (g1 - g2).abs().max() < 1e-5 * max(1.0, g1.abs().max())
+
+
+def test_static_collater_fixed_shapes():
+    ds = md17_shape_dataset_fast(32, seed=9)
+    nc, ec = compute_static_caps(ds, 8)
+    coll = StaticShapeCollater(nc, ec, pad_spacing=28.0)
+    shapes = None
+    for i in range(0, 32, 8):
+        b = coll(ds[i:i + 8])
+        s = {k: tuple(v.shape) for k, v in b.items()
+             if torch.is_tensor(v)}
+        if shapes is None:
+            shapes = s
+        assert s == shapes
+        assert b.num_nodes == nc and b.num_edges == ec
+        assert bool(b.get("static_shape_"))
+        assert int(b.get("num_real_graphs_")) == 8
+        w = b["loss_weight_g"]
+        assert w.shape == (9,) and float(w[-1]) == 0.0 \
+            and float(w[:-1].sum()) == 8.0
+        dst = b.edge_index[1]
+        assert (dst[1:] >= dst[:-1]).all(), "padding broke dst sort"
+        # pad edges never touch real nodes
+        n_real = sum(d.num_nodes for d in ds[i:i + 8])
+        pad_e = b.edge_index[:, -(ec - sum(d.num_edges
+                                           for d in ds[i:i + 8])):]
+        if pad_e.numel():
+            assert int(pad_e.min()) >= n_real
+
+
+def test_weighted_loss_matches_unweighted():
+    from hydragnn_amd.models.create import _make_weighted_loss
+    torch.manual_seed(0)
+    pred = torch.randn(12, 3)
+    true = torch.randn(12, 3)
+    w = torch.ones(12)
+    fn = _make_weighted_loss("mse", w)
+    ref = torch.nn.functional.mse_loss(pred, true)
+    assert torch.allclose(fn(pred, true), ref, atol=1e-6)
+    fn = _make_weighted_loss("mae", w)
+    ref = torch.nn.functional.l1_loss(pred, true)
+    assert torch.allclose(fn(pred, true), ref, atol=1e-6)
+    # zero-weight rows drop out
+    w2 = torch.cat([torch.ones(8), torch.zeros(4)])
+    fn = _make_weighted_loss("mse", w2)
+    ref = torch.nn.functional.mse_loss(pred[:8], true[:8])
+    assert torch.allclose(fn(pred, true), ref, atol=1e-6)
+
+
+def test_train_loop_with_padded_batches():
+    """train() (eager path on CPU) over a static-collated loader:
+    finite losses, sample counting excludes the pad graph."""
+    from torch.utils.data import DataLoader
+
+    from hydragnn_amd.train import train
+
+    torch.manual_seed(5)
+    ds = md17_shape_dataset(num_samples=8)
+    model = _small_mace(ds)
+    nc, ec = compute_static_caps(ds, 4)
+    loader = DataLoader(ds, batch_size=4,
+                        collate_fn=StaticShapeCollater(nc, ec),
+                        shuffle=False)
+    opt = torch.optim.AdamW(model.parameters(), lr=1e-3)
+    err, tasks = train(loader, model, opt, 0)
+    assert torch.isfinite(err).all()
+    assert torch.isfinite(tasks).all()
+
+
+def _grad_sync_worker(rank, world, port, q):
+    try:
+        os.environ.update(MASTER_ADDR="127.0.0.1",
+                          MASTER_PORT=str(port), RANK=str(rank),
+                          WORLD_SIZE=str(world), LOCAL_RANK=str(rank))
+        dist.init_process_group("gloo", rank=rank, world_size=world)
+        from hydragnn_amd.parallel import FlatGradSync
+        torch.manual_seed(100 + rank)
+        m = torch.nn.Linear(8, 8).to(torch.bfloat16)
+        x = torch.randn(4, 8).to(torch.bfloat16)
+        m(x).float().pow(2).sum().backward()
+        local = [p.grad.clone() for p in m.parameters()]
+        sync = FlatGradSync(m.parameters())
+        sync()
+        # expected: fp32 mean across ranks, cast back to bf16
+        for p, lg in zip(m.parameters(), local):
+            gathered = [torch.zeros_like(lg, dtype=torch.float32)
+                        for _ in range(world)]
+            dist.all_gather(gathered, lg.float())
+            want = (torch.stack(gathered).sum(0) / world).to(
+                torch.bfloat16)
+            assert torch.equal(p.grad, want), "fp32 mean mismatch"
+        q.put((rank, True, ""))
+        dist.destroy_process_group()
+    except Exception as e:  # pragma: no cover
+        q.put((rank, False, repr(e)))
+
+
+def test_flat_grad_sync_gloo():
+    import multiprocessing as mp
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    ps = [ctx.Process(target=_grad_sync_worker, args=(r, 2, 29661, q))
+          for r in range(2)]
+    for p in ps:
+        p.start()
+    results = [q.get(timeout=300) for _ in range(2)]
+    for p in ps:
+        p.join(timeout=60)
+    for rank, ok, info in results:
+        assert ok, f"rank {rank}: {info}"
+
+
+def test_bench_trainloop_cpu():
+    """bench.py default path runs through train() on CPU and prints
+    one valid JSON line."""
+    env = dict(os.environ, HYDRAGNN_BENCH_PRECISION="fp32",
+               HYDRAGNN_BENCH_WORKERS="0", MASTER_ADDR="127.0.0.1")
+    r = subprocess.run(
+        [sys.executable, os.path.join(REPO, "bench.py"), "--steps", "2",
+         "--warmup", "1", "--batch", "4"],
+        capture_output=True, text=True, timeout=420, cwd=REPO, env=env)
+    assert r.returncode == 0, r.stderr[-2000:]
+    line = [ln for ln in r.stdout.splitlines() if ln.startswith("{")][-1]
+    out = json.loads(line)
+    assert out["value"] > 0 and out["steps"] == 2
+    assert out["n_gpus"] == 1
+
+
+# (2-rank torchrun coverage of the same path lives in
+# tests/test_distributed.py::test_bench_two_rank_gloo.)
