@@ -251,7 +251,8 @@ def make_loaders(rank, steps, warmup, batch, use_cuda,
     total = (steps + warmup) * batch
     ds = md17_shape_dataset_fast(
         total, radius=MODEL_CONFIG["radius"], seed=seed + rank * 1000003)
-    node_cap, edge_cap = compute_static_caps(ds, batch)
+    node_cap, edge_cap = compute_static_caps(ds, batch,
+                                             sequential=True)
     coll = StaticShapeCollater(node_cap, edge_cap,
                                pad_spacing=4 * MODEL_CONFIG["radius"])
     nw = int(os.environ.get(
