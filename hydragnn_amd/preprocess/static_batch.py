@@ -166,7 +166,7 @@ def compute_static_caps(dataset, batch_size: int,
                         node_margin: int = 2,
                         edge_margin: int = 0,
                         sequential: bool = False,
-                        max_pad_edges_per_node: int = 16):
+                        max_pad_edges_per_node: int = 64):
     """(node, edge) capacities for batches of ``batch_size`` samples.
 
     ``sequential=True``: exact max over consecutive batches (loaders
