@@ -48,9 +48,13 @@ def _run_epochs(capture: bool, steps=6, batch=16, precision="bf16"):
             stepper = getattr(base, "_hip_captured_step", None)
             assert stepper not in (None, False), \
                 "captured path did not engage"
+            base._hip_captured_step = None
         return losses, params
     finally:
         os.environ.pop("HYDRAGNN_CAPTURE", None)
+        import gc
+        gc.collect()
+        torch.cuda.empty_cache()
 
 
 @needs_gpu
@@ -90,3 +94,8 @@ def test_captured_second_epoch_reuses_graph():
     train(loader, model, opt, 0, precision="bf16")
     s2 = getattr(model, "_hip_captured_step", None)
     assert s2 is s1, "graph was re-captured"
+    model._hip_captured_step = None
+    del s1, s2
+    import gc
+    gc.collect()
+    torch.cuda.empty_cache()
