@@ -144,12 +144,31 @@ def get_or_build_stepper(model, opt, device_batch, autocast,
             return None
         if cached.opt is opt and cached.matches(device_batch):
             return cached
+    stepper = None
+    err = None
     try:
         stepper = CapturedTrainStep(model, opt, device_batch, autocast,
                                     compute_loss, param_dtype)
     except Exception as e:  # pragma: no cover - GPU-only path
+        err = e
+
+    # All ranks must agree: a mix of captured (explicit FlatGradSync
+    # all-reduce) and eager (DDP-hook all-reduce) ranks would mismatch
+    # collectives and hang. Fall back everywhere unless all captured.
+    import torch.distributed as dist
+    if dist.is_initialized() and dist.get_world_size() > 1:
+        ok = torch.tensor([0 if stepper is None else 1],
+                          device=device_batch.pos.device
+                          if torch.is_tensor(device_batch.get("pos"))
+                          else None)
+        dist.all_reduce(ok, op=dist.ReduceOp.MIN)
+        if int(ok.item()) == 0:
+            stepper = None
+
+    if stepper is None:  # pragma: no cover - GPU-only path
         import sys
-        print(f"[captured] hipGraph capture failed ({e}); "
+        print(f"[captured] hipGraph capture disabled "
+              f"({err if err is not None else 'peer rank failed'}); "
               "eager fallback", file=sys.stderr)
         base._hip_captured_step = False
         return None
