@@ -25,7 +25,12 @@ from typing import Dict, List, Optional
 
 import torch
 from torch import nn
-from torch.nn import Linear, Module, ModuleDict, ModuleList, Sequential
+from torch.nn import Module, ModuleDict, ModuleList, Sequential
+
+# Drop-in nn.Linear that dispatches to the MFMA / narrow-output GEMV
+# HIP kernels when shapes qualify (falls back to split-K / F.linear):
+# keeps decoder-head GEMMs off hipBLASLt's degenerate N=1 tiles.
+from ..ops.mfma_linear import MFMALinear as Linear
 from torch.utils.checkpoint import checkpoint
 
 from ..ops import scatter

@@ -706,6 +706,9 @@ torch::Tensor irreps_linear(torch::Tensor X, torch::Tensor W,
 torch::Tensor irreps_linear_gw(torch::Tensor X, torch::Tensor G,
                                torch::Tensor lmap, long L,
                                long nblocks);
+// defined in gemv.hip
+torch::Tensor gemv_small_n(torch::Tensor A, torch::Tensor W,
+                           c10::optional<torch::Tensor> bias);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("etp_general", &etp_general, "fused ETP contraction (HIP)",
@@ -751,6 +754,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         pybind11::arg("bias") = pybind11::none(),
         pybind11::arg("trans_w") = false,
         pybind11::arg("add") = pybind11::none());
+  m.def("gemv_small_n", &gemv_small_n,
+        "narrow-output bf16 linear (HIP)",
+        pybind11::arg("A"), pybind11::arg("W"),
+        pybind11::arg("bias") = pybind11::none());
   m.def("irreps_linear_gw", &irreps_linear_gw,
         "irreps-linear weight-grad partials (HIP)",
         pybind11::arg("X"), pybind11::arg("G"), pybind11::arg("lmap"),

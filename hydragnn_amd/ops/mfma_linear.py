@@ -83,18 +83,56 @@ class _MFMALinearFn(torch.autograd.Function):
         return gx, gw, gb
 
 
+class _GemvLinearFn(torch.autograd.Function):
+    """Linear with out_features <= 8: forward on the gemv_small_n
+    kernel (hipBLASLt runs N=1 on MT1x4x256 tiles ~30x off roofline);
+    backward composed of differentiable ops so the force-training
+    double backward works unchanged."""
+
+    @staticmethod
+    def forward(ctx, x, weight, bias):
+        xb = x.to(torch.bfloat16).contiguous()
+        wb = weight.to(torch.bfloat16).contiguous()
+        ctx.save_for_backward(xb, wb)
+        ctx.has_bias = bias is not None
+        ctx.w_dtype = weight.dtype
+        ctx.x_dtype = x.dtype
+        ext = get_extension(required=True)
+        return ext.gemv_small_n(xb, wb,
+                                bias if bias is not None else None)
+
+    @staticmethod
+    def backward(ctx, g):
+        xb, wb = ctx.saved_tensors
+        g = g.contiguous().to(torch.bfloat16)
+        gx = gw = gb = None
+        if ctx.needs_input_grad[0]:
+            if wb.shape[0] == 1:
+                gx = (g * wb.reshape(-1)).to(ctx.x_dtype)
+            else:
+                gx = (g @ wb).to(ctx.x_dtype)
+        if ctx.needs_input_grad[1]:
+            gw = _splitk_weight_grad(xb, g, 64).to(ctx.w_dtype)
+        if ctx.has_bias and ctx.needs_input_grad[2]:
+            gb = g.float().sum(0)
+        return gx, gw, gb
+
+
 class MFMALinear(SplitKLinear):
     """nn.Linear drop-in: MFMA kernel when shapes/dtypes qualify,
-    split-K Linear otherwise."""
+    narrow-output GEMV kernel for out_features <= 8, split-K Linear
+    otherwise."""
 
     def forward(self, x):
-        use_mfma = (
+        hip_ok = (
             x.is_cuda and not use_eager()
             and (x.dtype == torch.bfloat16
                  or torch.is_autocast_enabled())
-            and x.dim() == 2
-            and _eligible(x.shape[0], self.out_features,
-                          self.in_features))
-        if use_mfma:
+            and x.dim() == 2)
+        if hip_ok and _eligible(x.shape[0], self.out_features,
+                                self.in_features):
             return _MFMALinearFn.apply(x, self.weight, self.bias)
+        if (hip_ok and self.out_features <= 8
+                and self.in_features % 64 == 0 and x.shape[0] >= 256):
+            return _GemvLinearFn.apply(x, self.weight, self.bias)
         return super().forward(x)

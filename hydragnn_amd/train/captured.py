@@ -69,25 +69,35 @@ class CapturedTrainStep:
 
         no_sync = getattr(model, "no_sync", nullcontext)
 
+        # Pre-assign every grad as a view into ONE flat buffer: the
+        # per-replay zeroing is a single fill kernel instead of ~100,
+        # and the DP sync / a fused optimizer can address the whole
+        # gradient contiguously.
+        params = [p for p in model.parameters() if p.requires_grad]
+        total = sum(p.numel() for p in params)
+        dev = next(iter(params)).device if params else "cuda"
+        self._flat_grad = torch.zeros(total, dtype=param_dtype,
+                                      device=dev)
+        off = 0
+        for p in params:
+            p.grad = self._flat_grad[off:off + p.numel()].view_as(p)
+            off += p.numel()
+
         def fwd_bwd():
-            if self._grads is not None:
-                torch._foreach_zero_(self._grads)
+            self._flat_grad.zero_()
             self.static.pos.requires_grad_(True)
             with autocast:
                 loss, tasks_loss, _ = compute_loss(self.static)
             loss.backward()
             return loss, tasks_loss
 
-        self._grads = None
         side = torch.cuda.Stream()
         side.wait_stream(torch.cuda.current_stream())
         with torch.cuda.stream(side), no_sync():
             for _ in range(warmup_iters):
                 loss, _ = fwd_bwd()
-                if self._grads is None:
-                    self._grads = [p.grad for p in model.parameters()
-                                   if p.grad is not None]
         torch.cuda.current_stream().wait_stream(side)
+        self._grads = [p.grad for p in params if p.grad is not None]
 
         self.graph = torch.cuda.CUDAGraph()
         with no_sync():

@@ -13,6 +13,7 @@ ext_modules = [
                  "hydragnn_amd/ops/csrc/etp.hip",
                  "hydragnn_amd/ops/csrc/mfma_linear.hip",
                  "hydragnn_amd/ops/csrc/varlen_attn.hip",
+                 "hydragnn_amd/ops/csrc/gemv.hip",
                  "hydragnn_amd/ops/csrc/irreps_linear.hip"],
         extra_compile_args={
             "cxx": ["-O3"],
