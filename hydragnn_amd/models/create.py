@@ -334,7 +334,7 @@ class EnhancedModelWrapper(torch.nn.Module):
             node_energy_pred = pred[0]
             graph_energy_pred = scatter(
                 node_energy_pred, data.batch, n_graphs,
-                "sum").squeeze(-1).float()
+                "sum", sorted_index=True).squeeze(-1).float()
         elif self.head_type[0] == "graph":
             if getattr(self.model, "graph_pooling", "mean") not in ("add",):
                 raise ValueError(
@@ -369,7 +369,8 @@ class EnhancedModelWrapper(torch.nn.Module):
 
         natoms = scatter(torch.ones_like(data.batch,
                                          dtype=graph_energy_pred.dtype),
-                         data.batch, n_graphs, "sum")
+                         data.batch, n_graphs, "sum",
+                         sorted_index=True)
         e_pa_pred = graph_energy_pred / natoms
         e_pa_true = graph_energy_true / natoms
         pa_loss = loss_fn(e_pa_pred, e_pa_true)

@@ -81,7 +81,7 @@ def _rowptr_from_sorted(index: torch.Tensor, dim_size: int):
 class _ScatterSum(torch.autograd.Function):
     @staticmethod
     def forward(ctx, src, index, dim_size, sorted_index=False, csr=None):
-        ctx.save_for_backward(index)
+        ctx.csr = None
         if _use_hip(src):
             import os
             ext = get_extension(required=True)
@@ -90,20 +90,31 @@ class _ScatterSum(torch.autograd.Function):
                 # precomputed index-sort: out[r] = sum src[perm[e]] —
                 # contention-free and deterministic for UNSORTED index
                 perm, rowptr = csr
+                ctx.csr = csr
+                ctx.save_for_backward(index)
                 return ext.segment_sum_csr(src.contiguous(), rowptr,
                                            perm)
             if sorted_index and hasattr(ext, "segment_sum_csr") and \
                     os.environ.get("HYDRAGNN_CSR_SCATTER", "1") == "1":
                 rowptr = _rowptr_from_sorted(index, dim_size)
+                # pass the rowptr down the chain: the double-backward
+                # scatter (gather's backward) reuses it instead of
+                # falling back to atomics
+                ctx.csr = (None, rowptr)
+                ctx.save_for_backward(index)
                 return ext.segment_sum_csr(src.contiguous(), rowptr)
             if os.environ.get("HYDRAGNN_DETERMINISTIC", "0") == "1":
                 # order-independent accumulation: sort once, CSR reduce
                 # (atomicAdd float accumulation is order-dependent)
                 perm = torch.argsort(index, stable=True)
                 rowptr = _rowptr_from_sorted(index[perm], dim_size)
+                ctx.csr = (perm, rowptr)
+                ctx.save_for_backward(index)
                 return ext.segment_sum_csr(
                     src.index_select(0, perm).contiguous(), rowptr)
+            ctx.save_for_backward(index)
             return ext.scatter_sum_fwd(src.contiguous(), index, dim_size)
+        ctx.save_for_backward(index)
         out = src.new_zeros((dim_size,) + src.shape[1:])
         out.index_add_(0, index, src)
         return out
@@ -111,7 +122,8 @@ class _ScatterSum(torch.autograd.Function):
     @staticmethod
     def backward(ctx, grad_out):
         (index,) = ctx.saved_tensors
-        return gather(grad_out, index), None, None, None, None
+        return (gather(grad_out, index, backward_csr=ctx.csr),
+                None, None, None, None)
 
 
 class _ScatterMean(torch.autograd.Function):
