@@ -29,6 +29,7 @@ from ...ops import (
     polynomial_cutoff,
     scatter,
 )
+from ...ops.mfma_linear import MFMALinear
 from .o3 import IrrepsLinear, dim, tp_paths, wigner_3j
 from .symmetric_contraction import SymmetricContraction
 
@@ -258,8 +259,8 @@ class RealAgnosticAttResidualInteractionBlock(
 
     def _setup_extra(self, num_channels: int):
         self._radial_extra = 2 * num_channels
-        self.linear_down = nn.Linear(num_channels, num_channels,
-                                     bias=False)
+        self.linear_down = MFMALinear(num_channels, num_channels,
+                                      bias=False)
 
     def _edge_weights(self, node_feats, src, dst, edge_radial, etp_meta):
         down = self.linear_down(node_feats[:, :, 0])  # [N, C] scalars
@@ -306,7 +307,7 @@ class LinearReadoutBlock(nn.Module):
 
     def __init__(self, num_channels: int, out_dim: int):
         super().__init__()
-        self.linear = nn.Linear(num_channels, out_dim)
+        self.linear = MFMALinear(num_channels, out_dim)
 
     def forward(self, node_feats: torch.Tensor) -> torch.Tensor:
         return self.linear(node_feats[:, :, 0])
@@ -316,9 +317,9 @@ class NonLinearReadoutBlock(nn.Module):
     def __init__(self, num_channels: int, hidden: int, out_dim: int,
                  act=None):
         super().__init__()
-        self.linear_1 = nn.Linear(num_channels, hidden)
+        self.linear_1 = MFMALinear(num_channels, hidden)
         self.act = act or nn.SiLU()
-        self.linear_2 = nn.Linear(hidden, out_dim)
+        self.linear_2 = MFMALinear(hidden, out_dim)
 
     def forward(self, node_feats: torch.Tensor) -> torch.Tensor:
         return self.linear_2(self.act(self.linear_1(node_feats[:, :, 0])))

@@ -80,7 +80,7 @@ torch::Tensor gemv_small_n(torch::Tensor A, torch::Tensor W,
   int N = W.size(0);
   TORCH_CHECK(W.size(1) == K, "inner dims mismatch");
   TORCH_CHECK(N >= 1 && N <= MAX_N, "gemv_small_n needs 1 <= N <= 8");
-  TORCH_CHECK(K % 64 == 0, "gemv_small_n needs K % 64 == 0");
+  TORCH_CHECK(K % 8 == 0, "gemv_small_n needs K % 8 == 0");
   auto C = torch::empty({M, (long)N}, A.options());
   if (M == 0) return C;
   const float* bias_ptr = nullptr;

@@ -133,6 +133,6 @@ class MFMALinear(SplitKLinear):
                                 self.in_features):
             return _MFMALinearFn.apply(x, self.weight, self.bias)
         if (hip_ok and self.out_features <= 8
-                and self.in_features % 64 == 0 and x.shape[0] >= 256):
+                and self.in_features % 8 == 0 and x.shape[0] >= 256):
             return _GemvLinearFn.apply(x, self.weight, self.bias)
         return super().forward(x)
