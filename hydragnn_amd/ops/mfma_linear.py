@@ -112,7 +112,15 @@ class _GemvLinearFn(torch.autograd.Function):
             else:
                 gx = (g @ wb).to(ctx.x_dtype)
         if ctx.needs_input_grad[1]:
-            gw = _splitk_weight_grad(xb, g, 64).to(ctx.w_dtype)
+            if wb.shape[0] == 1:
+                # dW[0,k] = sum_m g[m,0] x[m,k]: a weighted column sum
+                # (the split-K bmm here is a degenerate M=1 batched
+                # GEMM that hipBLASLt runs on MT1x4 tiles ~50x off
+                # roofline)
+                gw = (xb.float() * g.float()).sum(0, keepdim=True) \
+                    .to(ctx.w_dtype)
+            else:
+                gw = _splitk_weight_grad(xb, g, 64).to(ctx.w_dtype)
         if ctx.has_bias and ctx.needs_input_grad[2]:
             gb = g.float().sum(0)
         return gx, gw, gb

@@ -154,6 +154,8 @@ def md17_shape_dataset_fast(
     species: Sequence[int] = _MD17_SPECIES,
     spread: float = 2.5,
     chunk: int = 2048,
+    min_dist: float = 0.7,
+    max_push: int = 3,
 ):
     """Vectorized md17_shape_dataset: generates molecules in batched
     chunks (one cdist/LJ evaluation per chunk instead of per molecule)
@@ -170,10 +172,10 @@ def md17_shape_dataset_fast(
     for start in range(0, num_samples, chunk):
         m = min(chunk, num_samples - start)
         pos = (torch.rand(m, n, 3, generator=g) - 0.5) * 2 * spread
-        for _ in range(3):
+        for _ in range(max_push):
             d = torch.cdist(pos, pos) + eye * 10
             mind = d.flatten(1).min(dim=1).values
-            scale = torch.where(mind > 0.7,
+            scale = torch.where(mind > min_dist,
                                 torch.ones(m), torch.full((m,), 1.25))
             if (scale == 1.0).all():
                 break

@@ -1,0 +1,87 @@
+"""fp64 and fp16 paths on the GPU: MACE fp64 force parity vs CPU
+(ETP kernels dispatch double), and fp16 GradScaler training
+(VERDICT r1 items 6/7)."""
+
+import os
+import sys
+
+import pytest
+import torch
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+pytestmark = pytest.mark.gpu
+
+if not torch.cuda.is_available():
+    pytest.skip("requires MI355X GPU", allow_module_level=True)
+
+
+def _mace_model_and_batch(dtype, n_mols=8):
+    from hydragnn_amd.data import Batch
+    from hydragnn_amd.utils.datasets.synthetic import md17_shape_dataset
+    import bench as B
+
+    torch.manual_seed(2)
+    model = B.build_model("cpu", precision="fp32", seed=2).to(dtype)
+    ds = md17_shape_dataset(num_samples=n_mols, seed=6)
+    batch = Batch.from_data_list([d.clone() for d in ds])
+    for k in list(batch.keys()):
+        v = batch[k]
+        if torch.is_tensor(v) and torch.is_floating_point(v):
+            batch[k] = v.to(dtype)
+    return model, batch
+
+
+def _energy_forces(model, batch):
+    batch.pos.requires_grad_(True)
+    pred = model(batch)
+    from hydragnn_amd.ops import scatter
+    e = scatter(pred[0].double(), batch.batch, batch.num_graphs,
+                "sum").squeeze(-1)
+    f = -torch.autograd.grad(e.sum(), batch.pos)[0].double()
+    return e.detach(), f.detach()
+
+
+def test_mace_fp64_gpu_matches_cpu():
+    """fp64 MACE on GPU (kernels dispatch double) vs the same model on
+    CPU: tight agreement on energies and forces."""
+    model, batch = _mace_model_and_batch(torch.float64)
+    e_cpu, f_cpu = _energy_forces(model, batch.clone())
+    model_g = model.to("cuda")
+    batch_g = batch.clone().to("cuda")
+    e_gpu, f_gpu = _energy_forces(model_g, batch_g)
+    de = (e_cpu - e_gpu.cpu()).abs().max() / e_cpu.abs().max().clamp(min=1)
+    df = (f_cpu - f_gpu.cpu()).abs().max() / f_cpu.abs().max().clamp(min=1)
+    assert de < 1e-8, f"fp64 energy rel diff {de:.2e}"
+    assert df < 1e-6, f"fp64 force rel diff {df:.2e}"
+
+
+def test_fp16_scaler_training_gpu():
+    """fp16 + GradScaler: a few training steps stay finite and the
+    loss decreases (the reference's fp16 path,
+    /root/reference/hydragnn/train/train_validate_test.py:87-103)."""
+    from torch.utils.data import DataLoader
+
+    import bench as B
+    from hydragnn_amd.preprocess.load_data import _collate
+    from hydragnn_amd.train import train
+    from hydragnn_amd.train.train_validate_test import (
+        get_autocast_and_scaler)
+
+    autocast, scaler = get_autocast_and_scaler("fp16")
+    assert scaler is not None, "fp16 must return a GradScaler"
+
+    from hydragnn_amd.utils.datasets.synthetic import (
+        md17_shape_dataset_fast)
+    torch.manual_seed(3)
+    model = B.build_model("cuda:0", precision="fp16", seed=3)
+    opt = torch.optim.AdamW(model.parameters(), lr=1e-3)
+    ds = md17_shape_dataset_fast(64, seed=8)
+    loader = DataLoader(ds, batch_size=16, shuffle=False,
+                        collate_fn=_collate)
+    errs = []
+    for _ in range(4):
+        err, _ = train(loader, model, opt, 0, precision="fp16")
+        errs.append(float(err))
+    assert all(e == e and e < 1e30 for e in errs), errs
+    assert errs[-1] < errs[0], f"fp16 loss did not decrease: {errs}"
