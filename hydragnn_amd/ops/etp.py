@@ -103,7 +103,10 @@ def _kernel_ok(table: ETPTable, *tensors) -> bool:
         table.entries.shape[0] * 20
     if lds_bytes > 150 * 1024 or max(da, db, dg, do) > 192:
         return False
-    return (t.is_cuda and t.dtype in _KERNEL_DTYPES and not use_eager())
+    return (t.is_cuda and t.dtype in _KERNEL_DTYPES
+            and all(x.dtype == t.dtype for x in tensors
+                    if torch.is_tensor(x))
+            and not use_eager())
 
 
 class _ETPGeneral(torch.autograd.Function):

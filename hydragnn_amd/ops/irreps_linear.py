@@ -100,7 +100,8 @@ def irreps_linear_eligible(x: torch.Tensor, W: torch.Tensor) -> bool:
         return False
     if os.environ.get("HYDRAGNN_IRREPS_MFMA", "1") == "0":
         return False
-    if not (x.dtype == torch.bfloat16 or torch.is_autocast_enabled()):
+    from .mfma_linear import _bf16_ok
+    if not _bf16_ok(x):
         return False
     return irreps_kernel_ok(x.shape[0], x.shape[1], W.shape[2],
                             x.shape[2], W.shape[0])

@@ -15,6 +15,22 @@ from ._extension import get_extension, use_eager
 from .splitk_linear import SplitKLinear, _splitk_weight_grad
 
 
+
+
+def _bf16_ok(x):
+    """bf16 kernel eligibility: tensor already bf16, or autocast is on
+    WITH bf16 as the autocast dtype (an fp16 session must not be
+    silently rerouted through bf16 kernels)."""
+    if x.dtype == torch.bfloat16:
+        return True
+    if not torch.is_autocast_enabled():
+        return False
+    try:
+        return torch.get_autocast_dtype("cuda") == torch.bfloat16
+    except (AttributeError, TypeError):
+        return torch.get_autocast_gpu_dtype() == torch.bfloat16
+
+
 def _eligible(M: int, N: int, K: int) -> bool:
     # forward C[M,N] = A[M,K] @ B^T needs K%32, N%64; the gA backward
     # flips roles (K'=N, N'=K) so both must satisfy both constraints
@@ -134,8 +150,7 @@ class MFMALinear(SplitKLinear):
     def forward(self, x):
         hip_ok = (
             x.is_cuda and not use_eager()
-            and (x.dtype == torch.bfloat16
-                 or torch.is_autocast_enabled())
+            and _bf16_ok(x)
             and x.dim() == 2)
         if hip_ok and _eligible(x.shape[0], self.out_features,
                                 self.in_features):
