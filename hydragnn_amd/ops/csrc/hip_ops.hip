@@ -709,6 +709,12 @@ torch::Tensor irreps_linear_gw(torch::Tensor X, torch::Tensor G,
 // defined in gemv.hip
 torch::Tensor gemv_small_n(torch::Tensor A, torch::Tensor W,
                            c10::optional<torch::Tensor> bias);
+// defined in radius.hip
+std::vector<torch::Tensor> radius_pairs_t(torch::Tensor pos,
+                                          torch::Tensor batch,
+                                          torch::Tensor gptr, double r,
+                                          bool loop,
+                                          c10::optional<torch::Tensor> shifts);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("etp_general", &etp_general, "fused ETP contraction (HIP)",
@@ -746,6 +752,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("scatter_mean_fwd", &scatter_mean_fwd, "scatter-mean (HIP)");
   m.def("scatter_minmax_fwd", &scatter_minmax_fwd, "scatter-min/max (HIP)");
   m.def("radius_pairs", &radius_pairs, "radius pair enumeration (HIP)");
+  m.def("radius_pairs_t", &radius_pairs_t,
+        "tiled fp32/fp64 radius pairs, open or periodic (HIP)",
+        pybind11::arg("pos"), pybind11::arg("batch"),
+        pybind11::arg("gptr"), pybind11::arg("r"),
+        pybind11::arg("loop") = false,
+        pybind11::arg("shifts") = pybind11::none());
   m.def("varlen_attention", &varlen_attention,
         "segment-varlen attention (HIP)");
   m.def("irreps_linear", &irreps_linear,

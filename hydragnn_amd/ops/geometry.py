@@ -104,8 +104,13 @@ def radius_graph(
             gptr = torch.zeros(
                 counts.numel() + 1, dtype=torch.long, device=pos.device)
             gptr[1:] = counts.cumsum(0)
-        src, dst, dist = ext.radius_pairs(
-            pos.contiguous().float(), batch_t, gptr, float(r), bool(loop))
+        # distances in the POSITION dtype (fp64 stays fp64: boundary
+        # membership is deterministic for the fp64 science configs)
+        p = pos.contiguous()
+        if p.dtype not in (torch.float32, torch.float64):
+            p = p.float()
+        src, dst, dist, _ = ext.radius_pairs_t(
+            p, batch_t, gptr, float(r), bool(loop))
         if max_num_neighbors < n and dst.numel() > 0:
             # cap: keep the max_num_neighbors closest srcs per dst
             order = torch.argsort(dst * (dist.max() + 1.0) + dist)
@@ -145,6 +150,9 @@ def radius_graph_pbc(
     """
     device = pos.device
     dtype = pos.dtype
+    if pos.is_cuda and not use_eager():
+        return _radius_graph_pbc_hip(pos, r, cell, pbc,
+                                     max_num_neighbors, loop)
     p = pos.detach().cpu().double().numpy()
     c = cell.detach().cpu().double().numpy().reshape(3, 3)
     pbc = np.asarray(pbc, dtype=bool).reshape(3)
@@ -225,3 +233,60 @@ def radius_graph_pbc(
     edge_index = torch.from_numpy(np.stack([src, dst])).long().to(device)
     edge_shifts = torch.from_numpy(sh).to(dtype).to(device)
     return edge_index, edge_shifts
+
+
+def _radius_graph_pbc_hip(pos, r, cell, pbc, max_num_neighbors, loop):
+    """GPU periodic neighbor list: image offsets enumerated on the
+    host from the cell's perpendicular widths, pair enumeration on the
+    tiled HIP kernel in the position dtype (fp32 or fp64)."""
+    ext = get_extension(required=True)
+    device = pos.device
+    dtype = pos.dtype
+    n = pos.shape[0]
+    if n == 0:
+        return (torch.zeros(2, 0, dtype=torch.long, device=device),
+                torch.zeros(0, 3, dtype=dtype, device=device))
+    c = cell.detach().cpu().double().numpy().reshape(3, 3)
+    pbc_arr = np.asarray(pbc, dtype=bool).reshape(3)
+    vol = abs(np.linalg.det(c))
+    n_img = np.zeros(3, dtype=int)
+    for i in range(3):
+        if not pbc_arr[i]:
+            continue
+        j, k = (i + 1) % 3, (i + 2) % 3
+        cross = np.cross(c[j], c[k])
+        area = np.linalg.norm(cross)
+        h = vol / area if area > 0 else np.inf
+        n_img[i] = int(math.ceil(r / h)) if h > 0 and np.isfinite(h) \
+            else 0
+    grids = np.meshgrid(*[np.arange(-n_img[i], n_img[i] + 1)
+                          for i in range(3)], indexing="ij")
+    shifts_int = np.stack([g.ravel() for g in grids],
+                          axis=1).astype(np.float64)      # [S, 3]
+    shift_cart = shifts_int @ c                            # [S, 3]
+    p = pos.detach().contiguous()
+    if p.dtype not in (torch.float32, torch.float64):
+        p = p.double()
+    # kernel convention: membership |p[src] + s - p[dst]| <= r; the
+    # edge convention vec = p[dst] - p[src] + shift  =>  shift = -s
+    sh_dev = torch.from_numpy(shift_cart).to(device=device,
+                                             dtype=p.dtype)
+    batch_t = torch.zeros(n, dtype=torch.long, device=device)
+    gptr = torch.tensor([0, n], dtype=torch.long, device=device)
+    src, dst, dist, simg = ext.radius_pairs_t(
+        p, batch_t, gptr, float(r), bool(loop), sh_dev)
+    edge_shifts = -sh_dev[simg]
+    if max_num_neighbors < n * max(1, sh_dev.shape[0]) \
+            and dst.numel() > 0:
+        order = torch.argsort(dst * (dist.max() + 1.0) + dist)
+        src, dst, edge_shifts = src[order], dst[order], \
+            edge_shifts[order]
+        counts = torch.bincount(dst, minlength=n)
+        seg_start = torch.zeros(n, dtype=torch.long, device=device)
+        seg_start[1:] = counts.cumsum(0)[:-1]
+        pos_in_seg = (torch.arange(dst.numel(), device=device)
+                      - seg_start[dst])
+        keep = pos_in_seg < max_num_neighbors
+        src, dst, edge_shifts = src[keep], dst[keep], edge_shifts[keep]
+    return (torch.stack([src, dst], dim=0),
+            edge_shifts.to(dtype))
