@@ -76,7 +76,8 @@ def test_fp16_scaler_training_gpu():
     torch.manual_seed(3)
     model = B.build_model("cuda:0", precision="fp16", seed=3)
     opt = torch.optim.AdamW(model.parameters(), lr=1e-3)
-    ds = md17_shape_dataset_fast(64, seed=8)
+    ds = md17_shape_dataset_fast(64, seed=8, min_dist=0.95,
+                                 max_push=40)
     loader = DataLoader(ds, batch_size=16, shuffle=False,
                         collate_fn=_collate)
     errs = []

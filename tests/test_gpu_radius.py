@@ -24,7 +24,7 @@ def test_radius_graph_gpu_matches_cpu(dtype, n, graphs):
     torch.manual_seed(0)
     pos = torch.rand(n * graphs, 3, dtype=dtype) * 10
     batch = torch.arange(graphs).repeat_interleave(n)
-    ei_cpu = _radius_graph_torch(pos, 2.0, batch, max_num_neighbors=n)
+    ei_cpu = _radius_graph_torch(pos, 2.0, batch, n, False)
     ei_gpu = radius_graph(pos.cuda(), 2.0, batch.cuda(),
                           max_num_neighbors=n)
     assert _edge_set(ei_cpu) == _edge_set(ei_gpu)
@@ -70,7 +70,7 @@ def test_radius_graph_pbc_gpu_1024_atoms():
     torch.manual_seed(2)
     n = 1024
     cell = (torch.eye(3, dtype=torch.float64) * 14.0).cuda()
-    pos = (torch.rand(n, 3, dtype=torch.float64) @ cell).cuda()
+    pos = torch.rand(n, 3, dtype=torch.float64).cuda() @ cell
     ei, sh = radius_graph_pbc(pos, 5.0, cell)
     assert ei.shape[1] > n * 10  # dense periodic neighborhood
     vec = pos[ei[1]] - pos[ei[0]] + sh
