@@ -56,7 +56,7 @@ def lj_dataset(
     """Random near-lattice LJ configurations. x = [Z]; energy/forces
     analytic; edges from the PBC-aware radius graph with shift vectors."""
     g = torch.Generator().manual_seed(seed)
-    n_side = max(1, round(num_atoms ** (1 / 3)))
+    n_side = max(1, math.ceil(num_atoms ** (1 / 3) - 1e-9))
     spacing = cell_size / n_side
     base = torch.stack(torch.meshgrid(
         torch.arange(n_side), torch.arange(n_side), torch.arange(n_side),

@@ -92,8 +92,10 @@ CONFIGS = {
     "dimenet_fp64": dict(
         label="DimeNet fp64 + grad checkpointing, periodic cells",
         precision="fp64", local_batch=4, mlip=False,
-        data=lambda b: lj_dataset(num_samples=b, num_atoms=216,
-                                  cell_size=12.0, radius=3.0,
+        # 1024-atom FeSi-shape cells at the 216-atom run's density
+        # (12^3 for 216 -> 20.25^3 for 1024); BASELINE configs[4]
+        data=lambda b: lj_dataset(num_samples=b, num_atoms=1024,
+                                  cell_size=20.25, radius=3.0,
                                   pbc=True, seed=31,
                                   dtype=torch.float64),
         model=dict(mpnn_type="DimeNet", input_dim=1, hidden_dim=64,
