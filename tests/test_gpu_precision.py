@@ -57,24 +57,34 @@ def test_mace_fp64_gpu_matches_cpu():
 
 
 def test_fp16_scaler_training_gpu():
-    """fp16 + GradScaler: a few training steps stay finite and the
-    loss decreases (the reference's fp16 path,
-    /root/reference/hydragnn/train/train_validate_test.py:87-103)."""
+    """fp16 + GradScaler on a graph-head model (the reference's fp16
+    usage, /root/reference/hydragnn/train/train_validate_test.py:87-103
+    — force double-backward is a bf16/fp32 path): loss decreases and
+    the scaler is exercised."""
     from torch.utils.data import DataLoader
 
-    import bench as B
+    from hydragnn_amd.models.create import create_model
     from hydragnn_amd.preprocess.load_data import _collate
     from hydragnn_amd.train import train
     from hydragnn_amd.train.train_validate_test import (
         get_autocast_and_scaler)
+    from hydragnn_amd.utils.datasets.synthetic import (
+        md17_shape_dataset_fast)
 
     autocast, scaler = get_autocast_and_scaler("fp16")
     assert scaler is not None, "fp16 must return a GradScaler"
 
-    from hydragnn_amd.utils.datasets.synthetic import (
-        md17_shape_dataset_fast)
     torch.manual_seed(3)
-    model = B.build_model("cuda:0", precision="fp16", seed=3)
+    model = create_model(
+        mpnn_type="SchNet", input_dim=1, hidden_dim=32,
+        output_dim=[1], output_type=["graph"],
+        output_heads={"graph": [{"type": "branch-0", "architecture": {
+            "num_sharedlayers": 1, "dim_sharedlayers": 32,
+            "num_headlayers": 2, "dim_headlayers": [32, 32]}}]},
+        activation_function="silu", loss_function_type="mse",
+        task_weights=[1.0], num_conv_layers=2, radius=7.0,
+        max_neighbours=30, num_gaussians=16, num_filters=32,
+        use_gpu=False).to("cuda:0")
     opt = torch.optim.AdamW(model.parameters(), lr=1e-3)
     ds = md17_shape_dataset_fast(64, seed=8, min_dist=0.95,
                                  max_push=40)
