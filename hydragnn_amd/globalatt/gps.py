@@ -137,10 +137,9 @@ class HydraGPSConv(nn.Module):
         if attn.in_proj_weight is None or not attn.batch_first:
             return None
         num_graphs = int(batch[-1]) + 1 if batch.numel() else 0
+        # (r2: the K/V-tiled kernel has no segment-length cap, so no
+        # device sync to check max_seg)
         ptr = _rowptr_from_sorted(batch, num_graphs)
-        max_seg = int((ptr[1:] - ptr[:-1]).max()) if num_graphs else 0
-        if not varlen_eligible(head_dim, max_seg, x.device):
-            return None
         qkv = nn.functional.linear(x, attn.in_proj_weight,
                                    attn.in_proj_bias)
         q, k, v = qkv.view(x.shape[0], 3, self.heads,

@@ -22,8 +22,8 @@ import torch
 from ._extension import get_extension
 from ..data import to_dense_batch
 
-MAX_SEG = 256
-MAX_DH = 32
+MAX_SEG = None  # r2: K/V tiling removed the segment cap
+MAX_DH = 64
 
 
 def torch_varlen_attention(q: torch.Tensor, k: torch.Tensor,
@@ -78,10 +78,11 @@ def varlen_attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
     CPU."""
     if not q.is_cuda:
         return torch_varlen_attention(q, k, v, batch)
-    orig_dtype = q.dtype
-    q, k, v = (t.float() for t in (q, k, v))
-    out = _VarlenAttn.apply(q, k, v, ptr, batch)
-    return out.to(orig_dtype)
+    if q.dtype not in (torch.float32, torch.bfloat16):
+        orig_dtype = q.dtype
+        q, k, v = (t.float() for t in (q, k, v))
+        return _VarlenAttn.apply(q, k, v, ptr, batch).to(orig_dtype)
+    return _VarlenAttn.apply(q, k, v, ptr, batch)
 
 
 def varlen_eligible(head_dim: int, max_seg: int, device) -> bool:
@@ -90,4 +91,4 @@ def varlen_eligible(head_dim: int, max_seg: int, device) -> bool:
         return False
     if not (isinstance(device, torch.device) and device.type == "cuda"):
         return False
-    return head_dim <= MAX_DH and max_seg <= MAX_SEG
+    return head_dim <= MAX_DH

@@ -256,3 +256,27 @@ def test_scatter_mean_sorted_csr():
                              retain_graph=True)[0]
     g2 = torch.autograd.grad(out_atomic.square().sum(), src)[0]
     assert (g1 - g2).abs().max() < 1e-5
+
+
+@pytest.mark.parametrize("dtype,dh,sizes", [
+    (torch.bfloat16, 16, [5, 80, 37]),
+    (torch.float32, 64, [12, 300, 1]),
+    (torch.bfloat16, 64, [700, 4, 1200]),
+])
+def test_varlen_attention_bf16_large_segments(dtype, dh, sizes):
+    """r2 kernel envelope: bf16, head_dim up to 64, segments beyond the
+    r1 cap of 256 via K/V LDS tiling."""
+    from hydragnn_amd.ops.varlen_attn import (
+        varlen_attention, torch_varlen_attention)
+    from hydragnn_amd.ops.scatter import _rowptr_from_sorted
+    torch.manual_seed(1)
+    batch = torch.repeat_interleave(
+        torch.arange(len(sizes)), torch.tensor(sizes)).to("cuda")
+    ptr = _rowptr_from_sorted(batch, len(sizes))
+    N, H = int(sum(sizes)), 4
+    q, k, v = (torch.randn(N, H, dh, device="cuda", dtype=dtype)
+               for _ in range(3))
+    out = varlen_attention(q, k, v, ptr, batch)
+    ref = torch_varlen_attention(q.float(), k.float(), v.float(), batch)
+    tol = 2e-2 if dtype == torch.bfloat16 else 1e-5
+    assert (out.float() - ref).abs().max() < tol
