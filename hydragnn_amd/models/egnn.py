@@ -14,6 +14,8 @@ from typing import Optional
 import torch
 from torch import nn
 
+from ..ops.mfma_linear import MFMALinear
+
 from ..ops import gather, get_edge_vectors_and_lengths, scatter
 from .base import Base
 
@@ -26,16 +28,16 @@ class E_GCL(nn.Module):
         self.equivariant = equivariant
         in_edge = 2 * input_nf + 1 + edge_dim
         self.edge_mlp = nn.Sequential(
-            nn.Linear(in_edge, hidden_nf), act,
-            nn.Linear(hidden_nf, hidden_nf), act)
+            MFMALinear(in_edge, hidden_nf), act,
+            MFMALinear(hidden_nf, hidden_nf), act)
         self.node_mlp = nn.Sequential(
-            nn.Linear(input_nf + hidden_nf, hidden_nf), act,
-            nn.Linear(hidden_nf, output_nf))
+            MFMALinear(input_nf + hidden_nf, hidden_nf), act,
+            MFMALinear(hidden_nf, output_nf))
         if equivariant:
-            layer = nn.Linear(hidden_nf, 1, bias=False)
+            layer = MFMALinear(hidden_nf, 1, bias=False)
             nn.init.xavier_uniform_(layer.weight, gain=0.001)
             self.coord_mlp = nn.Sequential(
-                nn.Linear(hidden_nf, hidden_nf), act, layer)
+                MFMALinear(hidden_nf, hidden_nf), act, layer)
 
     def forward(self, h, pos, edge_index, edge_attr=None, edge_shifts=None):
         src, dst = edge_index[0], edge_index[1]
