@@ -223,6 +223,8 @@ def train(loader, model, opt, verbosity, precision="fp32", profiler=None):
         set_reshard_after_backward(model, False)
 
     from .captured import capture_enabled, get_or_build_stepper
+    from ..utils.distributed import is_deepspeed_engine
+    ds_engine = is_deepspeed_engine(model)
     stepper = None
 
     for ibatch, data in enumerate(iterate_tqdm(loader, verbosity)):
@@ -234,7 +236,7 @@ def train(loader, model, opt, verbosity, precision="fp32", profiler=None):
 
         # hipGraph-captured fast path (static-shape batches on GPU):
         # H2D copy-in + one graph replay per batch (captured.py).
-        if (scaler is None and stepper is None
+        if (scaler is None and stepper is None and not ds_engine
                 and capture_enabled(data)):
             dev_data = move_batch_to_device(data, param_dtype)
             stepper = get_or_build_stepper(
@@ -252,20 +254,27 @@ def train(loader, model, opt, verbosity, precision="fp32", profiler=None):
             tr.start("h2d")
             data = move_batch_to_device(data, param_dtype)
             tr.stop("h2d")
-            opt.zero_grad(set_to_none=True)
+            if not ds_engine:
+                opt.zero_grad(set_to_none=True)
             tr.start("forward")
             with autocast:
                 loss, tasks_loss, _ = _compute_loss(model, data, use_ia,
                                                     create_graph=True)
             tr.stop("forward")
             tr.start("backward")
-            if scaler is not None:
+            if ds_engine:
+                # DeepSpeed engine owns loss scaling / grad
+                # accumulation (reference train_validate_test.py:729)
+                model.backward(loss)
+            elif scaler is not None:
                 scaler.scale(loss).backward()
             else:
                 loss.backward()
             tr.stop("backward")
             tr.start("opt_step")
-            if scaler is not None:
+            if ds_engine:
+                model.step()
+            elif scaler is not None:
                 scaler.step(opt)
                 scaler.update()
             else:

@@ -228,7 +228,14 @@ def deepspeed_model_wrapper(model, optimizer, config):
     ds_config = parse_deepspeed_config(config)
     engine, optimizer, _, _ = deepspeed.initialize(
         model=model, optimizer=optimizer, config=ds_config)
+    # marker the train loop keys its engine.backward/engine.step hooks
+    # on (reference train_validate_test.py:729,780,797)
+    engine._hydragnn_deepspeed = True
     return engine, optimizer
+
+
+def is_deepspeed_engine(model) -> bool:
+    return bool(getattr(model, "_hydragnn_deepspeed", False))
 
 
 def distributed_model_wrapper(model, max_neighbours=None, verbosity: int = 0,

@@ -115,3 +115,72 @@ def create_ds_config(params, job_id, DEEPHYPER_LOG_DIR="."):
     with open(path, "w") as f:
         json.dump(cfg, f, indent=2)
     return path
+
+
+def run_deephyper_search(objective: Callable[[Dict], float], space: Dict,
+                         num_trials: int = 10, seed: int = 0,
+                         maximize: bool = False):
+    """DeepHyper CBO search (reference deephyper.py:15-187 uses
+    DeepHyper's HpProblem/CBO on the cluster).  Requires deephyper;
+    see run_search for the auto-fallback entry point.
+
+    Space grammar matches sample_config: list -> categorical,
+    (lo, hi) -> uniform int/float, (lo, hi, 'log') -> log-uniform."""
+    from deephyper.hpo import CBO, HpProblem
+    try:
+        import ConfigSpace.hyperparameters as csh
+    except ImportError:
+        csh = None
+
+    problem = HpProblem()
+    for name, spec in space.items():
+        if isinstance(spec, list):
+            problem.add_hyperparameter(spec, name)
+        elif isinstance(spec, tuple) and len(spec) == 3 \
+                and spec[2] == "log":
+            if csh is not None:
+                problem.add_hyperparameter(
+                    csh.UniformFloatHyperparameter(
+                        name=name, lower=spec[0], upper=spec[1],
+                        log=True), name)
+            else:
+                problem.add_hyperparameter((spec[0], spec[1]), name)
+        else:
+            problem.add_hyperparameter((spec[0], spec[1]), name)
+
+    # DeepHyper maximizes; flip the sign for minimization problems.
+    def dh_objective(cfg):
+        cfg = dict(cfg)
+        cfg.pop("job_id", None)
+        val = objective(cfg)
+        return val if maximize else -val
+
+    search = CBO(problem, dh_objective, random_state=seed)
+    results = search.search(max_evals=num_trials)
+    # results: DataFrame with 'objective' and 'p:<name>' columns
+    objs = results["objective"]
+    idx = objs.idxmax()
+    best_row = results.loc[idx]
+    best_cfg = {c[2:]: best_row[c] for c in results.columns
+                if c.startswith("p:")}
+    best_val = float(best_row["objective"])
+    return best_cfg, (best_val if maximize else -best_val)
+
+
+def run_search(objective: Callable[[Dict], float], space: Dict,
+               num_trials: int = 10, seed: int = 0,
+               maximize: bool = False, use_deephyper: bool = False):
+    """HPO entry point: DeepHyper CBO when requested and importable,
+    offline random search otherwise.  Returns (best_config,
+    best_value)."""
+    if use_deephyper:
+        try:
+            import deephyper  # noqa: F401
+            return run_deephyper_search(objective, space, num_trials,
+                                        seed, maximize)
+        except ImportError:
+            pass
+    best_cfg, best_val, _ = run_random_search(objective, space,
+                                              num_trials, seed,
+                                              maximize)
+    return best_cfg, best_val

@@ -106,11 +106,69 @@ class RocmEnergyTracer(TracerBackend):
                         f"{self.energy[name]:.3f} J\n")
 
 
-def initialize(energy: bool = False, verbose: bool = False):
+class GPTLTracer(TracerBackend):
+    """Adapter over gptl4py (the reference's GPTL backend,
+    reference tracer.py:29-120).  Import-gated: present on reference
+    clusters, absent in the MI355X image."""
+
+    def __init__(self):
+        import gptl4py as gp
+        self.gp = gp
+        gp.initialize()
+
+    def start(self, name):
+        self.gp.start(name)
+
+    def stop(self, name):
+        self.gp.stop(name)
+
+    def save(self, path, rank):
+        os.makedirs(path, exist_ok=True)
+        self.gp.pr_file(os.path.join(path, f"gp_timing.gptl.p{rank}"))
+        if rank == 0 and hasattr(self.gp, "pr_summary_file"):
+            self.gp.pr_summary_file(
+                os.path.join(path, "gp_timing.gptl.summary"))
+
+
+class ScorePTracer(TracerBackend):
+    """Adapter over scorep.user regions (reference tracer.py Score-P
+    backend).  Import-gated."""
+
+    def __init__(self):
+        import scorep.user as su
+        self.su = su
+
+    def start(self, name):
+        self.su.region_begin(name)
+
+    def stop(self, name):
+        self.su.region_end(name)
+
+    def save(self, path, rank):
+        pass  # Score-P writes its own experiment archive
+
+
+def initialize(energy: bool = False, verbose: bool = False,
+               extra_backends=None):
+    """Build the backend list: wall timer (+ per-call history) always;
+    ROCm energy on request; GPTL / Score-P adapters when named in
+    `extra_backends` or HYDRAGNN_TRACER_BACKENDS="gptl,scorep" AND
+    importable (silently skipped otherwise, matching the reference's
+    optional-import behavior)."""
     global _backends
     _backends = [WallTimer()]
     if energy:
         _backends.append(RocmEnergyTracer())
+    names = set(extra_backends or [])
+    env = os.environ.get("HYDRAGNN_TRACER_BACKENDS", "")
+    names |= {n.strip() for n in env.split(",") if n.strip()}
+    for name, cls in (("gptl", GPTLTracer), ("scorep", ScorePTracer)):
+        if name in names:
+            try:
+                _backends.append(cls())
+            except ImportError:
+                if verbose:
+                    print(f"[tracer] backend '{name}' unavailable")
 
 
 def enable():
