@@ -107,7 +107,8 @@ CONFIG = {
         "Training": {
             "num_epoch": 10, "perc_train": 0.8, "batch_size": 16,
             "loss_function_type": "mse",
-            "Optimizer": {"type": "AdamW", "learning_rate": 0.005},
+            "Optimizer": {"type": "AdamW", "learning_rate": float(
+                os.environ.get("HYDRAGNN_HPO_LR", 0.005))},
         },
     },
 }
