@@ -296,9 +296,16 @@ def main():
 
     precision = os.environ.get("HYDRAGNN_BENCH_PRECISION", PRECISION)
     model = build_model(device, precision=precision)
+    # FusedAdamW flattens params into one buffer (single-kernel step);
+    # construct BEFORE the DDP wrap so reducer bucket views are built
+    # over the flattened storages.
+    try:
+        from hydragnn_amd.ops.fused_adamw import FusedAdamW
+        optimizer = FusedAdamW(model.parameters(), lr=1e-3)
+    except TypeError:  # non-fp32 params (fp64 benches)
+        optimizer = torch.optim.AdamW(model.parameters(), lr=1e-3,
+                                      foreach=True)
     model = distributed_model_wrapper(model)
-    optimizer = torch.optim.AdamW(model.parameters(), lr=1e-3,
-                                  foreach=True)
 
     warm_loader, timed_loader = make_loaders(
         rank, args.steps, args.warmup, args.batch, use_cuda,

@@ -709,6 +709,10 @@ torch::Tensor irreps_linear_gw(torch::Tensor X, torch::Tensor G,
 // defined in gemv.hip
 torch::Tensor gemv_small_n(torch::Tensor A, torch::Tensor W,
                            c10::optional<torch::Tensor> bias);
+// defined in fused_adamw.hip
+void fused_adamw(torch::Tensor p, torch::Tensor g, torch::Tensor m,
+                 torch::Tensor v, torch::Tensor step, double lr,
+                 double beta1, double beta2, double eps, double wd);
 // defined in radius.hip
 std::vector<torch::Tensor> radius_pairs_t(torch::Tensor pos,
                                           torch::Tensor batch,
@@ -752,6 +756,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("scatter_mean_fwd", &scatter_mean_fwd, "scatter-mean (HIP)");
   m.def("scatter_minmax_fwd", &scatter_minmax_fwd, "scatter-min/max (HIP)");
   m.def("radius_pairs", &radius_pairs, "radius pair enumeration (HIP)");
+  m.def("fused_adamw", &fused_adamw,
+        "single-kernel flat AdamW (HIP)");
   m.def("radius_pairs_t", &radius_pairs_t,
         "tiled fp32/fp64 radius pairs, open or periodic (HIP)",
         pybind11::arg("pos"), pybind11::arg("batch"),

@@ -72,16 +72,20 @@ class CapturedTrainStep:
         # Pre-assign every grad as a view into ONE flat buffer: the
         # per-replay zeroing is a single fill kernel instead of ~100,
         # and the DP sync / a fused optimizer can address the whole
-        # gradient contiguously.
+        # gradient contiguously.  A FusedAdamW optimizer already owns
+        # such a buffer (ops/fused_adamw.py) — reuse it.
         params = [p for p in model.parameters() if p.requires_grad]
-        total = sum(p.numel() for p in params)
-        dev = next(iter(params)).device if params else "cuda"
-        self._flat_grad = torch.zeros(total, dtype=param_dtype,
-                                      device=dev)
-        off = 0
-        for p in params:
-            p.grad = self._flat_grad[off:off + p.numel()].view_as(p)
-            off += p.numel()
+        if getattr(opt, "flat_grad", None) is not None:
+            self._flat_grad = opt.flat_grad
+        else:
+            total = sum(p.numel() for p in params)
+            dev = next(iter(params)).device if params else "cuda"
+            self._flat_grad = torch.zeros(total, dtype=param_dtype,
+                                          device=dev)
+            off = 0
+            for p in params:
+                p.grad = self._flat_grad[off:off + p.numel()].view_as(p)
+                off += p.numel()
 
         def fwd_bwd():
             self._flat_grad.zero_()

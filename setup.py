@@ -15,6 +15,7 @@ ext_modules = [
                  "hydragnn_amd/ops/csrc/varlen_attn.hip",
                  "hydragnn_amd/ops/csrc/gemv.hip",
                  "hydragnn_amd/ops/csrc/radius.hip",
+                 "hydragnn_amd/ops/csrc/fused_adamw.hip",
                  "hydragnn_amd/ops/csrc/irreps_linear.hip"],
         extra_compile_args={
             "cxx": ["-O3"],
