@@ -69,7 +69,9 @@ MODEL_CONFIG = {
 }
 
 LOCAL_BATCH = int(os.environ.get("HYDRAGNN_BENCH_BATCH", "1024"))
-PRECISION = "bf16"
+# pure-bf16 training: bf16 params/activations (no autocast cast
+# traffic) + fp32 master weights in FusedAdamW — reported dtype bf16
+PRECISION = "bf16_pure"
 
 
 def build_model_and_batch(device="cuda:0", local_batch=LOCAL_BATCH,
@@ -370,7 +372,8 @@ def main():
             "higher_is_better": True,
             "scaling": os.environ.get("HYDRAGNN_BENCH_SCALING", "weak"),
             "vs_baseline": None,
-            "dtype": precision,
+            "dtype": "bf16" if precision.startswith("bf16")
+                     else precision,
             "data": "synthetic",
             "config": {
                 "model": "MACE-MLIP energy+forces",

@@ -17,6 +17,10 @@ from .stacks import CGCNNStack, GATStack, GINStack, MFCStack, PNAStack, SAGEStac
 
 PRECISION_MAP = {
     "bf16": {"param_dtype": torch.float32, "autocast_dtype": torch.bfloat16},
+    # pure bf16: parameters AND activations bf16, no autocast layer —
+    # fp32 optimizer master weights via FusedAdamW.  Removes the
+    # per-op cast traffic of autocast on memory-bound GNN steps.
+    "bf16_pure": {"param_dtype": torch.bfloat16, "autocast_dtype": None},
     "fp16": {"param_dtype": torch.float32, "autocast_dtype": torch.float16},
     "fp32": {"param_dtype": torch.float32, "autocast_dtype": None},
     "fp64": {"param_dtype": torch.float64, "autocast_dtype": None},
@@ -28,6 +32,7 @@ def resolve_precision(precision):
         precision = "fp32"
     prec = str(precision).lower()
     aliases = {"bfloat16": "bf16", "float16": "fp16", "half": "fp16",
+               "bf16-pure": "bf16_pure", "pure_bf16": "bf16_pure",
                "float32": "fp32", "float": "fp32",
                "float64": "fp64", "double": "fp64"}
     prec = aliases.get(prec, prec)

@@ -713,6 +713,11 @@ torch::Tensor gemv_small_n(torch::Tensor A, torch::Tensor W,
 void fused_adamw(torch::Tensor p, torch::Tensor g, torch::Tensor m,
                  torch::Tensor v, torch::Tensor step, double lr,
                  double beta1, double beta2, double eps, double wd);
+void fused_adamw_bf16(torch::Tensor p, torch::Tensor g,
+                      torch::Tensor master, torch::Tensor m,
+                      torch::Tensor v, torch::Tensor step, double lr,
+                      double beta1, double beta2, double eps,
+                      double wd);
 // defined in radius.hip
 std::vector<torch::Tensor> radius_pairs_t(torch::Tensor pos,
                                           torch::Tensor batch,
@@ -758,6 +763,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("radius_pairs", &radius_pairs, "radius pair enumeration (HIP)");
   m.def("fused_adamw", &fused_adamw,
         "single-kernel flat AdamW (HIP)");
+  m.def("fused_adamw_bf16", &fused_adamw_bf16,
+        "single-kernel flat AdamW, bf16 params + fp32 master (HIP)");
   m.def("radius_pairs_t", &radius_pairs_t,
         "tiled fp32/fp64 radius pairs, open or periodic (HIP)",
         pybind11::arg("pos"), pybind11::arg("batch"),

@@ -30,17 +30,24 @@ class FusedAdamW(torch.optim.Optimizer):
         params = [p for p in params if p.requires_grad]
         if not params:
             raise ValueError("FusedAdamW: no parameters")
-        if any(p.dtype != torch.float32 for p in params):
-            raise TypeError("FusedAdamW supports fp32 parameters only")
+        dtypes = {p.dtype for p in params}
+        if dtypes == {torch.float32}:
+            self.param_dtype = torch.float32
+        elif dtypes == {torch.bfloat16}:
+            # pure-bf16 training: bf16 working params + fp32 master
+            self.param_dtype = torch.bfloat16
+        else:
+            raise TypeError(
+                "FusedAdamW supports uniform fp32 or bf16 parameters")
         defaults = dict(lr=lr, betas=betas, eps=eps,
                         weight_decay=weight_decay)
         super().__init__(params, defaults)
 
         device = params[0].device
         total = sum(p.numel() for p in params)
-        self.flat_param = torch.empty(total, dtype=torch.float32,
+        self.flat_param = torch.empty(total, dtype=self.param_dtype,
                                       device=device)
-        self.flat_grad = torch.zeros(total, dtype=torch.float32,
+        self.flat_grad = torch.zeros(total, dtype=self.param_dtype,
                                      device=device)
         self.exp_avg = torch.zeros(total, dtype=torch.float32,
                                    device=device)
@@ -57,6 +64,8 @@ class FusedAdamW(torch.optim.Optimizer):
                 p.data = self.flat_param[off:off + n].view_as(p)
                 p.grad = self.flat_grad[off:off + n].view_as(p)
                 off += n
+        self.master = (self.flat_param.float()
+                       if self.param_dtype == torch.bfloat16 else None)
 
     def zero_grad(self, set_to_none: bool = True):
         # grads are views into the flat buffer: never drop them
@@ -70,22 +79,32 @@ class FusedAdamW(torch.optim.Optimizer):
         eps, wd = g["eps"], g["weight_decay"]
         if self.flat_param.is_cuda and not use_eager():
             ext = get_extension(required=True)
-            ext.fused_adamw(self.flat_param, self.flat_grad,
-                            self.exp_avg, self.exp_avg_sq, self.step_t,
-                            lr, b1, b2, eps, wd)
+            if self.master is None:
+                ext.fused_adamw(self.flat_param, self.flat_grad,
+                                self.exp_avg, self.exp_avg_sq,
+                                self.step_t, lr, b1, b2, eps, wd)
+            else:
+                ext.fused_adamw_bf16(self.flat_param, self.flat_grad,
+                                     self.master, self.exp_avg,
+                                     self.exp_avg_sq, self.step_t,
+                                     lr, b1, b2, eps, wd)
             return loss
         # CPU / eager fallback: identical math on the flat buffers
         self.step_t += 1
         t = float(self.step_t.item())
-        self.flat_param.mul_(1.0 - lr * wd)
-        self.exp_avg.mul_(b1).add_(self.flat_grad, alpha=1.0 - b1)
-        self.exp_avg_sq.mul_(b2).addcmul_(self.flat_grad,
-                                          self.flat_grad,
-                                          value=1.0 - b2)
+        work = self.master if self.master is not None \
+            else self.flat_param
+        g32 = self.flat_grad.float() if self.master is not None \
+            else self.flat_grad
+        work.mul_(1.0 - lr * wd)
+        self.exp_avg.mul_(b1).add_(g32, alpha=1.0 - b1)
+        self.exp_avg_sq.mul_(b2).addcmul_(g32, g32, value=1.0 - b2)
         bc1 = 1.0 - b1 ** t
         bc2 = 1.0 - b2 ** t
         denom = (self.exp_avg_sq / bc2).sqrt_().add_(eps)
-        self.flat_param.addcdiv_(self.exp_avg / bc1, denom, value=-lr)
+        work.addcdiv_(self.exp_avg / bc1, denom, value=-lr)
+        if self.master is not None:
+            self.flat_param.copy_(work)
         return loss
 
     def state_dict(self):
@@ -93,10 +112,14 @@ class FusedAdamW(torch.optim.Optimizer):
             "param_groups": self.param_groups,
             "flat": {"exp_avg": self.exp_avg,
                      "exp_avg_sq": self.exp_avg_sq,
-                     "step": self.step_t},
+                     "step": self.step_t,
+                     "master": self.master},
         }
 
     def load_state_dict(self, sd):
+        if sd["flat"].get("master") is not None \
+                and self.master is not None:
+            self.master.copy_(sd["flat"]["master"])
         self.exp_avg.copy_(sd["flat"]["exp_avg"])
         self.exp_avg_sq.copy_(sd["flat"]["exp_avg_sq"])
         self.step_t.copy_(sd["flat"]["step"])

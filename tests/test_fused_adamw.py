@@ -56,6 +56,32 @@ def test_fused_adamw_state_roundtrip():
     m2 = _model(3)
     o2 = FusedAdamW(m2.parameters(), lr=1e-3)
     o2.load_state_dict({"param_groups": sd["param_groups"],
-                        "flat": {k: v.clone()
+                        "flat": {k: (v.clone() if v is not None else None)
                                  for k, v in sd["flat"].items()}})
     assert float(o2.step_t) == 1.0
+
+
+def test_fused_adamw_bf16_master_mode_cpu():
+    """bf16 params + fp32 master: trajectory tracks an fp32
+    torch.optim.AdamW run within bf16 resolution."""
+    m_ref = _model(4)
+    m_bf = _model(4)
+    m_bf.load_state_dict(m_ref.state_dict())
+    m_bf = m_bf.to(torch.bfloat16)
+    o_ref = torch.optim.AdamW(m_ref.parameters(), lr=3e-3,
+                              weight_decay=0.01)
+    o_bf = FusedAdamW(m_bf.parameters(), lr=3e-3, weight_decay=0.01)
+    assert o_bf.master is not None
+    x = torch.randn(32, 6)
+    y = torch.randn(32, 3)
+    for _ in range(8):
+        o_ref.zero_grad()
+        torch.nn.functional.mse_loss(m_ref(x), y).backward()
+        o_ref.step()
+        o_bf.zero_grad()
+        torch.nn.functional.mse_loss(
+            m_bf(x.bfloat16()).float(), y).backward()
+        o_bf.step()
+    for p1, p2 in zip(m_ref.parameters(), m_bf.parameters()):
+        assert torch.allclose(p1, p2.float(), rtol=0.1, atol=0.05), \
+            (p1 - p2.float()).abs().max()
