@@ -66,9 +66,14 @@ class _MFMAMatmul(torch.autograd.Function):
             if ctx.needs_input_grad[0]:
                 gA = _MFMAMatmul.apply(g, B, True)
             if ctx.needs_input_grad[1]:
+                # _splitk_weight_grad(x2d, g2d) returns
+                # [g2d_cols, x2d_cols]: (g, A) -> [K, N] = A^T g
+                # directly (r2 fix: the old .t() silently transposed
+                # the second-order weight grad; square shapes passed
+                # the shape check with WRONG values)
                 gB = _splitk_weight_grad(g.reshape(-1, g.shape[-1]),
                                          A.reshape(-1, A.shape[-1]),
-                                         64).t().contiguous()
+                                         64).contiguous()
         return gA, gB, None
 
 

@@ -235,3 +235,29 @@ def test_gemv_linear_grads_match_fp32():
                           ref.weight.grad, rtol=5e-2, atol=1.0)
     assert torch.allclose(x1.grad.float(), x2.grad,
                           rtol=5e-2, atol=5e-2)
+
+
+@pytest.mark.parametrize("M,N,K", [(512, 128, 64), (512, 64, 128)])
+def test_mfma_second_order_weight_grad(M, N, K):
+    """Second-order weight gradient through _MFMAMatmul (the force
+    double-backward path) vs fp32 autograd — NON-square shapes so a
+    transposed result cannot hide (r2 regression test: the old
+    trans_b=False path returned g^T A instead of A^T g)."""
+    torch.manual_seed(0)
+    x = (_asym((M, K), 11) * 0.2).requires_grad_(True)
+    w = (_asym((N, K), 12) * 0.2).requires_grad_(True)
+    y = _MFMAMatmul.apply(x, w, True)       # x @ w^T
+    gx = torch.autograd.grad(y.float().square().sum(), x,
+                             create_graph=True)[0]
+    # second backward: gradient of ||gx||^2 w.r.t. w exercises the
+    # trans_b=False gB path
+    gw2 = torch.autograd.grad(gx.float().square().sum(), w)[0]
+
+    xr = x.detach().float().requires_grad_(True)
+    wr = w.detach().float().requires_grad_(True)
+    yr = xr @ wr.t()
+    gxr = torch.autograd.grad(yr.square().sum(), xr,
+                              create_graph=True)[0]
+    gw2r = torch.autograd.grad(gxr.square().sum(), wr)[0]
+    rel = (gw2.float() - gw2r).abs().max() / gw2r.abs().max().clamp(min=1e-6)
+    assert rel < 5e-2, f"rel {rel:.3e}"
