@@ -82,16 +82,18 @@ def test_captured_second_epoch_reuses_graph():
         md17_shape_dataset_fast)
 
     torch.manual_seed(5)
-    model = bench_mod.build_model("cuda:0", seed=5)
-    opt = torch.optim.AdamW(model.parameters(), lr=1e-3, foreach=True)
+    model = bench_mod.build_model("cuda:0", precision="bf16_pure",
+                                  seed=5)
+    from hydragnn_amd.ops.fused_adamw import FusedAdamW
+    opt = FusedAdamW(model.parameters(), lr=1e-3)
     ds = md17_shape_dataset_fast(32, seed=3)
     nc, ec = compute_static_caps(ds, 16)
     loader = DataLoader(ds, batch_size=16, shuffle=False,
                         collate_fn=StaticShapeCollater(nc, ec, 28.0))
-    train(loader, model, opt, 0, precision="bf16")
+    train(loader, model, opt, 0, precision="bf16_pure")
     s1 = getattr(model, "_hip_captured_step", None)
     assert s1 not in (None, False)
-    train(loader, model, opt, 0, precision="bf16")
+    train(loader, model, opt, 0, precision="bf16_pure")
     s2 = getattr(model, "_hip_captured_step", None)
     assert s2 is s1, "graph was re-captured"
     model._hip_captured_step = None
