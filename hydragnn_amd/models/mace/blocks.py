@@ -273,14 +273,18 @@ class RealAgnosticAttResidualInteractionBlock(
         # gathered as 64-wide outputs — identical numerics to
         # cat([radial, down[src], down[dst]]) @ W1^T at ~1/13 the GEMM
         # work (E >> N)
+        from ...ops.splitk_linear import _SplitKLinearFn
         lin1 = self.radial_mlp[0]
         W1 = lin1.weight
         rd = W1.shape[1] - 2 * down.shape[1]
         c = down.shape[1]
-        h = torch.nn.functional.linear(edge_radial,
-                                       W1[:, :rd], lin1.bias)
-        h_s = torch.nn.functional.linear(down, W1[:, rd:rd + c])
-        h_d = torch.nn.functional.linear(down, W1[:, rd + c:])
+        # split-K weight grads: F.linear's dW here is a [64, E] @
+        # [E, 8] GEMM with K = num_edges — hipBLASLt runs it on
+        # (M/16)x(N/16) workgroups, a serial-K crawl (measured
+        # ~1.4 ms/step at b1024)
+        h = _SplitKLinearFn.apply(edge_radial, W1[:, :rd], lin1.bias)
+        h_s = _SplitKLinearFn.apply(down, W1[:, rd:rd + c], None)
+        h_d = _SplitKLinearFn.apply(down, W1[:, rd + c:], None)
         h = h + gather(h_s, src, backward_csr=src_csr) \
             + gather(h_d, dst, backward_csr=dst_csr)
         return self.radial_mlp[1:](h)
