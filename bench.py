@@ -308,6 +308,10 @@ def main():
         optimizer = torch.optim.AdamW(model.parameters(), lr=1e-3,
                                       foreach=True)
     model = distributed_model_wrapper(model)
+    if getattr(optimizer, "master", None) is not None:
+        # DDP's construction-time broadcast may have overwritten the
+        # flat params; keep the fp32 master in sync
+        optimizer.master.copy_(optimizer.flat_param.float())
 
     warm_loader, timed_loader = make_loaders(
         rank, args.steps, args.warmup, args.batch, use_cuda,
