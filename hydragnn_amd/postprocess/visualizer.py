@@ -220,3 +220,107 @@ class Visualizer:
         fig.tight_layout()
         fig.savefig(os.path.join(self.outdir, "num_nodes.png"), dpi=120)
         plt.close(fig)
+
+    def create_plot_global(self, true_values, predicted_values,
+                           output_names=None):
+        """One parity panel per head in a single figure (reference
+        create_plot_global pattern)."""
+        if not _HAS_MPL:
+            return
+        n = len(true_values)
+        fig, axes = plt.subplots(1, max(n, 1),
+                                 figsize=(5 * max(n, 1), 4),
+                                 squeeze=False)
+        for ihead in range(n):
+            ax = axes[0][ihead]
+            t = true_values[ihead].detach().cpu().flatten().numpy()
+            p = predicted_values[ihead].detach().cpu().flatten().numpy()
+            ax.scatter(t, p, s=4, alpha=0.5)
+            self.add_identity(ax, "k--", lw=1)
+            name = (output_names[ihead] if output_names
+                    and ihead < len(output_names) else f"head{ihead}")
+            ax.set_title(name)
+            ax.set_xlabel("true")
+            ax.set_ylabel("predicted")
+        fig.tight_layout()
+        fig.savefig(os.path.join(self.outdir, "parity_global.png"),
+                    dpi=120)
+        plt.close(fig)
+
+    def create_parity_plot_and_error_histogram_scalar(
+            self, varname, true_values, predicted_values, iepoch=None):
+        """Scalar head: parity scatter + error PDF side by side; for
+        per-node-slot outputs (fixed-size graphs) a grid of per-slot
+        parity panels plus SUM panels (reference
+        create_parity_plot_and_error_histogram_scalar pattern)."""
+        if not _HAS_MPL:
+            return
+        import numpy as np
+        t = true_values.detach().cpu().numpy()
+        p = predicted_values.detach().cpu().numpy()
+        if t.ndim == 1 or t.shape[-1] == 1:
+            t, p = t.reshape(-1), p.reshape(-1)
+            fig, axes = plt.subplots(1, 2, figsize=(10, 4.5))
+            axes[0].scatter(t, p, s=4, alpha=0.5)
+            self.add_identity(axes[0], "k--", lw=1)
+            axes[0].set_title(varname)
+            axes[0].set_xlabel("true")
+            axes[0].set_ylabel("predicted")
+            hist, edges = np.histogram(p - t, bins=40, density=True)
+            axes[1].plot(0.5 * (edges[:-1] + edges[1:]), hist, "ro")
+            axes[1].set_title(f"{varname}: error PDF")
+        else:
+            nslots = t.shape[1]
+            ncol = int(np.ceil(np.sqrt(nslots + 1)))
+            nrow = int(np.ceil((nslots + 1) / ncol))
+            fig, axes = plt.subplots(nrow, ncol,
+                                     figsize=(3 * ncol, 3 * nrow),
+                                     squeeze=False)
+            flat = axes.flatten()
+            for s in range(nslots):
+                flat[s].scatter(t[:, s], p[:, s], s=4, alpha=0.5)
+                self.add_identity(flat[s], "k--", lw=1)
+                flat[s].set_title(f"slot {s}", fontsize=8)
+            flat[nslots].scatter(t.sum(1), p.sum(1), s=10)
+            self.add_identity(flat[nslots], "k--", lw=1)
+            flat[nslots].set_title("SUM", fontsize=8)
+            for s in range(nslots + 1, len(flat)):
+                flat[s].axis("off")
+        fig.tight_layout()
+        suffix = f"_epoch{iepoch}" if iepoch is not None else ""
+        fig.savefig(os.path.join(
+            self.outdir, f"parity_hist_{varname}{suffix}.png"), dpi=120)
+        plt.close(fig)
+
+    def create_parity_plot_per_node_vector(
+            self, varname, true_values, predicted_values,
+            num_nodes, iepoch=None):
+        """Vector node targets on fixed-size graphs: per-node-slot
+        parity of the vector magnitude (reference
+        create_parity_plot_per_node_vector pattern)."""
+        if not _HAS_MPL:
+            return
+        import numpy as np
+        t = true_values.detach().cpu().reshape(-1, num_nodes, 3).numpy()
+        p = (predicted_values.detach().cpu()
+             .reshape(-1, num_nodes, 3).numpy())
+        tm = np.linalg.norm(t, axis=-1)
+        pm = np.linalg.norm(p, axis=-1)
+        ncol = int(np.ceil(np.sqrt(num_nodes)))
+        nrow = int(np.ceil(num_nodes / ncol))
+        fig, axes = plt.subplots(nrow, ncol,
+                                 figsize=(3 * ncol, 3 * nrow),
+                                 squeeze=False)
+        flat = axes.flatten()
+        for s in range(num_nodes):
+            flat[s].scatter(tm[:, s], pm[:, s], s=4, alpha=0.5)
+            self.add_identity(flat[s], "k--", lw=1)
+            flat[s].set_title(f"node {s}", fontsize=8)
+        for s in range(num_nodes, len(flat)):
+            flat[s].axis("off")
+        fig.tight_layout()
+        suffix = f"_epoch{iepoch}" if iepoch is not None else ""
+        fig.savefig(os.path.join(
+            self.outdir, f"parity_pernode_{varname}{suffix}.png"),
+            dpi=120)
+        plt.close(fig)

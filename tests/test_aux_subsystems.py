@@ -257,8 +257,18 @@ def test_visualizer_plot_suite(tmp_path):
     viz.create_parity_plot_vector(torch.randn(30, 3),
                                   torch.randn(30, 3))
     viz.num_nodes_plot(nc)
+    viz.create_plot_global(t, p, output_names=["a", "b"])
+    viz.create_parity_plot_and_error_histogram_scalar(
+        "energy", t[0], p[0], iepoch=1)
+    viz.create_parity_plot_and_error_histogram_scalar(
+        "pernode", torch.randn(20, 6), torch.randn(20, 6))
+    viz.create_parity_plot_per_node_vector(
+        "forces", torch.randn(10, 4, 3), torch.randn(10, 4, 3), 4)
     out = tmp_path / "viz_test"
     for f in ["history.png", "scatter.png", "error_hist.png",
               "global_analysis.png", "error_hist_per_size.png",
-              "parity_forces.png", "num_nodes.png"]:
+              "parity_forces.png", "num_nodes.png",
+              "parity_global.png", "parity_hist_energy_epoch1.png",
+              "parity_hist_pernode.png",
+              "parity_pernode_forces.png"]:
         assert (out / f).exists(), f
