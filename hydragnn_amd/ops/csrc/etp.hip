@@ -91,6 +91,73 @@ __global__ void etp_general_kernel(
   }
 }
 
+// Register-accumulation variant (r2 default): the entry table is
+// sorted by output slot with per-output (start, count) ranges
+// (ETPTable.device_tensors), so each output accumulates in a REGISTER
+// and stores once — the r1 kernel's mo[q.w] += ... formed a serially
+// dependent LDS read-modify-write chain (consecutive entries hit the
+// same address; PMC: SQ busy only ~13% of wall time = latency-bound).
+// The LDS accumulator slice disappears too (smaller stride -> higher
+// occupancy).
+template <typename T>
+__global__ void etp_general_racc_kernel(
+    const T* __restrict__ A, const T* __restrict__ B,
+    const T* __restrict__ C, T* __restrict__ out,
+    const int4* __restrict__ entries,
+    const float* __restrict__ coefs,
+    const int2* __restrict__ o_ranges, int n_ent,
+    long NC, int nch, int da, int db, int dg, int do_,
+    const long* __restrict__ ai, const long* __restrict__ bi,
+    const long* __restrict__ ci) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  using ACC = typename acc_of<T>::type;
+  const int stride = ((da + db + dg) | 1);
+  ACC* slices = reinterpret_cast<ACC*>(smem);
+  int4* ent_lds = reinterpret_cast<int4*>(
+      smem + (size_t)blockDim.x * stride * sizeof(ACC));
+  float* coef_lds = reinterpret_cast<float*>(ent_lds + n_ent);
+  int2* rng_lds = reinterpret_cast<int2*>(coef_lds + n_ent);
+
+  for (int k = threadIdx.x; k < n_ent; k += blockDim.x) {
+    ent_lds[k] = entries[k];
+    coef_lds[k] = coefs[k];
+  }
+  for (int k = threadIdx.x; k < do_; k += blockDim.x)
+    rng_lds[k] = o_ranges[k];
+
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  ACC* my = slices + (size_t)threadIdx.x * stride;
+  ACC* ma = my;
+  ACC* mb = ma + da;
+  ACC* mc = mb + db;
+  if (i < NC) {
+    long e = i / nch;
+    int c = (int)(i - e * nch);
+    long ea = ai ? ai[e] : e;
+    long eb = bi ? bi[e] : e;
+    long ec = ci ? ci[e] : e;
+    const T* ap = A + (ea * nch + c) * da;
+    const T* bp = B + eb * db;
+    const T* cp = C + (ec * nch + c) * dg;
+    for (int k = 0; k < da; ++k) ma[k] = (ACC)ap[k];
+    for (int k = 0; k < db; ++k) mb[k] = (ACC)bp[k];
+    for (int k = 0; k < dg; ++k) mc[k] = (ACC)cp[k];
+  }
+  __syncthreads();
+  if (i >= NC) return;
+  T* op = out + i * do_;
+  for (int o = 0; o < do_; ++o) {
+    int s = rng_lds[o].x;
+    int cnt = rng_lds[o].y;
+    ACC acc = (ACC)0;
+    for (int k = s; k < s + cnt; ++k) {
+      int4 q = ent_lds[k];
+      acc += coef_lds[k] * ma[q.x] * mb[q.y] * mc[q.z];
+    }
+    op[o] = (T)acc;
+  }
+}
+
 // Occupancy variant of etp_general: A/B/C rows are PRIVATE per thread
 // and only ~1-3 cache lines each, so after first touch they are
 // L1-hot — staging them in LDS buys nothing but caps the block count
@@ -371,24 +438,49 @@ torch::Tensor etp_general(torch::Tensor A, torch::Tensor B, torch::Tensor C,
   size_t lds_full = (size_t)block * stride * accs + n_ent * 20;
   int stride_l1 = (int)do_ | 1;
   size_t lds_l1 = (size_t)block * stride_l1 * accs + n_ent * 20;
-  // Measured A/B (default bench, b1024): staged 32.9k g/s vs
-  // L1-operand 30.9k — the higher occupancy does NOT pay for the 3
-  // L1 loads + converts per entry, so staged is the default; the L1
-  // variant stays for shapes that exceed the staged LDS budget and
-  // for HYDRAGNN_ETP_L1=1 experiments.
-  bool use_l1 = lds_full > 150 * 1024;
-  const char* env = getenv("HYDRAGNN_ETP_L1");
-  if (env && env[0] == '0') use_l1 = false;
-  if (env && env[0] == '1') use_l1 = true;
-  size_t lds_bytes = use_l1 ? lds_l1 : lds_full;
+  // Variant choice (A/B'd on the default bench):
+  //  - racc (r2 default): register accumulation over the
+  //    output-sorted entry ranges — no LDS RMW dependency chain, no
+  //    output slice in LDS.
+  //  - staged (r1): full A/B/C/out slices in LDS.
+  //  - L1: operands from L1, only the accumulator in LDS.
+  // Fallback order by LDS budget; HYDRAGNN_ETP_VARIANT=staged|l1|racc
+  // overrides.
+  int stride_racc = (da + db + dg) | 1;
+  size_t lds_racc = (size_t)block * stride_racc * accs + n_ent * 20 +
+                    (size_t)do_ * 8;
+  int variant = lds_racc <= 150 * 1024 ? 2
+                : (lds_full <= 150 * 1024 ? 0 : 1);
+  const char* env = getenv("HYDRAGNN_ETP_VARIANT");
+  if (env) {
+    if (env[0] == 's') variant = 0;
+    else if (env[0] == 'l') variant = 1;
+    else if (env[0] == 'r') variant = 2;
+  }
+  size_t lds_bytes = variant == 2 ? lds_racc
+                     : (variant == 1 ? lds_l1 : lds_full);
   TORCH_CHECK(lds_bytes <= 150 * 1024, "etp LDS budget exceeded");
   long blocks = (NC + block - 1) / block;
+  auto orng = o_ranges.to(torch::kInt32).contiguous();
   // fp64 runs with double LDS slices (acc_of<double>); others stage fp32
   AT_DISPATCH_FLOATING_TYPES_AND2(
       at::ScalarType::BFloat16, at::ScalarType::Half, A.scalar_type(),
       "etp_general", [&] {
-        auto kern = use_l1 ? etp_general_l1_kernel<scalar_t>
-                           : etp_general_kernel<scalar_t>;
+        if (variant == 2) {
+          hipLaunchKernelGGL(
+              etp_general_racc_kernel<scalar_t>, dim3(blocks),
+              dim3(block), lds_bytes, etp_stream(),
+              A.data_ptr<scalar_t>(), B.data_ptr<scalar_t>(),
+              C.data_ptr<scalar_t>(), out.data_ptr<scalar_t>(),
+              reinterpret_cast<const int4*>(entries.data_ptr<int>()),
+              coefs.data_ptr<float>(),
+              reinterpret_cast<const int2*>(orng.data_ptr<int>()),
+              n_ent, NC, nch, da, db, dg, (int)do_,
+              idx_ptr(ai), idx_ptr(bi), idx_ptr(ci));
+          return;
+        }
+        auto kern = variant == 1 ? etp_general_l1_kernel<scalar_t>
+                                 : etp_general_kernel<scalar_t>;
         hipLaunchKernelGGL(
             kern, dim3(blocks), dim3(block),
             lds_bytes, etp_stream(),
