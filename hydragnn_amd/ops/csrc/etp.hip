@@ -66,44 +66,18 @@ __global__ void etp_general_kernel(
   ACC* mb = ma + da;
   ACC* mc = mb + db;
   ACC* mo = mc + dg;
-  const long base = (long)blockIdx.x * blockDim.x;
-  const long lim = NC - base < (long)blockDim.x ? NC - base
-                                                : (long)blockDim.x;
-  // A/C staging: when unindexed, the block's rows form ONE contiguous
-  // slab — stage cooperatively so each wave instruction reads 64
-  // consecutive elements (the per-thread row loop reads 64 addresses
-  // a row-stride apart: uncoalesced, and this kernel is latency-bound
-  // on exactly those loads — PMC SQ busy ~13%).
-  if (ai == nullptr) {
-    const T* gsrc = A + base * da;
-    for (long m = threadIdx.x; m < lim * da; m += blockDim.x) {
-      int t = (int)(m / da), k = (int)(m - (long)t * da);
-      slices[(size_t)t * stride + k] = (ACC)gsrc[m];
-    }
-  }
-  if (ci == nullptr) {
-    const T* gsrc = C + base * dg;
-    for (long m = threadIdx.x; m < lim * dg; m += blockDim.x) {
-      int t = (int)(m / dg), k = (int)(m - (long)t * dg);
-      slices[(size_t)t * stride + da + db + k] = (ACC)gsrc[m];
-    }
-  }
   if (i < NC) {
     long e = i / nch;
     int c = (int)(i - e * nch);
-    if (ai != nullptr) {
-      const T* ap = A + (ai[e] * nch + c) * da;
-      for (int k = 0; k < da; ++k) ma[k] = (ACC)ap[k];
-    }
-    if (ci != nullptr) {
-      const T* cp = C + (ci[e] * nch + c) * dg;
-      for (int k = 0; k < dg; ++k) mc[k] = (ACC)cp[k];
-    }
-    // B is per-edge: all 64 lanes of a wave read the same small row
-    // (broadcast from L1) — per-thread staging is already cheap
+    long ea = ai ? ai[e] : e;
     long eb = bi ? bi[e] : e;
+    long ec = ci ? ci[e] : e;
+    const T* ap = A + (ea * nch + c) * da;
     const T* bp = B + eb * db;
+    const T* cp = C + (ec * nch + c) * dg;
+    for (int k = 0; k < da; ++k) ma[k] = (ACC)ap[k];
     for (int k = 0; k < db; ++k) mb[k] = (ACC)bp[k];
+    for (int k = 0; k < dg; ++k) mc[k] = (ACC)cp[k];
     for (int k = 0; k < do_; ++k) mo[k] = 0.f;
   }
   __syncthreads();
@@ -112,15 +86,8 @@ __global__ void etp_general_kernel(
       int4 q = ent_lds[k];
       mo[q.w] += coef_lds[k] * ma[q.x] * mb[q.y] * mc[q.z];
     }
-  }
-  __syncthreads();
-  // cooperative coalesced store of the block's contiguous output slab
-  {
-    T* gdst = out + base * do_;
-    for (long m = threadIdx.x; m < lim * do_; m += blockDim.x) {
-      int t = (int)(m / do_), k = (int)(m - (long)t * do_);
-      gdst[m] = (T)slices[(size_t)t * stride + da + db + dg + k];
-    }
+    T* op = out + i * do_;
+    for (int k = 0; k < do_; ++k) op[k] = (T)mo[k];
   }
 }
 
