@@ -282,9 +282,16 @@ class RealAgnosticAttResidualInteractionBlock(
         # [E, 8] GEMM with K = num_edges — hipBLASLt runs it on
         # (M/16)x(N/16) workgroups, a serial-K crawl (measured
         # ~1.4 ms/step at b1024)
-        h = _SplitKLinearFn.apply(edge_radial, W1[:, :rd], lin1.bias)
-        h_s = _SplitKLinearFn.apply(down, W1[:, rd:rd + c], None)
-        h_d = _SplitKLinearFn.apply(down, W1[:, rd + c:], None)
+        if edge_radial.is_cuda:
+            h = _SplitKLinearFn.apply(edge_radial, W1[:, :rd],
+                                      lin1.bias)
+            h_s = _SplitKLinearFn.apply(down, W1[:, rd:rd + c], None)
+            h_d = _SplitKLinearFn.apply(down, W1[:, rd + c:], None)
+        else:
+            h = torch.nn.functional.linear(edge_radial, W1[:, :rd],
+                                           lin1.bias)
+            h_s = torch.nn.functional.linear(down, W1[:, rd:rd + c])
+            h_d = torch.nn.functional.linear(down, W1[:, rd + c:])
         h = h + gather(h_s, src, backward_csr=src_csr) \
             + gather(h_d, dst, backward_csr=dst_csr)
         return self.radial_mlp[1:](h)

@@ -21,7 +21,9 @@ def _splitk_weight_grad(x2d: torch.Tensor, g2d: torch.Tensor,
                         chunks: int) -> torch.Tensor:
     E = x2d.shape[0]
     S = min(chunks, max(1, E // 256))
-    if S <= 1:
+    if S <= 1 or not x2d.is_cuda:
+        # split-K only pays on the GPU (hipBLASLt's serial-K underfill);
+        # on CPU the chunked bmm is strictly slower than one mm
         return g2d.t() @ x2d
     pad = (S - E % S) % S
     if pad:
@@ -56,4 +58,9 @@ class _SplitKLinearFn(torch.autograd.Function):
 
 class SplitKLinear(nn.Linear):
     def forward(self, x):
+        if not x.is_cuda:
+            # plain autograd linear on CPU (no custom Function
+            # overhead; split-K is a GPU-shape fix)
+            return torch.nn.functional.linear(x, self.weight,
+                                              self.bias)
         return _SplitKLinearFn.apply(x, self.weight, self.bias)
