@@ -272,3 +272,44 @@ def test_visualizer_plot_suite(tmp_path):
               "parity_hist_pernode.png",
               "parity_pernode_forces.png"]:
         assert (out / f).exists(), f
+
+
+def test_xyz_to_graph_bond_perception():
+    """Own bond perception (the vendored-xyz2mol role): water, CO2 and
+    N2 get chemically correct graphs and bond orders."""
+    from hydragnn_amd.utils.descriptors_and_embeddings.xyz2graph import (
+        assign_bond_orders, perceive_bonds, xyz_to_graph)
+
+    # water: two O-H single bonds, no H-H bond
+    z = torch.tensor([8, 1, 1])
+    pos = torch.tensor([[0.0, 0.0, 0.0], [0.96, 0.0, 0.0],
+                        [-0.24, 0.93, 0.0]])
+    bonds, lengths = perceive_bonds(z, pos)
+    assert sorted(map(tuple, bonds.tolist())) == [(0, 1), (0, 2)]
+    orders, charges = assign_bond_orders(z, bonds, lengths)
+    assert orders.tolist() == [1, 1]
+    assert charges.abs().sum() == 0
+
+    # CO2: two C=O double bonds
+    z = torch.tensor([6, 8, 8])
+    pos = torch.tensor([[0.0, 0.0, 0.0], [1.16, 0.0, 0.0],
+                        [-1.16, 0.0, 0.0]])
+    bonds, lengths = perceive_bonds(z, pos)
+    orders, _ = assign_bond_orders(z, bonds, lengths)
+    assert sorted(orders.tolist()) == [2, 2]
+
+    # N2: triple bond
+    z = torch.tensor([7, 7])
+    pos = torch.tensor([[0.0, 0.0, 0.0], [1.10, 0.0, 0.0]])
+    bonds, lengths = perceive_bonds(z, pos)
+    orders, _ = assign_bond_orders(z, bonds, lengths)
+    assert orders.tolist() == [3]
+
+    # full pipeline -> trainable Data with dst-sorted edges
+    d = xyz_to_graph(torch.tensor([6, 1, 1, 1, 1]), torch.tensor(
+        [[0.0, 0, 0], [1.09, 0, 0], [-0.36, 1.03, 0],
+         [-0.36, -0.51, 0.89], [-0.36, -0.51, -0.89]]))
+    assert d.num_edges == 8  # 4 C-H bonds, both directions
+    dst = d.edge_index[1]
+    assert bool((dst[1:] >= dst[:-1]).all())
+    assert d.edge_attr.shape == (8, 2)
