@@ -48,7 +48,17 @@ class _SplitKLinearFn(torch.autograd.Function):
         x2d = x.reshape(-1, x.shape[-1]).to(g2d.dtype)
         gx = gw = gb = None
         if ctx.needs_input_grad[0]:
-            gx = (g2d @ weight.to(g2d.dtype)).view_as(x).to(x.dtype)
+            if g2d.is_cuda:
+                # gx = g2d @ W == linear(g2d, W^T): route through the
+                # split-K Function so the SECOND-order pass (force
+                # double backward differentiating this product by W)
+                # also gets a split-K weight grad — a plain mm here
+                # regenerates the serial-K [out, E] @ [E, in] GEMM
+                gx = _SplitKLinearFn.apply(
+                    g2d, weight.t().to(g2d.dtype), None
+                ).view_as(x).to(x.dtype)
+            else:
+                gx = (g2d @ weight.to(g2d.dtype)).view_as(x).to(x.dtype)
         if ctx.needs_input_grad[1]:
             gw = _splitk_weight_grad(x2d, g2d, 64).to(weight.dtype)
         if ctx.has_bias and ctx.needs_input_grad[2]:
