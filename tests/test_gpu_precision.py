@@ -96,3 +96,39 @@ def test_fp16_scaler_training_gpu():
         errs.append(float(err))
     assert all(e == e and e < 1e30 for e in errs), errs
     assert errs[-1] < errs[0], f"fp16 loss did not decrease: {errs}"
+
+
+def test_fused_adamw_kernel_matches_cpu_math():
+    """The single-kernel AdamW (fp32 and bf16-master variants) vs the
+    CPU flat-math fallback: same trajectory."""
+    from hydragnn_amd.ops.fused_adamw import FusedAdamW
+
+    for dtype in (torch.float32, torch.bfloat16):
+        torch.manual_seed(7)
+        m_gpu = torch.nn.Sequential(
+            torch.nn.Linear(8, 32), torch.nn.SiLU(),
+            torch.nn.Linear(32, 2)).to("cuda", dtype)
+        m_cpu = torch.nn.Sequential(
+            torch.nn.Linear(8, 32), torch.nn.SiLU(),
+            torch.nn.Linear(32, 2))
+        m_cpu.load_state_dict({k: v.float().cpu()
+                               for k, v in m_gpu.state_dict().items()})
+        m_cpu = m_cpu.to(dtype)
+        o_gpu = FusedAdamW(m_gpu.parameters(), lr=2e-3)
+        o_cpu = FusedAdamW(m_cpu.parameters(), lr=2e-3)
+        x = torch.randn(16, 8).to(dtype)
+        y = torch.randn(16, 2).to(dtype)
+        for _ in range(5):
+            o_gpu.zero_grad()
+            torch.nn.functional.mse_loss(
+                m_gpu(x.cuda()).float(), y.cuda().float()).backward()
+            o_gpu.step()
+            o_cpu.zero_grad()
+            torch.nn.functional.mse_loss(
+                m_cpu(x).float(), y.float()).backward()
+            o_cpu.step()
+        for pg, pc in zip(m_gpu.parameters(), m_cpu.parameters()):
+            tol = 1e-5 if dtype == torch.float32 else 2e-2
+            assert torch.allclose(pg.float().cpu(), pc.float(),
+                                  rtol=tol, atol=tol), \
+                (dtype, (pg.float().cpu() - pc.float()).abs().max())
