@@ -28,10 +28,12 @@ def main():
         train(timed, model, opt, 0, precision="bf16_pure")
         torch.cuda.synchronize()
     evs = prof.key_averages(group_by_input_shape=True)
+    import os as _os
+    kinds = _os.environ.get("DIAG_OPS", "mm,bmm,addmv,addmm,matmul,"
+                            "linear").split(",")
     rows = [(e.device_time_total, e.key, e.input_shapes, e.count)
             for e in evs
-            if any(k in e.key for k in ("mm", "bmm", "addmv", "addmm",
-                                        "matmul", "linear"))]
+            if any(k in e.key for k in kinds)]
     rows.sort(reverse=True)
     for t, k, shp, c in rows[:25]:
         print(f"{t/1000:9.2f} ms x{c:<4} {k:<18} {shp}")
