@@ -186,3 +186,59 @@ def test_bench_trainloop_cpu():
 
 # (2-rank torchrun coverage of the same path lives in
 # tests/test_distributed.py::test_bench_two_rank_gloo.)
+
+
+def _capture_agreement_worker(rank, world, port, q):
+    try:
+        os.environ.update(MASTER_ADDR="127.0.0.1",
+                          MASTER_PORT=str(port), RANK=str(rank),
+                          WORLD_SIZE=str(world), LOCAL_RANK=str(rank))
+        dist.init_process_group("gloo", rank=rank, world_size=world)
+        import hydragnn_amd.train.captured as cap
+        from hydragnn_amd.data import Data
+
+        class _FakeStepper:
+            def __init__(self, model, opt, batch, *a, **k):
+                # rank 1 "fails" capture; rank 0 succeeds
+                if dist.get_rank() == 1:
+                    raise RuntimeError("no capture on this rank")
+                self.opt = opt
+
+            def matches(self, data):
+                return True
+
+        orig = cap.CapturedTrainStep
+        cap.CapturedTrainStep = _FakeStepper
+        try:
+            model = torch.nn.Linear(2, 2)
+            opt = torch.optim.SGD(model.parameters(), lr=0.1)
+            batch = Data(pos=torch.zeros(3, 3),
+                         x=torch.zeros(3, 2),
+                         edge_index=torch.zeros(2, 0, dtype=torch.long))
+            stepper = cap.get_or_build_stepper(
+                model, opt, batch, None, None, torch.float32)
+        finally:
+            cap.CapturedTrainStep = orig
+        # one rank failed -> ALL ranks must fall back to eager
+        q.put((rank, stepper is None, ""))
+        dist.destroy_process_group()
+    except Exception as e:  # pragma: no cover
+        q.put((rank, False, repr(e)))
+
+
+def test_capture_rank_agreement_falls_back_everywhere():
+    """If capture fails on any rank, every rank must run eager —
+    a mixed fleet would mismatch collectives and hang the 8-GPU
+    scaling run."""
+    import multiprocessing as mp
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    ps = [ctx.Process(target=_capture_agreement_worker,
+                      args=(r, 2, 29681, q)) for r in range(2)]
+    for p in ps:
+        p.start()
+    results = [q.get(timeout=300) for _ in range(2)]
+    for p in ps:
+        p.join(timeout=60)
+    for rank, fell_back, info in results:
+        assert fell_back, f"rank {rank}: stepper not None ({info})"
