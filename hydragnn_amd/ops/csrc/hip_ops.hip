@@ -724,6 +724,10 @@ std::vector<torch::Tensor> radius_pairs_t(torch::Tensor pos,
                                           torch::Tensor gptr, double r,
                                           bool loop,
                                           c10::optional<torch::Tensor> shifts);
+std::vector<torch::Tensor> radius_pairs_cells(
+    torch::Tensor pos, torch::Tensor order, torch::Tensor cell_of,
+    torch::Tensor cell_start, long ncx, long ncy, long ncz, double r,
+    bool loop);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("etp_general", &etp_general, "fused ETP contraction (HIP)",
@@ -765,6 +769,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "single-kernel flat AdamW (HIP)");
   m.def("fused_adamw_bf16", &fused_adamw_bf16,
         "single-kernel flat AdamW, bf16 params + fp32 master (HIP)");
+  m.def("radius_pairs_cells", &radius_pairs_cells,
+        "cell-list radius pairs for large graphs (HIP)");
   m.def("radius_pairs_t", &radius_pairs_t,
         "tiled fp32/fp64 radius pairs, open or periodic (HIP)",
         pybind11::arg("pos"), pybind11::arg("batch"),

@@ -77,3 +77,31 @@ def test_radius_graph_pbc_gpu_1024_atoms():
     assert float(vec.norm(dim=-1).max()) <= 5.0 + 1e-9
     dst = ei[1]
     assert bool((dst[1:] >= dst[:-1]).all()), "dst-major order"
+
+
+@pytest.mark.parametrize("dtype", [torch.float32, torch.float64])
+def test_radius_graph_cell_list_matches_tiled(dtype):
+    """Large single graph: the cell-list path produces exactly the
+    tiled kernel's edge set (cell-list for 10k+ cells)."""
+    import os
+    torch.manual_seed(4)
+    n = 9000
+    pos = (torch.rand(n, 3, dtype=dtype) * 40.0).cuda()
+    ei_cells = radius_graph(pos, 2.5, max_num_neighbors=n)
+    os.environ["HYDRAGNN_CELL_LIST_MIN"] = str(10 ** 9)
+    try:
+        ei_tiled = radius_graph(pos, 2.5, max_num_neighbors=n)
+    finally:
+        os.environ.pop("HYDRAGNN_CELL_LIST_MIN", None)
+    assert _edge_set(ei_cells) == _edge_set(ei_tiled)
+    dst = ei_cells[1]
+    assert bool((dst[1:] >= dst[:-1]).all())
+
+
+def test_radius_graph_cell_list_cap():
+    torch.manual_seed(5)
+    n = 6000
+    pos = (torch.rand(n, 3) * 20.0).cuda()
+    ei = radius_graph(pos, 3.0, max_num_neighbors=12)
+    counts = torch.bincount(ei[1], minlength=n)
+    assert int(counts.max()) <= 12
