@@ -27,34 +27,6 @@ namespace {
 template <typename T> struct acc_of { using type = float; };
 template <> struct acc_of<double> { using type = double; };
 
-// Fast division by a small uniform constant d: row = (m * M) >> 25
-// with M = ceil(2^25 / d) — verified exact for the staging ranges
-// here (m < blockDim * dim <= 512 * 192, d <= 192).  Cooperative
-// staging needs row = m / d per element; the hardware idiv costs ~40
-// cycles and was THE regression in the first coalesced-staging
-// attempt.
-__device__ inline int fastdiv25(unsigned m, unsigned magic) {
-  return (int)(((unsigned long long)m * magic) >> 25);
-}
-
-// Cooperative coalesced stage of the block's contiguous slab
-// src[base*d .. (base+lim)*d) into per-thread LDS slices
-// slices[t*stride + off + k]: each wave instruction reads 64
-// CONSECUTIVE elements (one or two cache lines) instead of 64
-// row-strided 2-byte loads (measured 9x L1-transaction
-// amplification, etp_reduce at ~700 GB/s of an 8 TB/s roofline).
-template <typename T, typename ACC>
-__device__ inline void coop_stage(const T* __restrict__ gsrc,
-                                  ACC* __restrict__ slices, int stride,
-                                  int off, int d, long lim) {
-  unsigned magic = ((1u << 25) + (unsigned)d - 1) / (unsigned)d;
-  for (long m = threadIdx.x; m < lim * d; m += blockDim.x) {
-    int t = fastdiv25((unsigned)m, magic);
-    int k = (int)m - t * d;
-    slices[(size_t)t * stride + off + k] = (ACC)gsrc[m];
-  }
-}
-
 // out[i,c,o] = sum over entries(coef, a, b, g, o) of
 //              coef * A[i,c,a] * B[i,b] * C[i,c,g]
 //
@@ -94,31 +66,18 @@ __global__ void etp_general_kernel(
   ACC* mb = ma + da;
   ACC* mc = mb + db;
   ACC* mo = mc + dg;
-  const long base = (long)blockIdx.x * blockDim.x;
-  const long lim = NC - base < (long)blockDim.x ? NC - base
-                                                : (long)blockDim.x;
-  // A/C (and the output below): the block's rows form one contiguous
-  // slab when unindexed -> cooperative coalesced staging (the
-  // per-thread 2-byte row loads cost ~9x in L1 transactions)
-  if (ai == nullptr && lim > 0)
-    coop_stage<T, ACC>(A + base * da, slices, stride, 0, da, lim);
-  if (ci == nullptr && lim > 0)
-    coop_stage<T, ACC>(C + base * dg, slices, stride, da + db, dg,
-                       lim);
   if (i < NC) {
     long e = i / nch;
     int c = (int)(i - e * nch);
-    if (ai != nullptr) {
-      const T* ap = A + (ai[e] * nch + c) * da;
-      for (int k = 0; k < da; ++k) ma[k] = (ACC)ap[k];
-    }
-    if (ci != nullptr) {
-      const T* cp = C + (ci[e] * nch + c) * dg;
-      for (int k = 0; k < dg; ++k) mc[k] = (ACC)cp[k];
-    }
+    long ea = ai ? ai[e] : e;
     long eb = bi ? bi[e] : e;
-    const T* bp = B + eb * db;  // per-edge row: wave-broadcast reads
+    long ec = ci ? ci[e] : e;
+    const T* ap = A + (ea * nch + c) * da;
+    const T* bp = B + eb * db;
+    const T* cp = C + (ec * nch + c) * dg;
+    for (int k = 0; k < da; ++k) ma[k] = (ACC)ap[k];
     for (int k = 0; k < db; ++k) mb[k] = (ACC)bp[k];
+    for (int k = 0; k < dg; ++k) mc[k] = (ACC)cp[k];
     for (int k = 0; k < do_; ++k) mo[k] = 0.f;
   }
   __syncthreads();
@@ -127,17 +86,8 @@ __global__ void etp_general_kernel(
       int4 q = ent_lds[k];
       mo[q.w] += coef_lds[k] * ma[q.x] * mb[q.y] * mc[q.z];
     }
-  }
-  __syncthreads();
-  if (lim > 0) {
-    // cooperative coalesced store of the contiguous output slab
-    T* gdst = out + base * do_;
-    unsigned magic = ((1u << 25) + (unsigned)do_ - 1) / (unsigned)do_;
-    for (long m = threadIdx.x; m < lim * do_; m += blockDim.x) {
-      int t = fastdiv25((unsigned)m, magic);
-      int k = (int)m - t * do_;
-      gdst[m] = (T)slices[(size_t)t * stride + da + db + dg + k];
-    }
+    T* op = out + i * do_;
+    for (int k = 0; k < do_; ++k) op[k] = (T)mo[k];
   }
 }
 
@@ -410,33 +360,18 @@ __global__ void etp_reduce_idx_kernel(
   ACC* mc = ma + da;
   ACC* md = mc + dg;
   ACC* mb = md + do_;
-  const long base = (long)blockIdx.x * blockDim.x;
-  const long lim = NC - base < (long)blockDim.x ? NC - base
-                                                : (long)blockDim.x;
-  // cooperative coalesced staging of unindexed contiguous slabs
-  // (measured: per-thread row loads held this kernel at ~730 GB/s)
-  if (ai == nullptr && lim > 0)
-    coop_stage<T, ACC>(A + base * da, slices, stride, 0, da, lim);
-  if (ci == nullptr && lim > 0)
-    coop_stage<T, ACC>(C + base * dg, slices, stride, da, dg, lim);
-  if (di == nullptr && lim > 0)
-    coop_stage<T, ACC>(D + base * do_, slices, stride, da + dg, do_,
-                       lim);
   if (i < NC) {
     long e = i / nch;
     int c = (int)(i - e * nch);
-    if (ai != nullptr) {
-      const T* ap = A + (ai[e] * nch + c) * da;
-      for (int k = 0; k < da; ++k) ma[k] = (ACC)ap[k];
-    }
-    if (ci != nullptr) {
-      const T* cp = C + (ci[e] * nch + c) * dg;
-      for (int k = 0; k < dg; ++k) mc[k] = (ACC)cp[k];
-    }
-    if (di != nullptr) {
-      const T* dp = D + (di[e] * nch + c) * do_;
-      for (int k = 0; k < do_; ++k) md[k] = (ACC)dp[k];
-    }
+    long ea = ai ? ai[e] : e;
+    long ec = ci ? ci[e] : e;
+    long ed = di ? di[e] : e;
+    const T* ap = A + (ea * nch + c) * da;
+    const T* cp = C + (ec * nch + c) * dg;
+    const T* dp = D + (ed * nch + c) * do_;
+    for (int k = 0; k < da; ++k) ma[k] = (ACC)ap[k];
+    for (int k = 0; k < dg; ++k) mc[k] = (ACC)cp[k];
+    for (int k = 0; k < do_; ++k) md[k] = (ACC)dp[k];
     for (int k = 0; k < db; ++k) mb[k] = 0.f;
   }
   __syncthreads();
