@@ -82,7 +82,6 @@ def build_model_and_batch(device="cuda:0", local_batch=LOCAL_BATCH,
     from hydragnn_amd.models.create import create_model, resolve_precision
     from hydragnn_amd.train import get_autocast_and_scaler
     from hydragnn_amd.utils.datasets.synthetic import md17_shape_dataset
-    from hydragnn_amd.utils.optimizer import select_optimizer
 
     torch.manual_seed(seed)
     cfg = dict(MODEL_CONFIG)

@@ -15,9 +15,6 @@ over the flattened storages.
 
 from __future__ import annotations
 
-import math
-from typing import Iterable
-
 import torch
 
 from ._extension import get_extension, use_eager

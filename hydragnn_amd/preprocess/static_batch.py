@@ -22,7 +22,7 @@ kernels per batch; on MI355X we replay one hipGraph instead).
 from __future__ import annotations
 
 import math
-from typing import Optional, Sequence
+from typing import Sequence
 
 import torch
 

@@ -26,7 +26,7 @@ from __future__ import annotations
 
 import os
 from contextlib import nullcontext
-from typing import Callable, Dict, Optional
+from typing import Callable
 
 import torch
 

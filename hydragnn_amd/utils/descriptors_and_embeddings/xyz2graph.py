@@ -13,7 +13,7 @@ SMILES out of the perceived graph (``to_rdkit_mol``)."""
 
 from __future__ import annotations
 
-from typing import Dict, List, Optional, Tuple
+from typing import Dict, Optional, Tuple
 
 import torch
 
