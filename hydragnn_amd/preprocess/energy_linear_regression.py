@@ -41,12 +41,13 @@ def energy_linear_regression(dataset: Iterable,
         Atb += row * float(energy.flatten()[0])
     if distributed and dist.is_initialized() and \
             dist.get_world_size() > 1:
-        t = torch.from_numpy(AtA)
+        from ..utils.distributed import to_comm_device
+        t, _ = to_comm_device(torch.from_numpy(AtA))
         dist.all_reduce(t)
-        AtA = t.numpy()
-        t = torch.from_numpy(Atb)
+        AtA = t.cpu().numpy()
+        t, _ = to_comm_device(torch.from_numpy(Atb))
         dist.all_reduce(t)
-        Atb = t.numpy()
+        Atb = t.cpu().numpy()
     present = np.diag(AtA) > 0
     e = np.zeros(num_elements)
     if present.any():

@@ -71,7 +71,10 @@ def _gather_deg(dataset) -> torch.Tensor:
     for h in hists:
         out[: h.numel()] += h
     if torch.distributed.is_initialized():
-        torch.distributed.all_reduce(out)
+        from ..distributed import to_comm_device
+        t, moved = to_comm_device(out)
+        torch.distributed.all_reduce(t)
+        out = t.cpu() if moved else t
     return out
 
 
@@ -83,6 +86,8 @@ def _calculate_avg_deg(dataset) -> float:
         num_nodes += data.num_nodes
     t = torch.tensor([num_edges, num_nodes], dtype=torch.float64)
     if torch.distributed.is_initialized():
+        from ..distributed import to_comm_device
+        t, _ = to_comm_device(t)
         torch.distributed.all_reduce(t)
     return float(t[0] / t[1].clamp(min=1))
 

@@ -124,12 +124,16 @@ class AbstractRawDataset(AbstractBaseDataset):
                 gmin = torch.minimum(gmin, d.y.view(-1))
                 gmax = torch.maximum(gmax, d.y.view(-1))
         if self.dist and dist.is_initialized():
+            from ..distributed import to_comm_device
             for t, op in ((nmin, dist.ReduceOp.MIN),
                           (nmax, dist.ReduceOp.MAX),
                           (gmin, dist.ReduceOp.MIN),
                           (gmax, dist.ReduceOp.MAX)):
                 if t.numel():
-                    dist.all_reduce(t, op=op)
+                    tt, moved = to_comm_device(t)
+                    dist.all_reduce(tt, op=op)
+                    if moved:
+                        t.copy_(tt.cpu())
         for d in self.dataset:
             if num_node_f:
                 span = (nmax - nmin).clamp(min=1e-12)
@@ -160,7 +164,10 @@ class AbstractRawDataset(AbstractBaseDataset):
                     d.pos, d.edge_index, d.get("edge_shifts"))
                 max_len = torch.maximum(max_len, lengths.max().view(1))
         if self.dist and dist.is_initialized():
+            from ..distributed import to_comm_device
+            max_len, _ = to_comm_device(max_len)
             dist.all_reduce(max_len, op=dist.ReduceOp.MAX)
+            max_len = max_len.cpu()
         scale = float(max_len.clamp(min=1e-12))
         for d in self.dataset:
             add_edge_lengths(d, max_length=scale)

@@ -253,18 +253,20 @@ def allgatherv_numpy(array, comm=None):
     import torch.distributed as dist
     if not (dist.is_available() and dist.is_initialized()):
         return np.asarray(array)
+    from ..distributed import to_comm_device
     t = torch.from_numpy(np.ascontiguousarray(array))
-    n = torch.tensor([t.numel()], dtype=torch.long)
-    sizes = [torch.zeros(1, dtype=torch.long)
+    n, _ = to_comm_device(torch.tensor([t.numel()], dtype=torch.long))
+    sizes = [torch.zeros_like(n)
              for _ in range(dist.get_world_size())]
     dist.all_gather(sizes, n)
     maxn = int(max(s.item() for s in sizes))
     pad = torch.zeros(maxn, dtype=t.dtype)
     pad[:t.numel()] = t
-    outs = [torch.zeros(maxn, dtype=t.dtype)
+    pad, _ = to_comm_device(pad)
+    outs = [torch.zeros_like(pad)
             for _ in range(dist.get_world_size())]
     dist.all_gather(outs, pad)
-    return np.concatenate([o[:int(s.item())].numpy()
+    return np.concatenate([o[:int(s.item())].cpu().numpy()
                            for o, s in zip(outs, sizes)])
 
 
@@ -298,9 +300,10 @@ class ShardedDistDataset(AbstractBaseDataset):
         self._local = [_pickle.dumps(s) for s in local_samples]
         counts = [len(self._local)]
         if self.world > 1:
-            t = torch.tensor(counts, dtype=torch.long)
-            sizes = [torch.zeros(1, dtype=torch.long)
-                     for _ in range(self.world)]
+            from ..distributed import to_comm_device
+            t, _ = to_comm_device(torch.tensor(counts,
+                                               dtype=torch.long))
+            sizes = [torch.zeros_like(t) for _ in range(self.world)]
             dist.all_gather(sizes, t)
             counts = [int(s) for s in sizes]
             # per-owner request/reply groups: the rank-k service thread

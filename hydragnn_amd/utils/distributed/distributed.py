@@ -118,13 +118,29 @@ def nsplit(lst, n: int):
             for i in range(n))
 
 
+def to_comm_device(t: torch.Tensor) -> Tuple[torch.Tensor, bool]:
+    """Move a tensor onto the device the default process group
+    communicates on: RCCL ("nccl") groups reject CPU tensors, so
+    preprocessing-time collectives over CPU tensors (degree stats,
+    normalization, energy regression) hop through the GPU.  Returns
+    (tensor, was_moved)."""
+    if dist.is_initialized() and dist.get_backend() == "nccl" \
+            and not t.is_cuda:
+        return t.to(get_device()), True
+    return t, False
+
+
 def comm_reduce(value: torch.Tensor, op: str = "sum") -> torch.Tensor:
     if not dist.is_initialized():
         return value
     ops = {"sum": dist.ReduceOp.SUM, "max": dist.ReduceOp.MAX,
            "min": dist.ReduceOp.MIN}
-    dist.all_reduce(value, op=ops[op])
-    return value
+    t, moved = to_comm_device(value)
+    dist.all_reduce(t, op=ops[op])
+    if moved:
+        value.copy_(t.to(value.device))
+        return value
+    return t
 
 
 # ---------------------------------------------------------------------------
@@ -290,7 +306,10 @@ def check_remaining_time(start_time, epoch_time, broadcast: bool = True
         except Exception:
             pass
     if broadcast and dist.is_initialized():
+        should_stop, moved = to_comm_device(should_stop)
         dist.broadcast(should_stop, src=0)
+        if moved:
+            should_stop = should_stop.cpu()
     return bool(should_stop.item())
 
 

@@ -34,6 +34,8 @@ class Timer:
     def reduce(self):
         t = torch.tensor([self.elapsed])
         if dist.is_initialized() and dist.get_world_size() > 1:
+            from ..distributed import to_comm_device
+            t, _ = to_comm_device(t)
             tmin, tmax, tsum = t.clone(), t.clone(), t.clone()
             dist.all_reduce(tmin, op=dist.ReduceOp.MIN)
             dist.all_reduce(tmax, op=dist.ReduceOp.MAX)
