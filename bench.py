@@ -68,7 +68,7 @@ MODEL_CONFIG = {
     "distance_transform": None,
 }
 
-LOCAL_BATCH = int(os.environ.get("HYDRAGNN_BENCH_BATCH", "4096"))
+LOCAL_BATCH = int(os.environ.get("HYDRAGNN_BENCH_BATCH", "8192"))
 # pure-bf16 training: bf16 params/activations (no autocast cast
 # traffic) + fp32 master weights in FusedAdamW — reported dtype bf16
 PRECISION = "bf16_pure"

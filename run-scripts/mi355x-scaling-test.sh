@@ -11,7 +11,7 @@ set -euo pipefail
 MODE=${1:-weak}
 STEPS=${2:-20}
 WARMUP=${3:-5}
-GLOBAL_BATCH=${HYDRAGNN_BENCH_BATCH:-4096}
+GLOBAL_BATCH=${HYDRAGNN_BENCH_BATCH:-8192}
 
 export HSA_ENABLE_IPC_MODE_LEGACY=0
 export MASTER_ADDR=127.0.0.1
