@@ -239,9 +239,14 @@ def train(loader, model, opt, verbosity, precision="fp32", profiler=None):
         if (scaler is None and stepper is None and not ds_engine
                 and capture_enabled(data)):
             dev_data = move_batch_to_device(data, param_dtype)
+            # capture the UNWRAPPED module's forward: DDP's python
+            # pre/post-forward stays entirely out of the graph (its
+            # reducer is unused on this path — FlatGradSync owns the
+            # gradient sync), removing the main capture-failure risk
+            # at world > 1
             stepper = get_or_build_stepper(
                 model, opt, dev_data, autocast,
-                lambda b: _compute_loss(model, b, use_ia,
+                lambda b: _compute_loss(_unwrap(model), b, use_ia,
                                         create_graph=True),
                 param_dtype)
             if stepper is not None:
