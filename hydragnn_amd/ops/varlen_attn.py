@@ -26,6 +26,34 @@ MAX_SEG = None  # r2: K/V tiling removed the segment cap
 MAX_DH = 64
 
 
+def torch_varlen_attention_chunked(q, k, v, batch, chunk=512):
+    """Per-graph exact attention in query chunks: O(chunk * N_g)
+    transient memory instead of the dense [B, maxN, maxN] logits —
+    the recompute backward for LARGE segments (the dense reference
+    would materialize maxN^2 per graph)."""
+    N, H, dh = q.shape
+    out = torch.empty_like(q)
+    scale = 1.0 / math.sqrt(dh)
+    n_graphs = int(batch.max()) + 1 if batch.numel() else 0
+    counts = torch.bincount(batch, minlength=n_graphs)
+    starts = torch.zeros(n_graphs, dtype=torch.long,
+                         device=batch.device)
+    if n_graphs > 1:
+        starts[1:] = counts.cumsum(0)[:-1]
+    for g in range(n_graphs):
+        lo, n = int(starts[g]), int(counts[g])
+        if n == 0:
+            continue
+        kg = k[lo:lo + n].transpose(0, 1)          # [H, n, dh]
+        vg = v[lo:lo + n].transpose(0, 1)
+        for c0 in range(0, n, chunk):
+            qg = q[lo + c0:lo + min(c0 + chunk, n)].transpose(0, 1)
+            logits = qg @ kg.transpose(-1, -2) * scale
+            o = torch.softmax(logits, dim=-1) @ vg  # [H, c, dh]
+            out[lo + c0:lo + c0 + o.shape[1]] = o.transpose(0, 1)
+    return out
+
+
 def torch_varlen_attention(q: torch.Tensor, k: torch.Tensor,
                            v: torch.Tensor, batch: torch.Tensor):
     """Reference path: [N, H, dh] q/k/v + per-node graph index ->
@@ -61,8 +89,15 @@ class _VarlenAttn(torch.autograd.Function):
         # graph stays connected for force-style double backward
         q, k, v, batch = ctx.saved_tensors
         need = ctx.needs_input_grad[:3]
+        max_seg = int(torch.bincount(batch).max()) \
+            if batch.numel() else 0
         with torch.enable_grad():
-            out = torch_varlen_attention(q, k, v, batch)
+            if max_seg > 1024:
+                # dense recompute would materialize maxN^2 logits per
+                # graph; chunked exact path keeps memory bounded
+                out = torch_varlen_attention_chunked(q, k, v, batch)
+            else:
+                out = torch_varlen_attention(q, k, v, batch)
             inputs = [t for t, n in zip((q, k, v), need) if n]
             grads = iter(torch.autograd.grad(
                 out, inputs, g, create_graph=torch.is_grad_enabled(),
