@@ -54,6 +54,11 @@ class WallTimer(TracerBackend):
                 tot = sum(hist)
                 f.write(f"{name:<24}{len(hist):>8}{tot:>14.6f}"
                         f"{tot / len(hist):>14.6f}{max(hist):>14.6f}\n")
+        # per-call history (the reference's gp_full.p<rank>), capped
+        with open(os.path.join(path, f"gp_full.p{rank}"), "w") as f:
+            for name, hist in sorted(self.history.items()):
+                for i, dt in enumerate(hist[-10000:]):
+                    f.write(f"{name} {i} {dt:.6f}\n")
 
 
 class RocmEnergyTracer(TracerBackend):
