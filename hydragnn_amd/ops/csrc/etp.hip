@@ -395,6 +395,66 @@ __global__ void etp_reduce_idx_kernel(
   }
 }
 
+// L1-operand reduce variant: only the small db-accumulator lives in
+// LDS; A/C/D rows are read straight through L1 (each row is 1-2 hot
+// cache lines).  LDS per thread drops from (da+dg+do+db) to db words
+// -> ~6x more waves per CU than the staged variant; the staged reduce
+// measured only ~700-950 GB/s of the 8 TB/s roofline (latency-bound
+// at 1-2 workgroups/CU).
+template <typename T>
+__global__ void etp_reduce_l1_kernel(
+    const T* __restrict__ A, const T* __restrict__ C,
+    const T* __restrict__ D,
+    typename acc_of<T>::type* __restrict__ out,
+    const int4* __restrict__ entries,
+    const float* __restrict__ coefs, int n_ent,
+    long E, int nch, int da, int db, int dg, int do_,
+    const long* __restrict__ ai, const long* __restrict__ ci,
+    const long* __restrict__ di) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  using ACC = typename acc_of<T>::type;
+  const int stride = db | 1;
+  ACC* slices = reinterpret_cast<ACC*>(smem);
+  int4* ent_lds = reinterpret_cast<int4*>(
+      smem + (size_t)blockDim.x * stride * sizeof(ACC));
+  float* coef_lds = reinterpret_cast<float*>(ent_lds + n_ent);
+  for (int k = threadIdx.x; k < n_ent; k += blockDim.x) {
+    ent_lds[k] = entries[k];
+    coef_lds[k] = coefs[k];
+  }
+  long NC = E * nch;
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  ACC* mb = slices + (size_t)threadIdx.x * stride;
+  __syncthreads();
+  long e = i / nch;
+  if (i < NC) {
+    int c = (int)(i - e * nch);
+    long ea = ai ? ai[e] : e;
+    long ec = ci ? ci[e] : e;
+    long ed = di ? di[e] : e;
+    const T* ap = A + (ea * nch + c) * da;
+    const T* cp = C + (ec * nch + c) * dg;
+    const T* dp = D + (ed * nch + c) * do_;
+    for (int k = 0; k < db; ++k) mb[k] = 0.f;
+    for (int k = 0; k < n_ent; ++k) {
+      int4 q = ent_lds[k];
+      mb[q.y] += coef_lds[k] * (ACC)ap[q.x] * (ACC)cp[q.z] *
+                 (ACC)dp[q.w];
+    }
+  }
+  if (nch % 64 == 0) {
+    for (int b = 0; b < db; ++b) {
+      ACC v = (i < NC) ? mb[b] : (ACC)0;
+      for (int off = 32; off >= 1; off >>= 1)
+        v += __shfl_down(v, off, 64);
+      if ((threadIdx.x % 64) == 0 && i < NC)
+        atomicAdd(&out[e * db + b], v);
+    }
+  } else if (i < NC) {
+    for (int b = 0; b < db; ++b) atomicAdd(&out[e * db + b], mb[b]);
+  }
+}
+
 }  // namespace
 
 static hipStream_t etp_stream() {
@@ -568,13 +628,21 @@ torch::Tensor etp_reduce(torch::Tensor A, torch::Tensor C, torch::Tensor D,
   int stride = (da + dg + do_ + (int)db) | 1;
   size_t accs = A.scalar_type() == at::ScalarType::Double ? 8 : 4;
   size_t lds_bytes = (size_t)block * stride * accs + n_ent * 20;
+  // L1-operand variant (accumulator-only LDS, ~6x occupancy) unless
+  // HYDRAGNN_ETP_REDUCE_VARIANT=staged
+  const char* rv = getenv("HYDRAGNN_ETP_REDUCE_VARIANT");
+  bool use_l1 = !(rv && rv[0] == 's');
+  if (use_l1)
+    lds_bytes = (size_t)block * ((db | 1)) * accs + n_ent * 20;
   TORCH_CHECK(lds_bytes <= 150 * 1024, "etp_reduce LDS budget exceeded");
   long blocks = (E * nch + block - 1) / block;
   AT_DISPATCH_FLOATING_TYPES_AND2(
       at::ScalarType::BFloat16, at::ScalarType::Half, A.scalar_type(),
       "etp_reduce", [&] {
+        auto kern = use_l1 ? etp_reduce_l1_kernel<scalar_t>
+                           : etp_reduce_idx_kernel<scalar_t>;
         hipLaunchKernelGGL(
-            etp_reduce_idx_kernel<scalar_t>, dim3(blocks), dim3(block),
+            kern, dim3(blocks), dim3(block),
             lds_bytes, etp_stream(), A.data_ptr<scalar_t>(),
             C.data_ptr<scalar_t>(),
             D.data_ptr<scalar_t>(), out.data_ptr<typename acc_of<scalar_t>::type>(),
