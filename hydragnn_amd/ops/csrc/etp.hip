@@ -628,10 +628,14 @@ torch::Tensor etp_reduce(torch::Tensor A, torch::Tensor C, torch::Tensor D,
   int stride = (da + dg + do_ + (int)db) | 1;
   size_t accs = A.scalar_type() == at::ScalarType::Double ? 8 : 4;
   size_t lds_bytes = (size_t)block * stride * accs + n_ent * 20;
-  // L1-operand variant (accumulator-only LDS, ~6x occupancy) unless
-  // HYDRAGNN_ETP_REDUCE_VARIANT=staged
+  // staged is the measured default (963 GB/s vs the L1-operand
+  // variant's 381: repeated per-entry L1 hits lose to LDS reads);
+  // HYDRAGNN_ETP_REDUCE_VARIANT=l1 selects the low-LDS variant for
+  // over-budget shapes
   const char* rv = getenv("HYDRAGNN_ETP_REDUCE_VARIANT");
-  bool use_l1 = !(rv && rv[0] == 's');
+  size_t lds_l1 = (size_t)block * ((db | 1)) * accs + n_ent * 20;
+  bool use_l1 = (rv && rv[0] == 'l') ||
+                (lds_bytes > 150 * 1024 && lds_l1 <= 150 * 1024);
   if (use_l1)
     lds_bytes = (size_t)block * ((db | 1)) * accs + n_ent * 20;
   TORCH_CHECK(lds_bytes <= 150 * 1024, "etp_reduce LDS budget exceeded");
