@@ -91,7 +91,7 @@ __global__ void etp_general_kernel(
   }
 }
 
-// Register-accumulation variant (r2 default): the entry table is
+// Register-accumulation variant (LDS-pressure fallback): the entry table is
 // sorted by output slot with per-output (start, count) ranges
 // (ETPTable.device_tensors), so each output accumulates in a REGISTER
 // and stores once — the r1 kernel's mo[q.w] += ... formed a serially
