@@ -64,10 +64,40 @@ def create_dataloaders(trainset, valset, testset, batch_size: int,
 
     use_custom = os.getenv("HYDRAGNN_CUSTOM_DATALOADER", "0") == "1"
 
+    # Training.Batching.mode == "static_shape": fixed-capacity padded
+    # batches (preprocess/static_batch.py) so the train loop's
+    # hipGraph-captured step engages — caps computed once over the
+    # training set (worst case, shuffle-safe)
+    static_coll = None
+    if batching is not None and batching.get("mode") == "static_shape":
+        from .static_batch import StaticShapeCollater, compute_static_caps
+        node_cap, edge_cap = compute_static_caps(
+            trainset, batch_size,
+            sequential=bool(batching.get("sequential", False)))
+        static_coll = StaticShapeCollater(
+            node_cap, edge_cap,
+            pad_spacing=float(batching.get("pad_spacing", 30.0)))
+
     def make(ds, shuffle):
         if ds is None or len(ds) == 0:
             return TorchDataLoader([], batch_size=batch_size,
                                    collate_fn=_collate)
+        if static_coll is not None and shuffle:
+            # padded static batches for the TRAIN loader only (eval
+            # loaders keep exact batches)
+            if use_dist:
+                sampler = DistributedSampler(
+                    ds, shuffle=sampler_shuffle)
+                return TorchDataLoader(
+                    ds, batch_size=batch_size, sampler=sampler,
+                    collate_fn=static_coll, num_workers=num_workers,
+                    drop_last=True,
+                    pin_memory=torch.cuda.is_available())
+            return TorchDataLoader(
+                ds, batch_size=batch_size, shuffle=sampler_shuffle,
+                collate_fn=static_coll, num_workers=num_workers,
+                drop_last=True,
+                pin_memory=torch.cuda.is_available())
         if use_custom:
             from .dataloader import HydraDataLoader
             bs = None

@@ -242,3 +242,34 @@ def test_capture_rank_agreement_falls_back_everywhere():
         p.join(timeout=60)
     for rank, fell_back, info in results:
         assert fell_back, f"rank {rank}: stepper not None ({info})"
+
+
+def test_config_driven_static_shape_batching():
+    """Training.Batching.mode == 'static_shape' produces fixed-shape
+    padded train batches through create_dataloaders, and train() runs
+    over them (the config path to the captured step)."""
+    from hydragnn_amd.preprocess import create_dataloaders
+    from hydragnn_amd.train import train
+
+    torch.manual_seed(6)
+    ds = md17_shape_dataset(num_samples=16)
+    model = _small_mace(ds)
+    config = {"NeuralNetwork": {"Training": {
+        "batch_size": 4,
+        "Batching": {"mode": "static_shape", "pad_spacing": 28.0}}}}
+    tr_loader, val_loader, te_loader = create_dataloaders(
+        ds, ds, ds, 4, config=config)
+    shapes = None
+    for b in tr_loader:
+        assert bool(b.get("static_shape_"))
+        s = {k: tuple(v.shape) for k, v in b.items()
+             if torch.is_tensor(v)}
+        shapes = shapes or s
+        assert s == shapes
+    # eval loaders stay exact (no pad graph)
+    for b in val_loader:
+        assert b.get("static_shape_") is None
+        break
+    opt = torch.optim.AdamW(model.parameters(), lr=1e-3)
+    err, _ = train(tr_loader, model, opt, 0)
+    assert torch.isfinite(err).all()
