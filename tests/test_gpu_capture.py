@@ -101,3 +101,33 @@ def test_captured_second_epoch_reuses_graph():
     import gc
     gc.collect()
     torch.cuda.empty_cache()
+
+
+@needs_gpu
+def test_config_driven_capture_engages():
+    """Training.Batching.mode='static_shape' through create_dataloaders
+    engages the hipGraph-captured step on GPU."""
+    from hydragnn_amd.preprocess import create_dataloaders
+    from hydragnn_amd.train import train
+    from hydragnn_amd.utils.datasets.synthetic import (
+        md17_shape_dataset_fast)
+    import bench as bench_mod
+
+    torch.manual_seed(9)
+    model = bench_mod.build_model("cuda:0", precision="bf16_pure",
+                                  seed=9)
+    from hydragnn_amd.ops.fused_adamw import FusedAdamW
+    opt = FusedAdamW(model.parameters(), lr=1e-3)
+    ds = md17_shape_dataset_fast(48, seed=21)
+    config = {"NeuralNetwork": {"Training": {
+        "batch_size": 16,
+        "Batching": {"mode": "static_shape", "pad_spacing": 28.0}}}}
+    loaders = create_dataloaders(ds, ds, ds, 16, config=config)
+    err, _ = train(loaders[0], model, opt, 0, precision="bf16_pure")
+    assert torch.isfinite(err).all()
+    stepper = getattr(model, "_hip_captured_step", None)
+    assert stepper not in (None, False), "capture did not engage"
+    model._hip_captured_step = None
+    import gc
+    gc.collect()
+    torch.cuda.empty_cache()
