@@ -27,10 +27,30 @@ def main():
     device = "cuda:0" if torch.cuda.is_available() else "cpu"
     precision = os.environ.get("CONV_PRECISION", "bf16")
     epochs = int(os.environ.get("CONV_EPOCHS", "40"))
+    mpnn = os.environ.get("CONV_MODEL", "MACE")
     n_train, n_val = 512, 128
 
     torch.manual_seed(11)
-    model = B.build_model(device, precision=precision, seed=11)
+    if mpnn == "MACE":
+        model = B.build_model(device, precision=precision, seed=11)
+    else:
+        from hydragnn_amd.models.create import (create_model,
+                                                resolve_precision)
+        extra = {}
+        if mpnn == "SchNet":
+            extra = dict(num_gaussians=32, num_filters=64)
+        model = create_model(
+            mpnn_type=mpnn, input_dim=1, hidden_dim=64,
+            output_dim=[1], output_type=["node"],
+            output_heads=B.MODEL_CONFIG["output_heads"],
+            activation_function="silu", loss_function_type="mse",
+            task_weights=[1.0], num_conv_layers=3, num_nodes=21,
+            max_neighbours=30, radius=7.0,
+            enable_interatomic_potential=True, energy_weight=1.0,
+            energy_peratom_weight=1.0, force_weight=100.0,
+            equivariance=True, use_gpu=False, **extra)
+        _, pd, _ = resolve_precision(precision)
+        model = model.to(device=device, dtype=pd)
     opt = torch.optim.AdamW(model.parameters(), lr=2e-3, foreach=True)
     sched = torch.optim.lr_scheduler.CosineAnnealingLR(opt, epochs)
 
@@ -77,7 +97,7 @@ def main():
         print(json.dumps(rec), flush=True)
 
     out = {
-        "config": "MACE-MLIP bf16, MD17-shape synthetic (LJ targets), "
+        "config": f"{mpnn}-MLIP {precision}, MD17-shape synthetic (LJ), "
                   "512 train / 128 val, batch 32, AdamW 2e-3 cosine",
         "device": torch.cuda.get_device_name(0) if device != "cpu"
                   else "cpu",
@@ -89,7 +109,9 @@ def main():
                           / max(traj[-1]["val_force_mae"], 1e-12),
     }
     os.makedirs("gpurun_out", exist_ok=True)
-    with open("gpurun_out/convergence_mace_md17.json", "w") as f:
+    out_name = (f"gpurun_out/convergence_{mpnn.lower()}"
+                "_md17.json")
+    with open(out_name, "w") as f:
         json.dump(out, f, indent=1)
     print(f"loss drop x{out['loss_drop']:.1f}, "
           f"force MAE drop x{out['force_mae_drop']:.1f}")
