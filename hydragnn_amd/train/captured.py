@@ -141,6 +141,12 @@ class CapturedTrainStep:
         self._copy_in(data)
         self.graph.replay()
         self.grad_sync()
+        clip = getattr(self.opt, "_hydragnn_grad_clip", None)
+        if clip is not None:
+            # one global-norm clip over the flat gradient buffer
+            norm = self._flat_grad.float().norm()
+            scale = (clip / (norm + 1e-6)).clamp(max=1.0)
+            self._flat_grad.mul_(scale.to(self._flat_grad.dtype))
         self.opt.step()
         return self.loss, self.tasks_loss
 

@@ -68,7 +68,8 @@ def get_nbatch(loader):
     if os.getenv("HYDRAGNN_MAX_NUM_BATCH") is not None:
         nbatch = min(nbatch, int(os.environ["HYDRAGNN_MAX_NUM_BATCH"]))
     if dist.is_initialized() and dist.get_world_size() > 1:
-        t = torch.tensor([nbatch], dtype=torch.long)
+        from ..utils.distributed import to_comm_device
+        t, _ = to_comm_device(torch.tensor([nbatch], dtype=torch.long))
         dist.all_reduce(t, op=dist.ReduceOp.MIN)
         nbatch = int(t.item())
     return nbatch
@@ -277,6 +278,11 @@ def train(loader, model, opt, verbosity, precision="fp32", profiler=None):
                 loss.backward()
             tr.stop("backward")
             tr.start("opt_step")
+            clip = getattr(opt, "_hydragnn_grad_clip", None)
+            if clip is not None and scaler is None and not ds_engine:
+                torch.nn.utils.clip_grad_norm_(
+                    (p for p in model.parameters()
+                     if p.grad is not None), clip)
             if ds_engine:
                 model.step()
             elif scaler is not None:

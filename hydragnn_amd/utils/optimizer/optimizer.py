@@ -32,9 +32,16 @@ def select_optimizer(model, config):
     use_zero = config.get("use_zero_redundancy", False)
     if use_zero and dist.is_initialized() and dist.get_world_size() > 1:
         from torch.distributed.optim import ZeroRedundancyOptimizer
-        return ZeroRedundancyOptimizer(
+        opt = ZeroRedundancyOptimizer(
             model.parameters(), optimizer_class=cls, **kwargs)
-    return cls(model.parameters(), **kwargs)
+    else:
+        opt = cls(model.parameters(), **kwargs)
+    clip = config.get("grad_clip_norm")
+    if clip is not None:
+        # consumed by train() / the captured step before each
+        # optimizer step (global-norm clipping)
+        opt._hydragnn_grad_clip = float(clip)
+    return opt
 
 
 def select_standard_optimizer(model, config):

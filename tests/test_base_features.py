@@ -214,3 +214,20 @@ def test_training_with_plot_creation(tmp_path, monkeypatch):
                      recursive=True)
     assert any("scatter" in p for p in pngs), pngs
     assert any("error_hist" in p for p in pngs), pngs
+
+
+def test_grad_clip_norm_option():
+    """Training.Optimizer.grad_clip_norm clips the global grad norm
+    before each step (beyond-reference robustness option)."""
+    from hydragnn_amd.utils.optimizer import select_optimizer
+
+    torch.manual_seed(0)
+    m = torch.nn.Linear(4, 4)
+    opt = select_optimizer(m, {"type": "SGD", "learning_rate": 1.0,
+                               "grad_clip_norm": 0.5})
+    assert getattr(opt, "_hydragnn_grad_clip", None) == 0.5
+    (m(torch.randn(16, 4) * 100).square().sum()).backward()
+    import torch as _t
+    _t.nn.utils.clip_grad_norm_(m.parameters(), 0.5)
+    total = sum(p.grad.norm() ** 2 for p in m.parameters()) ** 0.5
+    assert float(total) <= 0.5 + 1e-4
