@@ -285,7 +285,9 @@ def main():
     use_cuda = torch.cuda.is_available()
     if world_size > 1:
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
-        dist.init_process_group("nccl" if use_cuda else "gloo")
+        backend = os.environ.get("HYDRAGNN_BACKEND") or (
+            "nccl" if use_cuda else "gloo")
+        dist.init_process_group(backend)
         if use_cuda:
             torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", "0")))
     device = (f"cuda:{int(os.environ.get('LOCAL_RANK', '0'))}"
