@@ -289,8 +289,10 @@ def main():
             "nccl" if use_cuda else "gloo")
         dist.init_process_group(backend)
         if use_cuda:
-            torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", "0")))
-    device = (f"cuda:{int(os.environ.get('LOCAL_RANK', '0'))}"
+            torch.cuda.set_device(
+                int(os.environ.get("LOCAL_RANK", "0"))
+                % torch.cuda.device_count())
+    device = (f"cuda:{int(os.environ.get('LOCAL_RANK', '0')) % torch.cuda.device_count()}"
               if use_cuda else "cpu")
 
     # The measured product is the real train loop: framework DDP
