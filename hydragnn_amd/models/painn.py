@@ -78,7 +78,10 @@ class PainnUpdate(nn.Module):
     def forward(self, node_scalar, node_vector):
         Uv = self.update_U(node_vector)
         Vv = self.update_V(node_vector)
-        Vv_norm = torch.linalg.norm(Vv, dim=1)
+        # eps-safe norm: linalg.norm's backward is v/|v| and NaNs on
+        # exactly-zero per-channel vectors (common: vector features
+        # start at 0) — fatal for the force double-backward
+        Vv_norm = torch.sqrt((Vv * Vv).sum(dim=1) + 1e-12)
         mlp_input = torch.cat([Vv_norm, node_scalar], dim=1)
         mlp_output = self.update_mlp(mlp_input)
         a_vv, a_sv, a_ss = torch.split(mlp_output, self.node_size, dim=1)

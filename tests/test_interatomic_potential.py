@@ -24,13 +24,17 @@ def _mlip_config(mpnn_type, num_epoch=6):
     arch["energy_peratom_weight"] = 0.0
     arch["force_weight"] = 10.0
     arch["radius"] = 2.5
-    arch["equivariance"] = False
+    arch["equivariance"] = mpnn_type in ("PAINN",)
+    if mpnn_type == "PAINN":
+        arch["num_radial"] = 8
     config["NeuralNetwork"]["Variables_of_interest"]["output_dim"] = [1]
     return config
 
 
-@pytest.mark.parametrize("mpnn_type", ["SchNet", "EGNN"])
+@pytest.mark.parametrize("mpnn_type", ["SchNet", "EGNN", "PAINN"])
 def test_energy_force_training(mpnn_type):
+    # PAINN regression: per-channel zero vectors NaN'd the force
+    # double-backward through linalg.norm (now eps-safe)
     torch.manual_seed(3)
     config = _mlip_config(mpnn_type)
     dataset = lj_dataset(num_samples=24, num_atoms=27, pbc=False)
