@@ -51,10 +51,15 @@ def main():
             equivariance=True, use_gpu=False, **extra)
         _, pd, _ = resolve_precision(precision)
         model = model.to(device=device, dtype=pd)
-    default_lr = {"MACE": 2e-3, "SchNet": 2e-3, "EGNN": 1e-3,
+    default_lr = {"MACE": 2e-3, "SchNet": 2e-3, "EGNN": 2e-3,
                   "PAINN": 3e-4}.get(mpnn, 1e-3)
     lr = float(os.environ.get("CONV_LR", default_lr))
     opt = torch.optim.AdamW(model.parameters(), lr=lr, foreach=True)
+    clip = os.environ.get("CONV_CLIP")
+    if clip is None and mpnn == "EGNN":
+        clip = "10.0"   # EGNN diverges unclipped at useful lr
+    if clip is not None:
+        opt._hydragnn_grad_clip = float(clip)
     sched = torch.optim.lr_scheduler.CosineAnnealingLR(opt, epochs)
 
     # tame LJ landscape (min pair distance 0.95 sigma) so the
