@@ -51,7 +51,7 @@ def main():
             equivariance=True, use_gpu=False, **extra)
         _, pd, _ = resolve_precision(precision)
         model = model.to(device=device, dtype=pd)
-    default_lr = {"MACE": 2e-3, "SchNet": 2e-3, "EGNN": 2e-3,
+    default_lr = {"MACE": 2e-3, "SchNet": 2e-3, "EGNN": 1e-3,
                   "PAINN": 3e-4}.get(mpnn, 1e-3)
     lr = float(os.environ.get("CONV_LR", default_lr))
     opt = torch.optim.AdamW(model.parameters(), lr=lr, foreach=True)
