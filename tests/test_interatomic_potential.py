@@ -24,14 +24,15 @@ def _mlip_config(mpnn_type, num_epoch=6):
     arch["energy_peratom_weight"] = 0.0
     arch["force_weight"] = 10.0
     arch["radius"] = 2.5
-    arch["equivariance"] = mpnn_type in ("PAINN",)
-    if mpnn_type == "PAINN":
+    arch["equivariance"] = mpnn_type in ("PAINN", "PNAEq")
+    if mpnn_type in ("PAINN", "PNAEq"):
         arch["num_radial"] = 8
     config["NeuralNetwork"]["Variables_of_interest"]["output_dim"] = [1]
     return config
 
 
-@pytest.mark.parametrize("mpnn_type", ["SchNet", "EGNN", "PAINN"])
+@pytest.mark.parametrize("mpnn_type", ["SchNet", "EGNN", "PAINN",
+                                       "PNAEq"])
 def test_energy_force_training(mpnn_type):
     # PAINN regression: per-channel zero vectors NaN'd the force
     # double-backward through linalg.norm (now eps-safe)
@@ -76,3 +77,28 @@ def test_md17_shape_dataset():
     assert ds[0].num_nodes == 21
     assert ds[0].forces.shape == (21, 3)
     assert torch.isfinite(ds[0].energy).all()
+
+
+def test_dimenet_energy_force_training():
+    """DimeNet MLIP force path (triplet angles under double backward)
+    stays finite and optimizes."""
+    torch.manual_seed(3)
+    config = _mlip_config("DimeNet")
+    arch = config["NeuralNetwork"]["Architecture"]
+    arch.update({"num_radial": 6, "num_spherical": 7,
+                 "basis_emb_size": 8, "int_emb_size": 64,
+                 "out_emb_size": 128, "num_before_skip": 1,
+                 "num_after_skip": 2})
+    dataset = lj_dataset(num_samples=24, num_atoms=27, pbc=False)
+    trainset, valset, testset = split_dataset(dataset, 0.8, seed=0)
+    loaders = create_dataloaders(trainset, valset, testset, 8,
+                                 config=config)
+    config = update_config(config, *loaders)
+    model = create_model_config(config["NeuralNetwork"], use_gpu=False)
+    opt = select_optimizer(
+        model, config["NeuralNetwork"]["Training"]["Optimizer"])
+    errs = []
+    for _ in range(4):
+        err, _ = train_fn(loaders[0], model, opt, 0)
+        errs.append(float(err))
+    assert all(e == e for e in errs), errs
