@@ -56,6 +56,11 @@ def pad_batch_static(
     Requires ``node_cap >= num_nodes + 2`` (the pad graph needs at
     least 2 nodes to host pad edges).
     """
+    if batch.get("y_loc") is not None:
+        raise NotImplementedError(
+            "static-shape padding does not support y_loc multihead "
+            "y-packing yet (the pad graph would shift the per-sample "
+            "offsets); use it with MLIP / single-head targets")
     n = batch.num_nodes
     e = batch.num_edges
     b = batch.num_graphs
