@@ -131,3 +131,47 @@ def test_config_driven_capture_engages():
     import gc
     gc.collect()
     torch.cuda.empty_cache()
+
+
+@needs_gpu
+def test_captured_mixed_shapes_falls_back_eager():
+    """A loader mixing static-collated and plain batches: the stepper
+    engages on matching shapes and the mismatched batch runs eager —
+    both finite, no crash."""
+    from torch.utils.data import DataLoader
+
+    import bench as bench_mod
+    from hydragnn_amd.data import Batch
+    from hydragnn_amd.preprocess.static_batch import (
+        StaticShapeCollater, compute_static_caps)
+    from hydragnn_amd.train import train
+    from hydragnn_amd.utils.datasets.synthetic import (
+        md17_shape_dataset_fast)
+
+    torch.manual_seed(13)
+    model = bench_mod.build_model("cuda:0", precision="bf16_pure",
+                                  seed=13)
+    from hydragnn_amd.ops.fused_adamw import FusedAdamW
+    opt = FusedAdamW(model.parameters(), lr=1e-3)
+    ds = md17_shape_dataset_fast(48, seed=31)
+    nc, ec = compute_static_caps(ds, 16)
+    coll = StaticShapeCollater(nc, ec, 28.0)
+
+    class MixedLoader:
+        def __iter__(self):
+            yield coll(ds[:16])                       # static
+            yield Batch.from_data_list(
+                [d.clone() for d in ds[16:28]])       # 12 graphs, raw
+            yield coll(ds[32:48])                     # static again
+
+        def __len__(self):
+            return 3
+
+    err, _ = train(MixedLoader(), model, opt, 0, precision="bf16_pure")
+    assert torch.isfinite(err).all()
+    stepper = getattr(model, "_hip_captured_step", None)
+    assert stepper not in (None, False)
+    model._hip_captured_step = None
+    import gc
+    gc.collect()
+    torch.cuda.empty_cache()
