@@ -17,7 +17,6 @@ import torch.distributed as dist
 
 sys.path.insert(0, os.path.join(os.path.dirname(__file__), "..", ".."))
 
-from hydragnn_amd.data import Batch
 from hydragnn_amd.models import create_model_config
 from hydragnn_amd.models.multitask_mp import MultiTaskModelMP
 from hydragnn_amd.preprocess import create_dataloaders

@@ -12,7 +12,6 @@ import sys
 
 import numpy as np
 import torch
-import torch.distributed as dist
 
 sys.path.insert(0, os.path.join(os.path.dirname(__file__), "..", ".."))
 

@@ -21,7 +21,7 @@ Differences from the reference by design:
 
 from __future__ import annotations
 
-from typing import Dict, List, Optional
+from typing import List, Optional
 
 import torch
 from torch import nn

@@ -10,7 +10,7 @@ layers; the last layer produces scalars only.
 
 from __future__ import annotations
 
-from typing import List, Optional
+from typing import Optional
 
 import torch
 from torch import nn
