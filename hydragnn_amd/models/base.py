@@ -58,6 +58,10 @@ def global_max_pool(x, batch, size=None):
 
 
 class Base(Module):
+    # stacks whose forward is shape-static given a fixed batch shape
+    # override this to True (see train/captured.py)
+    _hipgraph_capture_safe = False
+
     def __init__(
         self,
         input_dim: int,
@@ -122,6 +126,16 @@ class Base(Module):
         self.use_global_attn = bool(self.global_attn_engine) and \
             self.global_attn_engine.lower() not in ("", "none", "false")
         self.conv_checkpointing = False
+        # hipGraph-capture safety (train/captured.py): only stacks
+        # whose forward has NO data-dependent internal shapes may be
+        # replayed across batches (class attr _hipgraph_capture_safe,
+        # opt-in per stack).  Dense-batch global attention
+        # (to_dense_batch maxN varies per batch) force-disables it.
+        self.supports_hipgraph_capture = (
+            self._hipgraph_capture_safe and not (
+                bool(global_attn_engine)
+                and str(global_attn_engine).lower()
+                not in ("", "none", "false")))
         self.graph_pooling = graph_pooling
         self.use_graph_attr_conditioning = use_graph_attr_conditioning
         self.graph_attr_conditioning_mode = graph_attr_conditioning_mode.lower()

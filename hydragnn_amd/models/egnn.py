@@ -75,6 +75,7 @@ class _EGCLWrapper(nn.Module):
 
 
 class EGCLStack(Base):
+    _hipgraph_capture_safe = True  # uses only the given edge_index
     def __init__(self, edge_dim: Optional[int] = None,
                  max_neighbours: Optional[int] = None, **kwargs):
         self.is_edge_model = True

@@ -35,6 +35,7 @@ NUM_ELEMENTS = 118
 
 
 class MACEStack(Base):
+    _hipgraph_capture_safe = True  # uses only the given edge_index
     def __init__(
         self,
         r_max: Optional[float] = None,

@@ -133,6 +133,7 @@ class _PainnWrapper(nn.Module):
 
 
 class PAINNStack(Base):
+    _hipgraph_capture_safe = True  # uses only the given edge_index
     def __init__(self, edge_dim: Optional[int] = None,
                  num_radial: Optional[int] = None,
                  radius: Optional[float] = None, **kwargs):

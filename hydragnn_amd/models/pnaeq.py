@@ -89,6 +89,7 @@ class _PNAEqConv(nn.Module):
 
 
 class PNAEqStack(Base):
+    _hipgraph_capture_safe = True  # uses only the given edge_index
     def __init__(self, deg: List[int], edge_dim: Optional[int] = None,
                  num_radial: Optional[int] = None,
                  radius: Optional[float] = None, **kwargs):

@@ -25,7 +25,15 @@ from .layers import (
 )
 
 
-class GINStack(Base):
+class _StaticShapeStack(Base):
+    """Marker base: forward uses only the provided edge_index (no
+    dynamic graph rebuild / data-dependent shapes) -> safe to replay
+    as a hipGraph across same-shape batches."""
+
+    _hipgraph_capture_safe = True
+
+
+class GINStack(_StaticShapeStack):
     def get_conv(self, input_dim, output_dim, edge_dim=None):
         mlp = nn.Sequential(
             nn.Linear(input_dim, output_dim),
@@ -38,7 +46,7 @@ class GINStack(Base):
         return "GINStack"
 
 
-class SAGEStack(Base):
+class SAGEStack(_StaticShapeStack):
     def get_conv(self, input_dim, output_dim, edge_dim=None):
         return InvariantConvWrapper(SAGEConv(input_dim, output_dim))
 
@@ -46,7 +54,7 @@ class SAGEStack(Base):
         return "SAGEStack"
 
 
-class MFCStack(Base):
+class MFCStack(_StaticShapeStack):
     def __init__(self, max_degree: int = 10, **kwargs):
         self.max_degree = max_degree
         super().__init__(**kwargs)
@@ -59,7 +67,7 @@ class MFCStack(Base):
         return "MFCStack"
 
 
-class CGCNNStack(Base):
+class CGCNNStack(_StaticShapeStack):
     """CGConv preserves channel count; hidden_dim is forced equal to
     input_dim by the config normalizer (reference config_utils.py:97-104)."""
 
@@ -76,7 +84,7 @@ class CGCNNStack(Base):
         return "CGCNNStack"
 
 
-class PNAStack(Base):
+class PNAStack(_StaticShapeStack):
     def __init__(self, deg: List[int], edge_dim=None, **kwargs):
         self.deg = torch.tensor(deg, dtype=torch.float)
         self.is_edge_model = True
@@ -96,7 +104,7 @@ class PNAStack(Base):
         return "PNAStack"
 
 
-class GATStack(Base):
+class GATStack(_StaticShapeStack):
     """GATv2, 6 heads, concat on all but the last conv layer; BatchNorm
     dims are head-aware (reference GATStack.py:39-175)."""
 

@@ -37,10 +37,13 @@ _SKIP_COPY = {"edges_sorted_", "num_graphs_", "num_nodes_",
               "num_real_graphs_", "static_shape_"}
 
 
-def capture_enabled(batch=None) -> bool:
+def capture_enabled(batch=None, model=None) -> bool:
     """Captured stepping is on for CUDA devices when the batch is
-    static-shape collated (or forced via HYDRAGNN_CAPTURE=1), unless
-    disabled via HYDRAGNN_CAPTURE=0."""
+    static-shape collated AND the model declares its forward
+    shape-static (``supports_hipgraph_capture`` — replaying a graph
+    that rebuilds neighbor lists or triplets internally, e.g. SchNet /
+    DimeNet / dense-batch attention, would silently freeze
+    data-dependent shapes).  HYDRAGNN_CAPTURE=1 forces, =0 disables."""
     env = os.environ.get("HYDRAGNN_CAPTURE")
     if env == "0":
         return False
@@ -48,6 +51,10 @@ def capture_enabled(batch=None) -> bool:
         return False
     if env == "1":
         return True
+    if model is not None:
+        base = model.module if hasattr(model, "module") else model
+        if not getattr(base, "supports_hipgraph_capture", False):
+            return False
     return bool(batch is not None and batch.get("static_shape_"))
 
 

@@ -238,7 +238,7 @@ def train(loader, model, opt, verbosity, precision="fp32", profiler=None):
         # hipGraph-captured fast path (static-shape batches on GPU):
         # H2D copy-in + one graph replay per batch (captured.py).
         if (scaler is None and stepper is None and not ds_engine
-                and capture_enabled(data)):
+                and capture_enabled(data, model)):
             dev_data = move_batch_to_device(data, param_dtype)
             # capture the UNWRAPPED module's forward: DDP's python
             # pre/post-forward stays entirely out of the graph (its

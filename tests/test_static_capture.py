@@ -273,3 +273,36 @@ def test_config_driven_static_shape_batching():
     opt = torch.optim.AdamW(model.parameters(), lr=1e-3)
     err, _ = train(tr_loader, model, opt, 0)
     assert torch.isfinite(err).all()
+
+
+def test_capture_gate_requires_shape_static_model():
+    """Models with data-dependent internal shapes (SchNet dynamic
+    radius rebuild, DimeNet triplets, dense global attention) must NOT
+    enter the captured path even with static-collated batches."""
+    from hydragnn_amd.train.captured import capture_enabled
+    from hydragnn_amd.models.create import create_model
+
+    mace = _small_mace(md17_shape_dataset(num_samples=4))
+    assert mace.supports_hipgraph_capture
+    schnet = create_model(
+        mpnn_type="SchNet", input_dim=1, hidden_dim=16,
+        output_dim=[1], output_type=["graph"],
+        output_heads={"graph": [{"type": "branch-0", "architecture": {
+            "num_sharedlayers": 1, "dim_sharedlayers": 16,
+            "num_headlayers": 1, "dim_headlayers": [16]}}]},
+        activation_function="silu", loss_function_type="mse",
+        task_weights=[1.0], num_conv_layers=1, radius=2.5,
+        max_neighbours=10, num_gaussians=8, num_filters=16,
+        use_gpu=False)
+    assert not schnet.supports_hipgraph_capture
+    gin_gps = create_model(
+        mpnn_type="GIN", input_dim=1, hidden_dim=16,
+        output_dim=[1], output_type=["graph"],
+        output_heads={"graph": [{"type": "branch-0", "architecture": {
+            "num_sharedlayers": 1, "dim_sharedlayers": 16,
+            "num_headlayers": 1, "dim_headlayers": [16]}}]},
+        activation_function="silu", loss_function_type="mse",
+        task_weights=[1.0], num_conv_layers=1,
+        global_attn_engine="gps", global_attn_type="multihead",
+        global_attn_heads=2, pe_dim=2, use_gpu=False)
+    assert not gin_gps.supports_hipgraph_capture
