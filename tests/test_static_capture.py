@@ -306,3 +306,24 @@ def test_capture_gate_requires_shape_static_model():
         global_attn_engine="gps", global_attn_type="multihead",
         global_attn_heads=2, pe_dim=2, use_gpu=False)
     assert not gin_gps.supports_hipgraph_capture
+
+
+def test_config_driven_bf16_pure_precision():
+    """Training.precision: bf16_pure casts the model to bf16 (no
+    autocast layer) through the standard config flow."""
+    from hydragnn_amd.models.create import resolve_precision
+    prec, pd, ac = resolve_precision("bf16_pure")
+    assert prec == "bf16_pure" and pd == torch.bfloat16 and ac is None
+    prec2, pd2, _ = resolve_precision("bf16-pure")
+    assert prec2 == "bf16_pure" and pd2 == torch.bfloat16
+
+    from torch.utils.data import DataLoader
+    from hydragnn_amd.preprocess.load_data import _collate
+    from hydragnn_amd.train import train
+
+    ds = md17_shape_dataset(num_samples=8)
+    model = _small_mace(ds).to(torch.bfloat16)
+    loader = DataLoader(ds, batch_size=4, collate_fn=_collate)
+    opt = torch.optim.AdamW(model.parameters(), lr=1e-3)
+    err, _ = train(loader, model, opt, 0, precision="bf16_pure")
+    assert torch.isfinite(err.float()).all()
