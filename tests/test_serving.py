@@ -55,3 +55,25 @@ def test_serving_mlip_energy_forces():
     # malformed input -> 400
     r = client.post("/predict", json={"samples": [{"pos": "bad"}]})
     assert r.status_code == 400
+
+
+@pytest.mark.gpu
+def test_serving_mlip_on_gpu():
+    """Serving runs MLIP energy+forces on cuda:0 (device autoselect)."""
+    if not torch.cuda.is_available():
+        pytest.skip("requires GPU")
+    from test_mace_model import _build, _mace_config
+    from hydragnn_amd.utils.datasets.synthetic import md17_shape_dataset
+    from hydragnn_amd.serve import create_app
+    ds = md17_shape_dataset(num_samples=8)
+    model, config, _ = _build(_mace_config(), ds)
+    app = create_app(model, config)  # autoselects cuda
+    client = TestClient(app)
+    r = client.get("/health")
+    assert r.json()["device"].startswith("cuda")
+    r = client.post("/predict", json={"samples": [_sample(6, 2)],
+                                      "radius": 7.0})
+    assert r.status_code == 200, r.text
+    body = r.json()
+    assert "energy" in body and "forces" in body
+    assert all(e == e for e in body["energy"])
