@@ -32,8 +32,19 @@ def _unwrap(model):
     return model.module if hasattr(model, "module") else model
 
 
-def move_batch_to_device(data, param_dtype):
-    device = get_device()
+def _model_device(model):
+    """Device of the model's parameters — batches must follow the
+    MODEL, not the globally preferred device (a CPU-built model on a
+    GPU box would otherwise receive cuda batches and fail in
+    F.linear)."""
+    for p in model.parameters():
+        return p.device
+    return get_device()
+
+
+def move_batch_to_device(data, param_dtype, device=None):
+    if device is None:
+        device = get_device()
     for key in list(data.keys()):
         v = data[key]
         if torch.is_tensor(v) and torch.is_floating_point(v):
@@ -41,13 +52,14 @@ def move_batch_to_device(data, param_dtype):
     return data.to(device, non_blocking=True)
 
 
-def get_autocast_and_scaler(precision):
+def get_autocast_and_scaler(precision, device=None):
     """Autocast context + loss scaler per precision (reference
     train_validate_test.py:87-110 + GradScaler machinery :783-801):
     bf16 -> autocast, no scaler; fp16 -> autocast + GradScaler (loss
     scaling against underflow; GPU only); fp32/fp64 -> neither."""
     precision, _, autocast_dtype = resolve_precision(precision)
-    device = get_device()
+    if device is None:
+        device = get_device()
     if precision == "bf16":
         use_bf16 = device.type == "cuda" or bool(
             getattr(torch.backends.cpu, "has_bf16", False))
@@ -207,13 +219,13 @@ def train(loader, model, opt, verbosity, precision="fp32", profiler=None):
         profiler = Profiler()
     m = _unwrap(model)
     num_tasks = len(m.loss_weights) if not _use_interatomic(m) else 3
-    device = get_device()
+    device = _model_device(m)
     total_error = torch.zeros(1, device=device)
     tasks_error = torch.zeros(num_tasks, device=device)
     num_samples_local = 0
     model.train()
     _, param_dtype, _ = resolve_precision(precision)
-    autocast, scaler = get_autocast_and_scaler(precision)
+    autocast, scaler = get_autocast_and_scaler(precision, device)
     nbatch = get_nbatch(loader)
     use_ia = _use_interatomic(m)
     dataset = getattr(loader, "dataset", None)
@@ -239,7 +251,7 @@ def train(loader, model, opt, verbosity, precision="fp32", profiler=None):
         # H2D copy-in + one graph replay per batch (captured.py).
         if (scaler is None and stepper is None and not ds_engine
                 and capture_enabled(data, model)):
-            dev_data = move_batch_to_device(data, param_dtype)
+            dev_data = move_batch_to_device(data, param_dtype, device)
             # capture the UNWRAPPED module's forward: DDP's python
             # pre/post-forward stays entirely out of the graph (its
             # reducer is unused on this path — FlatGradSync owns the
@@ -258,7 +270,7 @@ def train(loader, model, opt, verbosity, precision="fp32", profiler=None):
             tr.stop("captured_step")
         else:
             tr.start("h2d")
-            data = move_batch_to_device(data, param_dtype)
+            data = move_batch_to_device(data, param_dtype, device)
             tr.stop("h2d")
             if not ds_engine:
                 opt.zero_grad(set_to_none=True)
@@ -320,13 +332,13 @@ def _eval_pass(loader, model, verbosity, precision, return_samples=False):
     if _window:
         _ds.epoch_begin()
     num_tasks = len(m.loss_weights) if not use_ia else 3
-    device = get_device()
+    device = _model_device(m)
     total_error = torch.zeros(1, device=device)
     tasks_error = torch.zeros(num_tasks, device=device)
     num_samples_local = 0
     model.eval()
     _, param_dtype, _ = resolve_precision(precision)
-    autocast, _ = get_autocast_and_scaler(precision)
+    autocast, _ = get_autocast_and_scaler(precision, device)
     nbatch = get_nbatch(loader)
     true_values = [[] for _ in range(m.num_heads)]
     predicted_values = [[] for _ in range(m.num_heads)]
@@ -334,7 +346,7 @@ def _eval_pass(loader, model, verbosity, precision, return_samples=False):
     for ibatch, data in enumerate(loader):
         if ibatch >= nbatch:
             break
-        data = move_batch_to_device(data, param_dtype)
+        data = move_batch_to_device(data, param_dtype, device)
         if use_ia:
             with torch.enable_grad():
                 data.pos.requires_grad_(True)
