@@ -185,7 +185,10 @@ def build_model_and_batch(device="cuda:0", local_batch=LOCAL_BATCH,
                     eager_step()
             torch.cuda.current_stream().wait_stream(s)
             g = torch.cuda.CUDAGraph()
-            with torch.cuda.graph(g):
+            # capture on the warmup stream: autograd nodes surviving
+            # warmup then have a matching canonical stream (see
+            # train/captured.py)
+            with torch.cuda.graph(g, stream=s):
                 static_loss = fwd_bwd()
 
             def graph_step():

@@ -107,12 +107,22 @@ class CapturedTrainStep:
         with torch.cuda.stream(side), no_sync():
             for _ in range(warmup_iters):
                 loss, _ = fwd_bwd()
+        # Drop every reference to the warmup autograd graph BEFORE
+        # capturing: a live warmup loss keeps the side-stream
+        # AccumulateGrad nodes alive, the capture-time backward reuses
+        # them, and the resulting cross-stream sync during capture
+        # intermittently corrupts the captured gradients (observed as
+        # a ~1e30 loss on the sanity replay -> eager fallback).
+        del loss
         torch.cuda.current_stream().wait_stream(side)
         self._grads = [p.grad for p in params if p.grad is not None]
 
         self.graph = torch.cuda.CUDAGraph()
         with no_sync():
-            with torch.cuda.graph(self.graph):
+            # capture on the SAME stream the warmups ran on, so any
+            # autograd node that does survive warmup has a matching
+            # canonical stream
+            with torch.cuda.graph(self.graph, stream=side):
                 self.loss, self.tasks_loss = fwd_bwd()
 
         self.grad_sync = FlatGradSync(
