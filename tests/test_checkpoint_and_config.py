@@ -89,6 +89,23 @@ def test_move_batch_dtype():
     assert b.edge_index.dtype == torch.long  # ints untouched
 
 
+def test_batches_follow_model_device():
+    """Batches go to the MODEL's device, not the globally preferred one
+    (a CPU-built model on a GPU box must receive CPU batches)."""
+    import torch.nn as nn
+
+    from hydragnn_amd.train.train_validate_test import _model_device
+
+    m = nn.Linear(3, 3)
+    assert _model_device(m) == next(m.parameters()).device
+    ds = make_deterministic_dataset(num_samples=2, num_heads_node=0)
+    b = Batch.from_data_list(ds)
+    b = move_batch_to_device(b, torch.float32, _model_device(m))
+    assert b.x.device == next(m.parameters()).device
+    # no parameters -> falls back to the global device (never raises)
+    assert _model_device(nn.Identity()) is not None
+
+
 def test_update_multibranch_heads():
     out = update_multibranch_heads({"graph": {"num_sharedlayers": 1,
                                               "dim_sharedlayers": 4,
