@@ -20,14 +20,17 @@ from ...ops import radius_graph, radius_graph_pbc, scatter
 
 def _lj_energy_forces(pos: torch.Tensor, edge_index: torch.Tensor,
                       shifts: Optional[torch.Tensor], epsilon: float,
-                      sigma: float):
+                      sigma: float, r_min: float = 0.0):
     """Pairwise LJ over the given edge list (each pair appears twice —
-    once per direction — so use 0.5x for energy)."""
+    once per direction — so use 0.5x for energy).  ``r_min`` soft-cores
+    the potential: distances below it are clamped in the energy/force
+    evaluation so near-coincident atoms (random geometries) cannot
+    produce astronomically large targets."""
     src, dst = edge_index[0], edge_index[1]
     vec = pos[dst] - pos[src]
     if shifts is not None:
         vec = vec + shifts
-    r2 = (vec * vec).sum(-1).clamp(min=1e-12)
+    r2 = (vec * vec).sum(-1).clamp(min=max(1e-12, r_min * r_min))
     inv_r2 = (sigma * sigma) / r2
     inv_r6 = inv_r2 ** 3
     inv_r12 = inv_r6 ** 2
@@ -131,7 +134,7 @@ def md17_shape_dataset(
         edge_index = radius_graph(pos, radius,
                                   max_num_neighbors=max_neighbours)
         energy, forces = _lj_energy_forces(
-            pos.double(), edge_index, None, 0.05, 1.0)
+            pos.double(), edge_index, None, 0.05, 1.0, r_min=0.7)
         d = Data(
             x=z.to(dtype).view(-1, 1),
             z=z.clone(),
@@ -186,6 +189,12 @@ def md17_shape_dataset_fast(
         vec = posd.unsqueeze(1) - posd.unsqueeze(1).transpose(1, 2)
         r2 = (vec * vec).sum(-1).clamp(min=1e-12)
         within = (r2 < radius * radius) & ~eye
+        # soft-core at min_dist: the 1.25x push loop cannot separate
+        # near-coincident pairs (a multiplicative rescale leaves tiny
+        # distances tiny), and an unclamped LJ at r~0.05 produces
+        # ~1e15 force targets that blow the initial loss up to ~1e30
+        # (observed tripping the captured-step sanity check).
+        r2 = r2.clamp(min=min_dist * min_dist)
         sigma2 = 1.0
         inv_r2 = sigma2 / r2
         inv_r6 = inv_r2 ** 3
