@@ -458,3 +458,48 @@ def test_sharded_dataset_inside_train_loop():
         p.join(timeout=60)
     for rank, ok, info in results:
         assert ok, f"rank {rank}: {info}"
+
+
+def test_compositional_stratified_splitting_behavior():
+    """Every composition category lands in every split where possible,
+    singletons are duplicated rather than dropped, and no sample is
+    lost (reference compositional_data_splitting.py:19-156)."""
+    import torch
+
+    from hydragnn_amd.data import Data
+    from hydragnn_amd.preprocess.compositional_splitting import (
+        compositional_stratified_split, compositional_stratified_splitting,
+        create_dataset_categories)
+
+    def mol(zs):
+        z = torch.tensor(zs, dtype=torch.long)
+        d = Data(x=z.float().view(-1, 1), z=z,
+                 pos=torch.randn(len(zs), 3),
+                 y=torch.zeros(1, 1))
+        d.num_nodes = len(zs)
+        return d
+
+    torch.manual_seed(0)
+    # two well-populated compositions + one singleton
+    ds = [mol([1, 1, 8]) for _ in range(10)] \
+        + [mol([6, 6, 6, 1]) for _ in range(10)] + [mol([7, 7])]
+
+    cats = create_dataset_categories(ds)
+    assert len(set(cats)) == 3
+
+    tr, va, te = compositional_stratified_splitting(ds, 0.8)
+    # singleton duplicated -> 22 total, nothing lost
+    assert len(tr) + len(va) + len(te) == len(ds) + 1
+
+    def comps(split):
+        return {tuple(sorted(set(d.z.tolist()))) for d in split}
+
+    # every category reaches the train split
+    assert comps(tr) == {(1, 8), (1, 6), (7,)}
+    # the two populated categories appear in val+test too
+    assert {(1, 8), (1, 6)} <= comps(va) | comps(te)
+
+    # sklearn-backed entry point: proportional sizes, no loss
+    tr2, va2, te2 = compositional_stratified_split(ds, 0.8, seed=1)
+    assert len(tr2) + len(va2) + len(te2) == len(ds)
+    assert abs(len(tr2) - 0.8 * len(ds)) <= 2
