@@ -180,7 +180,15 @@ def _allreduce_sum(t):
 def gather_tensor_ranks(head_values):
     if not dist.is_initialized() or dist.get_world_size() == 1:
         return head_values
-    head_values = head_values.to(get_device())
+    orig_device = head_values.device
+    # gather on the device the backend communicates on: RCCL ("nccl")
+    # needs cuda tensors, gloo needs cpu (ROCm gloo has no CUDA-tensor
+    # support)
+    if dist.get_backend() == "nccl":
+        from ..utils.distributed import to_comm_device
+        head_values, _ = to_comm_device(head_values)
+    elif head_values.is_cuda:
+        head_values = head_values.cpu()
     size_local = torch.tensor([head_values.shape[0]], dtype=torch.int64,
                               device=head_values.device)
     size_all = [torch.ones_like(size_local)
@@ -194,7 +202,8 @@ def gather_tensor_ranks(head_values):
     gathered = [torch.zeros_like(padded)
                 for _ in range(dist.get_world_size())]
     dist.all_gather(gathered, padded)
-    return torch.cat([g[: int(s)] for g, s in zip(gathered, size_all)], 0)
+    return torch.cat([g[: int(s)] for g, s in zip(gathered, size_all)],
+                     0).to(orig_device)
 
 
 # ---------------------------------------------------------------------------
