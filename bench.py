@@ -316,7 +316,9 @@ def main():
     except TypeError:  # non-fp32 params (fp64 benches)
         optimizer = torch.optim.AdamW(model.parameters(), lr=1e-3,
                                       foreach=True)
-    model = distributed_model_wrapper(model)
+    # the bench MLIP uses every parameter each step; skip DDP's
+    # per-iteration unused-parameter traversal on the eager path
+    model = distributed_model_wrapper(model, find_unused_parameters=False)
     if getattr(optimizer, "master", None) is not None:
         # DDP's construction-time broadcast may have overwritten the
         # flat params; keep the fp32 master in sync

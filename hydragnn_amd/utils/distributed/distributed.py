@@ -255,20 +255,25 @@ def is_deepspeed_engine(model) -> bool:
 
 
 def distributed_model_wrapper(model, max_neighbours=None, verbosity: int = 0,
-                              find_unused_parameters: bool = False,
+                              find_unused_parameters=None,
                               sync_batch_norm: bool = False):
-    """Reference distributed.py:489: move to device and wrap."""
+    """Reference distributed.py:489: move to device and wrap.
+
+    ``find_unused_parameters``: None (default) auto-enables for MLIP
+    ``EnhancedModelWrapper`` models (their force double-backward can
+    leave head params unused in the first pass — reference toggle);
+    an explicit True/False from the caller is respected (False skips
+    DDP's per-iteration unused-parameter graph traversal when the
+    caller knows every parameter is used, e.g. bench.py)."""
     device = get_device()
     if sync_batch_norm and dist.is_initialized() and \
             dist.get_world_size() > 1 and device.type == "cuda":
         model = torch.nn.SyncBatchNorm.convert_sync_batchnorm(model)
     model = model.to(device)
     if dist.is_initialized() and dist.get_world_size() >= 1:
-        # EnhancedModelWrapper's second autograd pass can leave params
-        # unused in the first; mirror the reference's toggle.
-        from ...models.create import EnhancedModelWrapper
-        if isinstance(model, EnhancedModelWrapper):
-            find_unused_parameters = True
+        if find_unused_parameters is None:
+            from ...models.create import EnhancedModelWrapper
+            find_unused_parameters = isinstance(model, EnhancedModelWrapper)
         model = get_distributed_model(
             model, verbosity, find_unused_parameters=find_unused_parameters)
     return model
