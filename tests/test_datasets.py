@@ -503,3 +503,22 @@ def test_compositional_stratified_splitting_behavior():
     tr2, va2, te2 = compositional_stratified_split(ds, 0.8, seed=1)
     assert len(tr2) + len(va2) + len(te2) == len(ds)
     assert abs(len(tr2) - 0.8 * len(ds)) <= 2
+
+
+def test_md17_shape_targets_bounded():
+    """Soft-cored LJ targets stay bounded even when the random
+    geometry contains near-coincident pairs (unclamped LJ produced
+    ~1e15 force targets and a ~1e30 initial loss that tripped the
+    captured-step sanity check)."""
+    import torch
+
+    from hydragnn_amd.utils.datasets.synthetic import (
+        md17_shape_dataset, md17_shape_dataset_fast)
+
+    for ds in (md17_shape_dataset_fast(512, seed=13),
+               md17_shape_dataset(64, seed=13)):
+        y = torch.cat([d.y.reshape(-1) for d in ds])
+        f = torch.cat([d.forces.reshape(-1) for d in ds])
+        assert y.abs().max() < 1e3, float(y.abs().max())
+        assert f.abs().max() < 1e5, float(f.abs().max())
+        assert torch.isfinite(y).all() and torch.isfinite(f).all()
