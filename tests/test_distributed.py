@@ -264,3 +264,30 @@ def test_fsdp2_force_grad_regression_multirank(world_size, port):
         p.join(timeout=60)
     for rank, ok, info in results:
         assert ok, f"rank {rank}: {info}"
+
+
+def test_ddp_wrapper_find_unused_tristate():
+    """None auto-enables find_unused_parameters for MLIP wrappers;
+    an explicit False from the caller is respected."""
+    import os
+
+    import torch.distributed as dist
+
+    from hydragnn_amd.utils.distributed import distributed_model_wrapper
+    from test_mace_model import _build, _mace_config
+    from hydragnn_amd.utils.datasets.synthetic import md17_shape_dataset
+
+    os.environ.update(MASTER_ADDR="127.0.0.1", MASTER_PORT="29721",
+                      RANK="0", WORLD_SIZE="1", LOCAL_RANK="0")
+    dist.init_process_group("gloo", rank=0, world_size=1)
+    try:
+        ds = md17_shape_dataset(num_samples=4)
+        model, _, _ = _build(_mace_config(), ds)
+        auto = distributed_model_wrapper(model)
+        assert auto.find_unused_parameters is True
+        model2, _, _ = _build(_mace_config(), ds)
+        off = distributed_model_wrapper(model2,
+                                        find_unused_parameters=False)
+        assert off.find_unused_parameters is False
+    finally:
+        dist.destroy_process_group()
