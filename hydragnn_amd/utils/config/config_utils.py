@@ -286,16 +286,23 @@ def get_log_name_config(config) -> str:
 
 
 def parse_deepspeed_config(config) -> Dict:
-    """The reference can emit a DeepSpeed config
-    (config_utils.py:455); we keep the entry point for API parity. On
-    MI355X the native path is DDP/FSDP over RCCL."""
-    training = config["NeuralNetwork"]["Training"]
-    return {
-        "train_batch_size": training.get("batch_size", 32),
-        "optimizer": {
-            "type": training.get("Optimizer", {}).get("type", "AdamW"),
-            "params": {
-                "lr": training.get("Optimizer", {}).get("learning_rate", 1e-3)
-            },
-        },
-    }
+    """DeepSpeed engine config (reference config_utils.py:455-471):
+    a user-supplied ``NeuralNetwork.ds_config`` section passes
+    through; otherwise the per-GPU micro batch defaults to
+    ``Training.batch_size`` with no gradient accumulation.  The
+    PER-GPU key matters: a global ``train_batch_size`` equal to the
+    local batch fails DeepSpeed's micro*accum*world consistency check
+    at world_size > 1.  No ``optimizer`` section — the wrapper passes
+    the already-built optimizer instance to ``deepspeed.initialize``
+    and DeepSpeed rejects specifying both.  On MI355X the native path
+    is DDP/FSDP over RCCL."""
+    ds_config = dict(config["NeuralNetwork"].get("ds_config", {}))
+    if "train_micro_batch_size_per_gpu" not in ds_config and \
+            "train_batch_size" not in ds_config:
+        training = config["NeuralNetwork"]["Training"]
+        ds_config["train_micro_batch_size_per_gpu"] = \
+            training.get("batch_size", 32)
+        ds_config["gradient_accumulation_steps"] = 1
+    if "steps_per_print" not in ds_config:
+        ds_config["steps_per_print"] = int(1e9)  # disable printing
+    return ds_config
