@@ -233,8 +233,22 @@ def test_parse_deepspeed_config():
         "batch_size": 16,
         "Optimizer": {"type": "AdamW", "learning_rate": 5e-4}}}}
     ds = parse_deepspeed_config(cfg)
-    assert ds["train_batch_size"] == 16
-    assert ds["optimizer"]["params"]["lr"] == 5e-4
+    # per-GPU micro batch (a global train_batch_size equal to the
+    # local batch fails DeepSpeed's consistency check at world > 1)
+    assert ds["train_micro_batch_size_per_gpu"] == 16
+    assert ds["gradient_accumulation_steps"] == 1
+    assert ds["steps_per_print"] >= 1e9
+    # no optimizer section: the wrapper passes the built optimizer
+    # instance to deepspeed.initialize, and both together are rejected
+    assert "optimizer" not in ds
+    # user ds_config passes through untouched
+    cfg["NeuralNetwork"]["ds_config"] = {
+        "train_micro_batch_size_per_gpu": 4, "zero_optimization":
+        {"stage": 1}}
+    ds = parse_deepspeed_config(cfg)
+    assert ds["train_micro_batch_size_per_gpu"] == 4
+    assert ds["zero_optimization"]["stage"] == 1
+    assert "gradient_accumulation_steps" not in ds
 
 
 def test_visualizer_plot_suite(tmp_path):
