@@ -158,3 +158,9 @@ def compute_batch_statistics(sampler) -> BatchStatistics:
         max_cost=max(batch_costs) if batch_costs else 0,
         mean_cost=(sum(batch_costs) / len(batch_costs)
                    if batch_costs else 0.0))
+
+
+def graph_node_cost(data) -> int:
+    """Cost of a single graph (reference batch_sampler singular
+    form); see graph_node_costs for the dataset vectorized variant."""
+    return int(data.num_nodes)

@@ -187,3 +187,10 @@ def update_atom_features(atom_features, data: Data):
            for f in atom_features]
     data.x = data.x[:, idx]
     return data
+
+
+# reference-named aliases (reference splits dist/MPI variants;
+# ours picks the plane via HYDRAGNN_AGGR_BACKEND)
+gather_deg_dist = gather_deg
+gather_deg_mpi = gather_deg
+check_if_graph_size_variable_mpi = check_if_graph_size_variable_dist

@@ -546,3 +546,9 @@ def get_head_indices_node_or_mixed(model, data):
 
 reduce_values_ranks_dist = reduce_values_ranks
 reduce_values_ranks_mpi = reduce_values_ranks
+
+
+# reference-named aliases (reference train_validate_test.py:609-655
+# has separate torch-dist / MPI reduction entry points)
+reduce_values_ranks_dist = reduce_values_ranks
+reduce_values_ranks_mpi = reduce_values_ranks

@@ -35,10 +35,18 @@ def get_node_attribute_name(types=None) -> Tuple[List[str], List[int]]:
 def generate_graphdata_from_smilestr(smilestr: str, ytarget,
                                      types=None) -> Data:
     Chem = _require_rdkit()
-    types = types or ["C", "N", "O", "F", "H"]
     mol = Chem.MolFromSmiles(smilestr)
     if mol is None:
         raise ValueError(f"invalid SMILES: {smilestr}")
+    return generate_graphdata_from_rdkit_molecule(mol, ytarget, types)
+
+
+def generate_graphdata_from_rdkit_molecule(mol, ytarget,
+                                           types=None) -> Data:
+    """Graph Data from an rdkit molecule (reference
+    smiles_utils.py:59)."""
+    Chem = _require_rdkit()
+    types = types or ["C", "N", "O", "F", "H"]
     mol = Chem.AddHs(mol)
     n = mol.GetNumAtoms()
     feats = []

@@ -83,3 +83,7 @@ def find_bin(value: float, edges) -> int:
         if edges[i] <= value < edges[i + 1]:
             return i
     return len(edges) - 2
+
+
+# reference-named alias
+compute_formation_enthalpy = get_formation_enthalpy

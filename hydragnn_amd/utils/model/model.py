@@ -247,3 +247,12 @@ def multitask_optim_state_dict(dual_optimizer):
     if not out and hasattr(dual_optimizer, "state_dict"):
         out = {"optimizer": dual_optimizer.state_dict()}
     return out
+
+
+# reference-named aliases: the reference exposes separate torch-dist /
+# MPI reduction variants (utils/model/model.py:357-440); ours selects
+# the aggregation plane via HYDRAGNN_AGGR_BACKEND inside one function.
+calculate_PNA_degree_dist = calculate_PNA_degree
+calculate_PNA_degree_mpi = calculate_PNA_degree
+calculate_avg_deg_dist = calculate_avg_deg
+calculate_avg_deg_mpi = calculate_avg_deg

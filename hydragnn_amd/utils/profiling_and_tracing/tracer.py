@@ -272,3 +272,8 @@ def pr_summary_file(path: str):
 def nvtx_timer(name: str):
     with timer(name):
         yield
+
+
+# reference-named alias: the reference's abstract tracer base class is
+# `Tracer` (tracer.py:29)
+Tracer = TracerBackend
