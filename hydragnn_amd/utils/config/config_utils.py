@@ -133,32 +133,7 @@ def update_config(config, train_loader, val_loader, test_loader):
 
     # --- output dims from data / y_loc ---
     data0 = train_loader.dataset[0]
-    output_type = var["type"]
-    if arch.get("enable_interatomic_potential", False):
-        dims_list = var["output_dim"]
-    elif data0.get("y_loc") is not None:
-        dims_list = []
-        for ihead in range(len(output_type)):
-            span = int(data0.y_loc[0, ihead + 1]) - int(data0.y_loc[0, ihead])
-            if output_type[ihead] == "graph":
-                dims_list.append(span)
-            elif output_type[ihead] == "node":
-                if (graph_size_variable and
-                        arch["output_heads"]["node"][0]["architecture"]["type"]
-                        == "mlp_per_node"):
-                    raise ValueError(
-                        "mlp_per_node not allowed for variable graph size")
-                dims_list.append(span // data0.num_nodes)
-            else:
-                raise ValueError(f"Unknown output type {output_type[ihead]}")
-    else:
-        for t in output_type:
-            if t != "graph":
-                raise ValueError("y_loc needed for non-graph outputs")
-        dims_list = var["output_dim"]
-    arch["output_dim"] = dims_list
-    arch["output_type"] = output_type
-    arch["num_nodes"] = data0.num_nodes
+    update_config_NN_outputs(config, data0, graph_size_variable)
 
     var.setdefault("denormalize_output", False)
 
@@ -195,6 +170,59 @@ def update_config(config, train_loader, val_loader, test_loader):
     arch.setdefault("enable_interatomic_potential", False)
 
     # --- edge dim rules ---
+    update_config_edge_dim(config)
+
+    arch.setdefault("equivariance", None)
+    arch.setdefault("freeze_conv_layers", False)
+    arch.setdefault("initial_bias", None)
+    arch.setdefault("activation_function", "relu")
+    arch.setdefault("SyncBatchNorm", False)
+    training.setdefault("conv_checkpointing", False)
+    training.setdefault("loss_function_type", "mse")
+    training.setdefault("Optimizer", {"type": "AdamW", "learning_rate": 1e-3})
+    training.setdefault("precision", "fp32")
+    return config
+
+
+def update_config_NN_outputs(config, data, graph_size_variable):
+    """Derive per-head output dims from the sample's ``y_loc`` packing
+    (reference config_utils.py:313-355)."""
+    nn = config["NeuralNetwork"]
+    arch = nn["Architecture"]
+    var = nn["Variables_of_interest"]
+    output_type = var["type"]
+    if arch.get("enable_interatomic_potential", False):
+        dims_list = var["output_dim"]
+    elif data.get("y_loc") is not None:
+        dims_list = []
+        for ihead in range(len(output_type)):
+            span = int(data.y_loc[0, ihead + 1]) - int(data.y_loc[0, ihead])
+            if output_type[ihead] == "graph":
+                dims_list.append(span)
+            elif output_type[ihead] == "node":
+                if (graph_size_variable and
+                        arch["output_heads"]["node"][0]["architecture"]["type"]
+                        == "mlp_per_node"):
+                    raise ValueError(
+                        "mlp_per_node not allowed for variable graph size")
+                dims_list.append(span // data.num_nodes)
+            else:
+                raise ValueError(f"Unknown output type {output_type[ihead]}")
+    else:
+        for t in output_type:
+            if t != "graph":
+                raise ValueError("y_loc needed for non-graph outputs")
+        dims_list = var["output_dim"]
+    arch["output_dim"] = dims_list
+    arch["output_type"] = output_type
+    arch["num_nodes"] = data.num_nodes
+    return config
+
+
+def update_config_edge_dim(config):
+    """Edge-feature dimension rules per architecture (reference
+    config_utils.py:265-292)."""
+    arch = config["NeuralNetwork"]["Architecture"]
     arch["edge_dim"] = None
     edge_models = ["GAT", "PNA", "PNAPlus", "PAINN", "PNAEq", "CGCNN",
                    "SchNet", "EGNN", "DimeNet", "MACE"]
@@ -206,16 +234,104 @@ def update_config(config, train_loader, val_loader, test_loader):
             "Edge features cannot be combined with interatomic potentials")
     elif arch["mpnn_type"] == "CGCNN":
         arch["edge_dim"] = 0
+    return config
 
-    arch.setdefault("equivariance", None)
-    arch.setdefault("freeze_conv_layers", False)
-    arch.setdefault("initial_bias", None)
-    arch.setdefault("activation_function", "relu")
-    arch.setdefault("SyncBatchNorm", False)
-    training.setdefault("conv_checkpointing", False)
-    training.setdefault("loss_function_type", "mse")
-    training.setdefault("Optimizer", {"type": "AdamW", "learning_rate": 1e-3})
-    training.setdefault("precision", "fp32")
+
+def update_config_equivariance(arch):
+    """Equivariance flag normalization (reference config_utils.py:252):
+    only EGNN toggles behavior on it; other architectures are either
+    inherently equivariant (PaiNN/PNAEq/MACE) or ignore the flag, so a
+    stray setting warns instead of raising."""
+    import warnings
+    toggled = ["EGNN"]
+    inherently = ["SchNet", "PAINN", "PNAEq", "MACE"]
+    if "equivariance" in arch:
+        if arch.get("equivariance") and arch["mpnn_type"] not in (
+                toggled + inherently):
+            warnings.warn(
+                f"E(3) equivariance is only toggled for {toggled}; "
+                f"setting it for {arch['mpnn_type']} has no effect")
+    else:
+        arch["equivariance"] = None
+    return arch
+
+
+def check_output_dim_consistent(data, config):
+    """Assert the sample's y_loc spans match the Dataset feature dims
+    named by output_index (reference config_utils.py:295-310)."""
+    var = config["NeuralNetwork"]["Variables_of_interest"]
+    output_type = var["type"]
+    output_index = var["output_index"]
+    if data.get("y_loc") is None:
+        return
+    for ihead in range(len(output_type)):
+        span = int(data.y_loc[0, ihead + 1]) - int(data.y_loc[0, ihead])
+        if output_type[ihead] == "graph":
+            expect = config["Dataset"]["graph_features"]["dim"][
+                output_index[ihead]]
+            assert span == expect, (ihead, span, expect)
+        elif output_type[ihead] == "node":
+            expect = config["Dataset"]["node_features"]["dim"][
+                output_index[ihead]]
+            assert span // data.num_nodes == expect, (ihead, span, expect)
+
+
+def update_config_minmax(dataset_path, var_config):
+    """Populate ``x_minmax``/``y_minmax`` for output denormalization
+    (reference config_utils.py:381-405): minmax arrays come from the
+    config when present, else from the serialized dataset container's
+    leading two pickle objects (minmax node features, minmax graph
+    features)."""
+    import pickle
+
+    import numpy as np
+    if var_config.get("minmax_node_feature") is not None and \
+            var_config.get("minmax_graph_feature") is not None:
+        node_minmax = np.asarray(var_config["minmax_node_feature"])
+        graph_minmax = np.asarray(var_config["minmax_graph_feature"])
+    else:
+        with open(dataset_path, "rb") as f:
+            node_minmax = np.asarray(pickle.load(f))
+            graph_minmax = np.asarray(pickle.load(f))
+    var_config["x_minmax"] = [
+        node_minmax[:, i].tolist()
+        for i in var_config["input_node_features"]]
+    var_config["y_minmax"] = []
+    for out_t, out_i in zip(var_config["type"],
+                            var_config["output_index"]):
+        if out_t == "graph":
+            var_config["y_minmax"].append(graph_minmax[:, out_i].tolist())
+        elif out_t == "node":
+            var_config["y_minmax"].append(node_minmax[:, out_i].tolist())
+        else:
+            raise ValueError(f"Unknown output type {out_t}")
+    return var_config
+
+
+def normalize_output_config(config):
+    """Resolve the denormalization minmax source from the Dataset
+    paths and fill Variables_of_interest (reference
+    config_utils.py:357-378)."""
+    var = config["NeuralNetwork"]["Variables_of_interest"]
+    if var.get("denormalize_output"):
+        if var.get("minmax_node_feature") is not None and \
+                var.get("minmax_graph_feature") is not None:
+            dataset_path = None
+        else:
+            paths = config["Dataset"]["path"]
+            first = str(list(paths.values())[0])
+            if first.endswith(".pkl"):
+                dataset_path = first
+            else:
+                base = os.environ.get("SERIALIZED_DATA_PATH", os.getcwd())
+                name = config["Dataset"]["name"]
+                suffix = "" if "total" in paths else "_train"
+                dataset_path = (f"{base}/serialized_dataset/"
+                                f"{name}{suffix}.pkl")
+        config["NeuralNetwork"]["Variables_of_interest"] = \
+            update_config_minmax(dataset_path, var)
+    else:
+        var["denormalize_output"] = False
     return config
 
 

@@ -244,3 +244,65 @@ def test_checkpoint_best_metric_gating():
     es = EarlyStopping(patience=2, min_delta=0.0)
     assert not es(1.0) and not es(1.1)
     assert es(1.2)                             # patience exceeded
+
+
+def test_update_config_minmax_and_normalize(tmp_path):
+    """x/y minmax for output denormalization resolve from the config
+    or the serialized container (reference config_utils.py:357-405)."""
+    import pickle
+
+    import numpy as np
+
+    from hydragnn_amd.utils.config import (normalize_output_config,
+                                           update_config_minmax)
+
+    node_minmax = np.array([[0.0, 1.0, 2.0], [10.0, 11.0, 12.0]])
+    graph_minmax = np.array([[-1.0], [5.0]])
+    p = tmp_path / "total.pkl"
+    with open(p, "wb") as f:
+        pickle.dump(node_minmax, f)
+        pickle.dump(graph_minmax, f)
+        pickle.dump([], f)
+
+    var = {"input_node_features": [0, 2], "type": ["graph", "node"],
+           "output_index": [0, 1]}
+    out = update_config_minmax(str(p), dict(var))
+    assert out["x_minmax"] == [[0.0, 10.0], [2.0, 12.0]]
+    assert out["y_minmax"] == [[-1.0, 5.0], [1.0, 11.0]]
+
+    cfg = {"Dataset": {"path": {"total": str(p)}, "name": "t"},
+           "NeuralNetwork": {"Variables_of_interest": {
+               **var, "denormalize_output": True}}}
+    cfg = normalize_output_config(cfg)
+    assert cfg["NeuralNetwork"]["Variables_of_interest"]["y_minmax"] \
+        == [[-1.0, 5.0], [1.0, 11.0]]
+
+    # minmax provided inline -> no file read
+    cfg2 = {"Dataset": {"path": {}, "name": "t"},
+            "NeuralNetwork": {"Variables_of_interest": {
+                **var, "denormalize_output": True,
+                "minmax_node_feature": node_minmax.tolist(),
+                "minmax_graph_feature": graph_minmax.tolist()}}}
+    cfg2 = normalize_output_config(cfg2)
+    assert cfg2["NeuralNetwork"]["Variables_of_interest"]["x_minmax"] \
+        == [[0.0, 10.0], [2.0, 12.0]]
+
+
+def test_check_output_dim_consistent():
+    import torch
+
+    from hydragnn_amd.data import Data
+    from hydragnn_amd.utils.config import check_output_dim_consistent
+
+    d = Data(x=torch.zeros(4, 1), y=torch.zeros(6, 1),
+             y_loc=torch.tensor([[0, 2, 6]]))
+    d.num_nodes = 4
+    cfg = {"Dataset": {"graph_features": {"dim": [2]},
+                       "node_features": {"dim": [0, 1]}},
+           "NeuralNetwork": {"Variables_of_interest": {
+               "type": ["graph", "node"], "output_index": [0, 1]}}}
+    check_output_dim_consistent(d, cfg)  # consistent -> no raise
+    cfg["Dataset"]["graph_features"]["dim"] = [3]
+    import pytest as _pytest
+    with _pytest.raises(AssertionError):
+        check_output_dim_consistent(d, cfg)
