@@ -186,3 +186,137 @@ def main(argv=None):  # pragma: no cover - CLI shell
 
 if __name__ == "__main__":  # pragma: no cover
     main()
+
+
+class AdiosDataset(torch.utils.data.Dataset):
+    """Map-style dataset over one label of a reference ``.bp`` store
+    (reference adiosdataset.py:365-790 AdiosDataset, preload mode).
+    Samples load eagerly at construction; global attributes
+    (``pna_deg``, minmax features, ``dataset_name``) are exposed as
+    attributes for update_config."""
+
+    def __init__(self, filename: str, label: str = "trainset",
+                 keys: Optional[Sequence[str]] = None, file_cls=None):
+        file_cls = file_cls or _Adios2File
+        with file_cls(filename) as f:
+            self.samples = read_adios_samples(f, label, keys=keys)
+            attrs = read_global_attributes(f)
+        self.label = label
+        self.pna_deg = attrs.get("pna_deg")
+        self.minmax_node_feature = attrs.get("minmax_node_feature")
+        self.minmax_graph_feature = attrs.get("minmax_graph_feature")
+        self.dataset_name = attrs.get("dataset_name")
+
+    def __len__(self):
+        return len(self.samples)
+
+    def __getitem__(self, idx):
+        return self.samples[idx]
+
+
+class AdiosMultiDataset(torch.utils.data.Dataset):
+    """Concatenation of one label across several ``.bp`` stores
+    (reference adiosdataset.py:1128 AdiosMultiDataset)."""
+
+    def __init__(self, filenames: Sequence[str], label: str = "trainset",
+                 keys: Optional[Sequence[str]] = None, file_cls=None):
+        self.datasets = [AdiosDataset(fn, label, keys=keys,
+                                      file_cls=file_cls)
+                         for fn in filenames]
+        self._lens = [len(d) for d in self.datasets]
+        self.pna_deg = next((d.pna_deg for d in self.datasets
+                             if d.pna_deg is not None), None)
+
+    def __len__(self):
+        return sum(self._lens)
+
+    def __getitem__(self, idx):
+        for d, n in zip(self.datasets, self._lens):
+            if idx < n:
+                return d[idx]
+            idx -= n
+        raise IndexError(idx)
+
+
+class AdiosWriter:
+    """Write datasets in the reference ``.bp`` layout (reference
+    adiosdataset.py:120-287 AdiosWriter): per label a concatenated
+    global array per key plus variable_count/offset/dim metadata, and
+    global attributes.  The IO backend is injectable for testing; the
+    default uses the adios2 high-level Stream API."""
+
+    def __init__(self, filename: str, backend=None):
+        self.filename = filename
+        self.backend = backend
+        self._labels: Dict[str, List[Data]] = {}
+        self.attributes: Dict[str, object] = {}
+
+    def add_global(self, name: str, value) -> None:
+        self.attributes[name] = value
+
+    def add(self, label: str, samples: Sequence[Data]) -> None:
+        self._labels.setdefault(label, []).extend(samples)
+
+    @staticmethod
+    def _pack(samples: Sequence[Data], key: str):
+        """Concatenate one key across samples along its variable dim
+        (dim 1 for edge_index-style [fixed, n] tensors, else dim 0)
+        with count/offset metadata."""
+        arrs = [np.asarray(s[key].detach().cpu().numpy())
+                for s in samples]
+        vdim = 1 if (arrs[0].ndim == 2 and key == "edge_index") else 0
+        counts = np.array([a.shape[vdim] for a in arrs], dtype=np.int64)
+        offsets = np.concatenate([[0], np.cumsum(counts)[:-1]])
+        return np.concatenate(arrs, axis=vdim), counts, offsets, vdim
+
+    def save(self) -> None:
+        backend = self.backend
+        if backend is None:
+            backend = _Adios2StreamBackend(self.filename)
+        with backend:
+            for name, value in self.attributes.items():
+                backend.write_attribute(name, value)
+            for label, samples in self._labels.items():
+                if not samples:
+                    continue
+                keys = [k for k in samples[0].keys()
+                        if torch.is_tensor(samples[0][k])]
+                backend.write_attribute(f"{label}/keys", keys)
+                backend.write_attribute(f"{label}/ndata",
+                                        np.int64(len(samples)))
+                for k in keys:
+                    arr, counts, offsets, vdim = self._pack(samples, k)
+                    backend.write_array(f"{label}/{k}", arr)
+                    backend.write_array(f"{label}/{k}/variable_count",
+                                        counts)
+                    backend.write_array(f"{label}/{k}/variable_offset",
+                                        offsets)
+                    backend.write_array(f"{label}/{k}/variable_dim",
+                                        np.array([vdim], dtype=np.int64))
+
+
+class _Adios2StreamBackend:  # pragma: no cover - needs adios2
+    """Default AdiosWriter backend over adios2's high-level API."""
+
+    def __init__(self, filename: str):
+        import adios2
+        self._stream = adios2.Stream(filename, "w")
+
+    def __enter__(self):
+        return self
+
+    def __exit__(self, *a):
+        self._stream.close()
+
+    def write_attribute(self, name, value):
+        if isinstance(value, (list, tuple)) and value and \
+                isinstance(value[0], str):
+            self._stream.write_attribute(name, list(value))
+        else:
+            self._stream.write_attribute(name, np.asarray(value))
+
+    def write_array(self, name, arr):
+        arr = np.ascontiguousarray(arr)
+        self._stream.write(name, arr, shape=list(arr.shape),
+                           start=[0] * arr.ndim,
+                           count=list(arr.shape))
