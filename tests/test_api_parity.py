@@ -132,3 +132,48 @@ def test_top_level_package_surfaces():
               "save_model", "load_existing_model", "setup_log",
               "get_device", "distributed_model_wrapper"):
         assert hasattr(u, n), n
+
+
+def test_reference_named_aliases():
+    """Reference-named entry points that are aliases of unified
+    implementations resolve and are callable/usable."""
+    from hydragnn_amd.preprocess.batch_sampler import graph_node_cost
+    from hydragnn_amd.preprocess.graph_samples_checks_and_updates import (
+        check_if_graph_size_variable_mpi, gather_deg_dist, gather_deg_mpi)
+    from hydragnn_amd.train.train_validate_test import (
+        reduce_values_ranks_dist, reduce_values_ranks_mpi)
+    from hydragnn_amd.utils.config import (check_output_dim_consistent,
+                                           normalize_output_config,
+                                           update_config_NN_outputs,
+                                           update_config_edge_dim,
+                                           update_config_equivariance,
+                                           update_config_minmax)
+    from hydragnn_amd.utils.datasets.adios_reader import (
+        AdiosDataset, AdiosMultiDataset, AdiosWriter)
+    from hydragnn_amd.utils.datasets.download import download_file
+    from hydragnn_amd.utils.datasets.rawloaders import (
+        AbstractRawDataLoader, CFG_RawDataLoader, LSMS_RawDataLoader)
+    from hydragnn_amd.utils.descriptors_and_embeddings.smiles_utils import (
+        generate_graphdata_from_rdkit_molecule)
+    from hydragnn_amd.utils.lsms.lsms import compute_formation_enthalpy
+    from hydragnn_amd.utils.model.model import (calculate_PNA_degree_dist,
+                                                calculate_PNA_degree_mpi,
+                                                calculate_avg_deg_dist,
+                                                calculate_avg_deg_mpi)
+    from hydragnn_amd.utils.profiling_and_tracing.tracer import Tracer
+
+    for f in (graph_node_cost, gather_deg_dist, gather_deg_mpi,
+              reduce_values_ranks_dist, reduce_values_ranks_mpi,
+              calculate_PNA_degree_dist, calculate_PNA_degree_mpi,
+              calculate_avg_deg_dist, calculate_avg_deg_mpi,
+              check_if_graph_size_variable_mpi, download_file,
+              compute_formation_enthalpy, normalize_output_config,
+              update_config_minmax, update_config_NN_outputs,
+              update_config_edge_dim, update_config_equivariance,
+              check_output_dim_consistent,
+              generate_graphdata_from_rdkit_molecule):
+        assert callable(f), f
+    for c in (AdiosDataset, AdiosMultiDataset, AdiosWriter, Tracer,
+              AbstractRawDataLoader, CFG_RawDataLoader,
+              LSMS_RawDataLoader):
+        assert isinstance(c, type), c
