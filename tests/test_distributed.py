@@ -277,9 +277,12 @@ def test_ddp_wrapper_find_unused_tristate():
     from test_mace_model import _build, _mace_config
     from hydragnn_amd.utils.datasets.synthetic import md17_shape_dataset
 
-    os.environ.update(MASTER_ADDR="127.0.0.1", MASTER_PORT="29721",
-                      RANK="0", WORLD_SIZE="1", LOCAL_RANK="0")
-    dist.init_process_group("gloo", rank=0, world_size=1)
+    import tempfile
+    if dist.is_initialized():  # leftover group from another test
+        dist.destroy_process_group()
+    rdv = tempfile.NamedTemporaryFile(delete=False)
+    dist.init_process_group("gloo", rank=0, world_size=1,
+                            init_method=f"file://{rdv.name}")
     try:
         ds = md17_shape_dataset(num_samples=4)
         model, _, _ = _build(_mace_config(), ds)
