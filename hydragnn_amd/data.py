@@ -92,6 +92,31 @@ class Data:
     def to_dict(self) -> Dict[str, Any]:
         return dict(self._store)
 
+    @classmethod
+    def from_dict(cls, mapping: Dict[str, Any]) -> "Data":
+        out = cls()
+        for k, v in mapping.items():
+            out[k] = v
+        return out
+
+    @classmethod
+    def from_pyg(cls, data) -> "Data":
+        """Convert a torch_geometric ``Data``-like object (anything
+        exposing per-key tensor attributes via ``keys``/attribute
+        access — duck-typed, no PyG import) so reference-era datasets
+        drop straight into this framework's loaders."""
+        out = cls()
+        keys = data.keys() if callable(getattr(data, "keys", None)) \
+            else getattr(data, "keys", [])
+        for k in list(keys):
+            v = data[k] if hasattr(data, "__getitem__") \
+                else getattr(data, k)
+            out[k] = v
+        n = getattr(data, "num_nodes", None)
+        if n is not None:
+            out.num_nodes = int(n)
+        return out
+
     def clone(self) -> "Data":
         out = self.__class__()
         for k, v in self._store.items():

@@ -231,3 +231,32 @@ def test_grad_clip_norm_option():
     _t.nn.utils.clip_grad_norm_(m.parameters(), 0.5)
     total = sum(p.grad.norm() ** 2 for p in m.parameters()) ** 0.5
     assert float(total) <= 0.5 + 1e-4
+
+
+def test_data_from_pyg_ducktype():
+    """Data.from_pyg converts any PyG-shaped object (duck-typed keys +
+    item access) so reference-era datasets drop into our loaders."""
+    import torch
+
+    from hydragnn_amd.data import Batch, Data
+
+    class _PygLike:
+        def __init__(self):
+            self._d = {"x": torch.randn(5, 2),
+                       "edge_index": torch.zeros(2, 3, dtype=torch.long),
+                       "y": torch.randn(1, 1)}
+            self.num_nodes = 5
+
+        def keys(self):
+            return self._d.keys()
+
+        def __getitem__(self, k):
+            return self._d[k]
+
+    d = Data.from_pyg(_PygLike())
+    assert d.num_nodes == 5 and d.x.shape == (5, 2)
+    b = Batch.from_data_list([d, Data.from_pyg(_PygLike())])
+    assert b.num_graphs == 2 and b.num_nodes == 10
+
+    d2 = Data.from_dict({"x": torch.ones(3, 1)})
+    assert d2.x.shape == (3, 1)
