@@ -327,3 +327,39 @@ def test_config_driven_bf16_pure_precision():
     opt = torch.optim.AdamW(model.parameters(), lr=1e-3)
     err, _ = train(loader, model, opt, 0, precision="bf16_pure")
     assert torch.isfinite(err.float()).all()
+
+
+@pytest.mark.parametrize("n_samples,extra_nodes,extra_edges", [
+    (1, 0, 0),      # single-graph batch, minimum pad (2 nodes, 0 edges)
+    (1, 3, 1),      # pad edges on a tiny pad graph
+    (2, 0, 64),     # pad edges without extra nodes beyond the margin
+    (3, 17, 0),     # many pad nodes, zero pad edges
+])
+def test_pad_batch_edge_cases(n_samples, extra_nodes, extra_edges):
+    """Padding equivalence holds at the boundary shapes: minimum pad
+    graph (2 nodes), zero pad edges, single-sample batches."""
+    torch.manual_seed(11)
+    ds = md17_shape_dataset(num_samples=n_samples)
+    model = _small_mace(ds)
+    b1 = Batch.from_data_list([d.clone() for d in ds])
+    b2 = Batch.from_data_list([d.clone() for d in ds])
+    nc = b1.num_nodes + 2 + extra_nodes
+    ec = b1.num_edges + extra_edges
+    b2 = pad_batch_static(b2, nc, ec, pad_spacing=30.0)
+    assert b2.num_nodes == nc and b2.num_edges == ec
+    l1, t1, g1 = _loss_and_grads(model, b1)
+    l2, t2, g2 = _loss_and_grads(model, b2)
+    assert torch.isfinite(l2)
+    assert abs(float(l1) - float(l2)) < 1e-5 * max(1.0, abs(float(l1)))
+    assert (g1 - g2).abs().max() < 1e-5 * max(1.0, g1.abs().max())
+
+
+def test_pad_batch_rejects_impossible_caps():
+    ds = md17_shape_dataset(num_samples=2)
+    b = Batch.from_data_list(list(ds))
+    with pytest.raises(ValueError, match="node_cap"):
+        pad_batch_static(Batch.from_data_list(list(ds)),
+                         b.num_nodes + 1, b.num_edges + 8)
+    with pytest.raises(ValueError, match="edge_cap"):
+        pad_batch_static(Batch.from_data_list(list(ds)),
+                         b.num_nodes + 8, b.num_edges - 1)
