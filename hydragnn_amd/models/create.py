@@ -287,6 +287,12 @@ def _make_weighted_loss(loss_type: str, w: torch.Tensor):
             err = err * err
         elif lt in ("mae", "l1"):
             err = err.abs()
+        elif lt in ("smooth_l1", "huber"):
+            # beta/delta = 1.0, matching the nn module defaults
+            a = err.abs()
+            err = torch.where(a < 1.0, 0.5 * err * err, a - 0.5)
+        elif lt == "rmse":
+            err = err * err  # sqrt applied after the weighted mean
         else:
             raise NotImplementedError(
                 f"loss_weight_g not supported for loss '{loss_type}'")
@@ -294,7 +300,10 @@ def _make_weighted_loss(loss_type: str, w: torch.Tensor):
         while ww.dim() < err.dim():
             ww = ww.unsqueeze(-1)
         ww = ww.expand_as(err)
-        return (err * ww).sum() / ww.sum().clamp_min(1e-12)
+        out = (err * ww).sum() / ww.sum().clamp_min(1e-12)
+        if lt == "rmse":
+            out = torch.sqrt(out + 1e-12)
+        return out
 
     return fn
 

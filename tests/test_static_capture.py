@@ -363,3 +363,35 @@ def test_pad_batch_rejects_impossible_caps():
     with pytest.raises(ValueError, match="edge_cap"):
         pad_batch_static(Batch.from_data_list(list(ds)),
                          b.num_nodes + 8, b.num_edges - 1)
+
+
+@pytest.mark.parametrize("loss_type,ref_fn", [
+    ("mse", torch.nn.MSELoss()),
+    ("mae", torch.nn.L1Loss()),
+    ("smooth_l1", torch.nn.SmoothL1Loss()),
+    ("huber", torch.nn.HuberLoss()),
+])
+def test_weighted_loss_all_types(loss_type, ref_fn):
+    """All-ones weights reproduce the torch loss; zero-weight rows are
+    excluded exactly (pads contribute nothing for any loss type)."""
+    from hydragnn_amd.models.create import _make_weighted_loss
+    torch.manual_seed(5)
+    pred = torch.randn(6, 3)
+    true = torch.randn(6, 3)
+    ones = _make_weighted_loss(loss_type, torch.ones(6))
+    assert torch.allclose(ones(pred, true), ref_fn(pred, true),
+                          atol=1e-6)
+    w = torch.tensor([1.0, 1.0, 0.0, 1.0, 0.0, 1.0])
+    masked = _make_weighted_loss(loss_type, w)
+    keep = w.bool()
+    assert torch.allclose(masked(pred, true),
+                          ref_fn(pred[keep], true[keep]), atol=1e-6)
+
+
+def test_weighted_loss_rmse():
+    from hydragnn_amd.models.create import _make_weighted_loss
+    torch.manual_seed(6)
+    pred, true = torch.randn(5, 2), torch.randn(5, 2)
+    fn = _make_weighted_loss("rmse", torch.ones(5))
+    ref = torch.sqrt(torch.nn.functional.mse_loss(pred, true) + 1e-12)
+    assert torch.allclose(fn(pred, true), ref, atol=1e-6)
