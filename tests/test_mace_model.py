@@ -151,3 +151,32 @@ def test_mace_checkpointing_with_forces():
     loss.backward()
     grads = [p.grad for p in model.parameters() if p.grad is not None]
     assert grads and all(torch.isfinite(g).all() for g in grads)
+
+
+def test_single_atom_and_zero_edge_graphs():
+    """Edgeless graphs (single atoms) batch and train without NaNs —
+    the scatter/ETP chain must tolerate empty segments."""
+    import copy
+
+    from hydragnn_amd.data import Batch
+    from hydragnn_amd.utils.datasets.synthetic import md17_shape_dataset
+
+    ds = md17_shape_dataset(num_samples=3)
+    one = copy.deepcopy(ds[0])
+    one.x = one.x[:1]
+    one.z = one.z[:1]
+    one.pos = one.pos[:1]
+    one.edge_index = torch.zeros(2, 0, dtype=torch.long)
+    one.forces = one.forces[:1] * 0
+    one.energy = one.energy * 0
+    one.y = one.y * 0
+    one.num_nodes = 1
+    model, config, _ = _build(_mace_config(), ds + [one])
+    b = Batch.from_data_list(list(ds) + [one])
+    b.pos.requires_grad_(True)
+    pred = model(b)
+    loss, _ = model.energy_force_loss(pred, b, create_graph=False)
+    loss.backward()
+    assert torch.isfinite(loss)
+    assert all(torch.isfinite(p.grad).all() for p in model.parameters()
+               if p.grad is not None)
