@@ -11,6 +11,12 @@ testable against torch.optim.AdamW without a GPU.
 
 Construct BEFORE any DDP wrapping so reducer bucket views are built
 over the flattened storages.
+
+Role parity: the reference trains with stock torch.optim selected by
+``utils/optimizer`` (reference hydragnn/utils/optimizer/optimizer.py);
+this fused optimizer is the MI355X-native drop-in used by the
+launch-bound captured path (select_optimizer still provides every
+reference optimizer type).
 """
 
 from __future__ import annotations
