@@ -80,6 +80,9 @@ class Data:
     def __setitem__(self, key: str, value: Any) -> None:
         self._store[key] = value
 
+    def __delitem__(self, key: str) -> None:
+        del self._store[key]
+
     def get(self, key: str, default: Any = None) -> Any:
         return self._store.get(key, default)
 

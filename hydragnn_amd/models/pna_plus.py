@@ -86,6 +86,11 @@ class _PNAPlusWrapper(nn.Module):
 
 
 class PNAPlusStack(Base):
+    # shape-static forward: uses only the provided edge_index, and the
+    # Bessel envelope zeroes edges beyond the cutoff, so padded static
+    # batches replay exactly (pad edges are all longer than cutoff)
+    _hipgraph_capture_safe = True
+
     def __init__(self, deg: List[int], edge_dim: Optional[int] = None,
                  envelope_exponent: Optional[int] = None,
                  num_radial: Optional[int] = None,

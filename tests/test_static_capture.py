@@ -395,3 +395,40 @@ def test_weighted_loss_rmse():
     fn = _make_weighted_loss("rmse", torch.ones(5))
     ref = torch.sqrt(torch.nn.functional.mse_loss(pred, true) + 1e-12)
     assert torch.allclose(fn(pred, true), ref, atol=1e-6)
+
+
+def test_pnaplus_pad_equivalence():
+    """PNAPlus is capture-safe: the Bessel envelope zeroes pad edges
+    (all longer than cutoff), so padded batches match unpadded ones."""
+    import sys
+    sys.path.insert(0, os.path.dirname(__file__))
+    from deterministic_graph_data import base_config, make_deterministic_dataset
+    from hydragnn_amd.models import create_model_config
+    from hydragnn_amd.preprocess import create_dataloaders
+    from hydragnn_amd.utils.config import update_config
+
+    torch.manual_seed(2)
+    cfg = base_config("PNAPlus", heads=("graph",))
+    ds = make_deterministic_dataset(num_samples=4, num_heads_node=0,
+                                    include_graph_head=True)
+    loaders = create_dataloaders(ds, ds, ds, 4, config=cfg)
+    cfg = update_config(cfg, *loaders)
+    model = create_model_config(cfg["NeuralNetwork"], use_gpu=False)
+    assert model.supports_hipgraph_capture
+
+    def strip(d):
+        d = d.clone()
+        if "y_loc" in d.keys():
+            delattr(d, "y_loc")
+        return d
+
+    b1 = Batch.from_data_list([strip(d) for d in ds])
+    b2 = Batch.from_data_list([strip(d) for d in ds])
+    nc, ec = compute_static_caps(ds, 4)
+    b2 = pad_batch_static(b2, nc + 4, ec + 9, pad_spacing=30.0)
+    model.eval()
+    with torch.no_grad():
+        p1 = model(b1)
+        p2 = model(b2)
+    # real-graph predictions bitwise unaffected by padding
+    assert torch.allclose(p1[0], p2[0][: p1[0].shape[0]], atol=1e-6)
