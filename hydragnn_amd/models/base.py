@@ -315,6 +315,11 @@ class Base(Module):
         self.num_branches = 1
         if "graph" in self.config_heads:
             self.num_branches = len(self.config_heads["graph"])
+            if self.num_branches > 1:
+                # multi-branch forwards partition by data.dataset_name
+                # with boolean masks (data-dependent shapes) — never
+                # replay them from a captured hipGraph
+                self.supports_hipgraph_capture = False
             for branchdict in self.config_heads["graph"]:
                 arch = branchdict["architecture"]
                 dim_shared = arch["dim_sharedlayers"]

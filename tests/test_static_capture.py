@@ -432,3 +432,30 @@ def test_pnaplus_pad_equivalence():
         p2 = model(b2)
     # real-graph predictions bitwise unaffected by padding
     assert torch.allclose(p1[0], p2[0][: p1[0].shape[0]], atol=1e-6)
+
+
+def test_multibranch_disables_capture():
+    """Multi-branch (multidataset) models partition by dataset_name
+    with data-dependent masks — the capture gate must exclude them."""
+    import sys
+    sys.path.insert(0, os.path.dirname(__file__))
+    from deterministic_graph_data import base_config, make_deterministic_dataset
+    from hydragnn_amd.models import create_model_config
+    from hydragnn_amd.preprocess import create_dataloaders
+    from hydragnn_amd.utils.config import update_config
+
+    cfg = base_config("GIN", heads=("graph",))
+    heads = cfg["NeuralNetwork"]["Architecture"]["output_heads"]
+    base_branch = heads["graph"]
+    # two branches -> multidataset-style branched heads
+    if isinstance(base_branch, dict):
+        heads["graph"] = [
+            {"type": f"branch-{i}", "architecture": dict(base_branch)}
+            for i in range(2)]
+    ds = make_deterministic_dataset(num_samples=4, num_heads_node=0,
+                                    include_graph_head=True)
+    loaders = create_dataloaders(ds, ds, ds, 2, config=cfg)
+    cfg = update_config(cfg, *loaders)
+    model = create_model_config(cfg["NeuralNetwork"], use_gpu=False)
+    assert model.num_branches == 2
+    assert not model.supports_hipgraph_capture
