@@ -401,6 +401,9 @@ class Base(Module):
 
     def enable_conv_checkpointing(self):
         self.conv_checkpointing = True
+        # torch checkpointing's recompute-in-backward + RNG stashing
+        # is not replayable from a captured hipGraph
+        self.supports_hipgraph_capture = False
 
     # ------------------------------------------------------------------
     # forward

@@ -459,3 +459,12 @@ def test_multibranch_disables_capture():
     model = create_model_config(cfg["NeuralNetwork"], use_gpu=False)
     assert model.num_branches == 2
     assert not model.supports_hipgraph_capture
+
+
+def test_conv_checkpointing_disables_capture():
+    ds = md17_shape_dataset(num_samples=2)
+    model = _small_mace(ds)
+    core = model
+    assert core.supports_hipgraph_capture
+    core.enable_conv_checkpointing()
+    assert not core.supports_hipgraph_capture
