@@ -468,3 +468,27 @@ def test_conv_checkpointing_disables_capture():
     assert core.supports_hipgraph_capture
     core.enable_conv_checkpointing()
     assert not core.supports_hipgraph_capture
+
+
+def test_capture_safety_marker_matrix():
+    """Class-level capture-safety markers: every shape-static stack
+    opts in; dynamic-shape stacks stay out."""
+    from hydragnn_amd.models.egnn import EGCLStack
+    from hydragnn_amd.models.mace.stack import MACEStack
+    from hydragnn_amd.models.painn import PAINNStack
+    from hydragnn_amd.models.pna_plus import PNAPlusStack
+    from hydragnn_amd.models.pnaeq import PNAEqStack
+    from hydragnn_amd.models.stacks import (CGCNNStack, GATStack,
+                                            GINStack, MFCStack, PNAStack,
+                                            SAGEStack)
+    from hydragnn_amd.models.schnet import SCFStack
+    from hydragnn_amd.models.dimenet import DIMEStack
+
+    safe = (MACEStack, EGCLStack, PAINNStack, PNAEqStack, PNAPlusStack,
+            GINStack, SAGEStack, MFCStack, CGCNNStack, PNAStack,
+            GATStack)
+    unsafe = (SCFStack, DIMEStack)
+    for cls in safe:
+        assert cls._hipgraph_capture_safe, cls.__name__
+    for cls in unsafe:
+        assert not cls._hipgraph_capture_safe, cls.__name__
