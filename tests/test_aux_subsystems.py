@@ -327,3 +327,64 @@ def test_xyz_to_graph_bond_perception():
     dst = d.edge_index[1]
     assert bool((dst[1:] >= dst[:-1]).all())
     assert d.edge_attr.shape == (8, 2)
+
+
+def test_oversampling_dataloader():
+    """oversampling=True draws num_samples per epoch with replacement
+    (reference load_data.py RandomSampler path)."""
+    import sys as _sys
+    _sys.path.insert(0, os.path.dirname(__file__))
+    from deterministic_graph_data import make_deterministic_dataset
+    from hydragnn_amd.preprocess import create_dataloaders
+
+    ds = make_deterministic_dataset(num_samples=6, num_heads_node=0)
+    tr, va, te = create_dataloaders(ds, ds, ds, batch_size=4,
+                                    oversampling=True, num_samples=20)
+    seen = sum(b.num_graphs for b in tr)
+    assert seen == 20
+
+
+def _zero_worker(rank, world, port, q):
+    try:
+        os.environ.update(MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port),
+                          RANK=str(rank), WORLD_SIZE=str(world),
+                          LOCAL_RANK=str(rank))
+        import torch.distributed as dist
+
+        from hydragnn_amd.utils.optimizer import select_optimizer
+        dist.init_process_group("gloo", rank=rank, world_size=world)
+        torch.manual_seed(0)
+        m = torch.nn.Linear(4, 4)
+        opt = select_optimizer(m, {"type": "AdamW",
+                                   "learning_rate": 1e-3,
+                                   "use_zero_redundancy": True})
+        from torch.distributed.optim import ZeroRedundancyOptimizer
+        assert isinstance(opt, ZeroRedundancyOptimizer)
+        torch.manual_seed(10 + rank)
+        m(torch.randn(2, 4)).sum().backward()
+        opt.step()
+        opt.consolidate_state_dict(to=0)
+        if rank == 0:
+            sd = opt.state_dict()
+            assert "state" in sd and len(sd["state"]) == 2
+        dist.destroy_process_group()
+        q.put((rank, True, ""))
+    except Exception as e:  # pragma: no cover
+        q.put((rank, False, repr(e)))
+
+
+def test_zero_redundancy_optimizer_two_rank():
+    """use_zero_redundancy shards optimizer state at world > 1 and
+    consolidates for checkpointing (reference optimizer.py:53-111)."""
+    import multiprocessing as mp
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    ps = [ctx.Process(target=_zero_worker, args=(r, 2, 29731, q))
+          for r in range(2)]
+    for p in ps:
+        p.start()
+    results = [q.get(timeout=300) for _ in range(2)]
+    for p in ps:
+        p.join(timeout=60)
+    for rank, ok, info in results:
+        assert ok, f"rank {rank}: {info}"
