@@ -67,7 +67,10 @@ def run_training(config: Union[str, dict], dataset=None,
     # restart support (reference model.py:204-211)
     load_existing_model_config(
         model, config["NeuralNetwork"]["Training"], optimizer=optimizer)
-    model = distributed_model_wrapper(model, verbosity=verbosity)
+    model = distributed_model_wrapper(
+        model, verbosity=verbosity,
+        sync_batch_norm=config["NeuralNetwork"]["Architecture"].get(
+            "SyncBatchNorm", False))
     scheduler = torch.optim.lr_scheduler.ReduceLROnPlateau(
         optimizer, mode="min", factor=0.5, patience=5)
     writer = get_summary_writer(log_name)
