@@ -60,6 +60,11 @@ def create_dataloaders(trainset, valset, testset, batch_size: int,
     if config is not None:
         batching = config["NeuralNetwork"]["Training"].get("Batching")
 
+    if oversampling and batching is not None and \
+            batching.get("mode") == "node_budget":
+        raise ValueError(
+            "cost-aware batching cannot be combined with oversampling")
+
     use_dist = dist.is_initialized() and dist.get_world_size() > 1
 
     use_custom = os.getenv("HYDRAGNN_CUSTOM_DATALOADER", "0") == "1"
